@@ -1,0 +1,223 @@
+"""Solver — the trainer/solver loop: epoch loop,
+per-iter poly LR, (loss-scaled) fp16/bf16 step, custom gradient engine,
+checkpoint save/resume, distributed eval with MAE/F-measure.
+
+One step = load -> fwd -> hybrid loss -> bwd (buckets all-reduce on the side
+stream as they fill) -> fused SGD.  The 1/world average and any loss-scale
+unscale fold into the SGD grad_scale.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from .. import ops
+from ..config import Config
+from ..data import create_loader
+from ..losses import HybridLoss
+from ..metrics import MAEMeter, PRFMeter
+from ..models import build_model, convert_precision, set_syncbn
+from ..parallel import GradReducer, setup, get_rank, get_world_size
+from ..utils import AvgMeter, Timer, get_logger, seed_everything
+from ..utils.png import write_gray_png
+from ..utils.recorder import ResultsRecorder, TBRecorder
+from .checkpoint import exp_paths, load_checkpoint, make_exp_tree, save_checkpoint
+
+_DTYPES = {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32}
+
+
+class Solver:
+    def __init__(self, cfg: Config):
+        self.cfg = cfg
+        self.device = setup(cfg.backend, cfg.device)
+        self.rank = get_rank()
+        self.world = get_world_size()
+        self.distributed = self.world > 1
+        self.dtype = _DTYPES[cfg.dtype]
+        self.is_gpu = self.device.type == "cuda"
+
+        self.paths = exp_paths(cfg.output_root, cfg.exp_name)
+        if self.rank == 0:
+            make_exp_tree(self.paths)
+            with open(self.paths["cfg"], "w") as f:
+                f.write(cfg.to_json())
+        self.log = get_logger("sodkit", self.paths["log"] if self.rank == 0 else None)
+
+        # identical seed on every rank -> identical random init
+        seed_everything(cfg.seed)
+        model = build_model(cfg.model)
+        if self.is_gpu:
+            model = model.to(self.device)
+            if cfg.channels_last:
+                model = model.to(memory_format=torch.channels_last)
+            if self.dtype != torch.float32:
+                convert_precision(model, self.dtype)
+        if cfg.sync_bn and self.distributed:
+            set_syncbn(model, None)  # default process group
+        self.model = model
+
+        self.reducer = GradReducer(model, bucket_mb=cfg.bucket_mb,
+                                   overlap=cfg.overlap_comm)
+        self.reducer.broadcast_params(model)
+
+        decay, no_decay = ops.split_decay_groups(model)
+        groups = [{"params": decay, "weight_decay": cfg.weight_decay}]
+        if cfg.sgd_trick:
+            groups.append({"params": no_decay, "weight_decay": 0.0})
+        else:
+            groups[0]["params"] = decay + no_decay
+        self.optimizer = ops.SodSGD(groups, momentum=cfg.momentum,
+                                    weight_decay=cfg.weight_decay,
+                                    nesterov=cfg.nesterov)
+
+        self.criterion = HybridLoss(cfg.w_bce, cfg.w_iou, cfg.w_ssim)
+        self.loss_scale = cfg.loss_scale if cfg.dtype == "fp16" else 1.0
+
+        self.start_epoch = 0
+        self.iter_count = 0
+        self.total_iters = 1
+        self.tb = None
+
+    # ------------------------------------------------------------------
+    def _lr_at(self, it: int) -> float:
+        cfg = self.cfg
+        if cfg.warmup_iters and it < cfg.warmup_iters:
+            return cfg.lr * (it + 1) / cfg.warmup_iters
+        if cfg.lr_type == "poly":
+            frac = min(it, self.total_iters - 1) / max(self.total_iters, 1)
+            return cfg.lr * (1.0 - frac) ** cfg.lr_power
+        return cfg.lr
+
+    def _to_device(self, t: torch.Tensor, cast: bool) -> torch.Tensor:
+        t = t.to(self.device, non_blocking=True)
+        if cast and self.is_gpu and self.dtype != torch.float32:
+            t = t.to(self.dtype)
+        if self.is_gpu and self.cfg.channels_last and t.ndim == 4:
+            t = t.contiguous(memory_format=torch.channels_last)
+        return t
+
+    # ------------------------------------------------------------------
+    def step(self, img: torch.Tensor, mask: torch.Tensor,
+             depth: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """One optimization step on already-on-device tensors.  Returns the
+        (detached, device-side) unscaled loss."""
+        self.reducer.zero_grad()
+        logits = self.model(img, depth) if depth is not None else self.model(img)
+        loss = self.criterion(logits, mask)
+        (loss * self.loss_scale).backward() if self.loss_scale != 1.0 else loss.backward()
+        self.reducer.finalize()
+        lr = self._lr_at(self.iter_count)
+        self.optimizer.step(lr, self.reducer.grad_scale / self.loss_scale)
+        self.iter_count += 1
+        return loss.detach()
+
+    # ------------------------------------------------------------------
+    def train(self) -> None:
+        cfg = self.cfg
+        loader, sampler = create_loader(cfg, "train", self.distributed,
+                                        self.rank, self.world)
+        self.total_iters = max(1, cfg.epoch_num * len(loader))
+        if cfg.resume and os.path.exists(self.paths["checkpoint_final"]):
+            self.start_epoch = load_checkpoint(self.paths, self.model,
+                                               self.optimizer,
+                                               map_location=self.device)
+            self.iter_count = self.start_epoch * len(loader)
+            self.log.info(f"resumed at epoch {self.start_epoch}")
+        if self.rank == 0 and cfg.tb:
+            self.tb = TBRecorder(self.paths["tb"], enabled=True)
+
+        self.model.train()
+        loss_meter = AvgMeter()
+        ips_meter = AvgMeter()
+        for epoch in range(self.start_epoch, cfg.epoch_num):
+            if sampler is not None:
+                sampler.set_epoch(epoch)
+            loader.dataset.set_epoch(epoch)
+            t_last = time.perf_counter()
+            for i, batch in enumerate(loader):
+                if len(batch) == 4:
+                    img, depth, mask, _ = batch
+                    depth = self._to_device(depth, cast=True)
+                else:
+                    img, mask, _ = batch
+                    depth = None
+                img = self._to_device(img, cast=True)
+                mask = self._to_device(mask, cast=False)
+                loss = self.step(img, mask, depth)
+                now = time.perf_counter()
+                ips = img.shape[0] * self.world / (now - t_last)
+                t_last = now
+                ips_meter.update(ips)
+                if (i + 1) % cfg.log_interval == 0 or i + 1 == len(loader):
+                    loss_meter.update(loss.item())
+                    self.log.info(
+                        f"ep {epoch} it {i + 1}/{len(loader)} "
+                        f"loss {loss_meter.val:.4f} (avg {loss_meter.avg:.4f}) "
+                        f"lr {self._lr_at(self.iter_count - 1):.5f} "
+                        f"imgs/s {ips_meter.avg:.1f}")
+                    if self.tb is not None:
+                        self.tb.scalar("train/loss", loss_meter.val, self.iter_count)
+                        self.tb.scalar("train/lr", self._lr_at(self.iter_count - 1),
+                                       self.iter_count)
+            if self.rank == 0:
+                save_checkpoint(self.paths, epoch, self.model, self.optimizer,
+                                {"loss_scale": self.loss_scale})
+            self._barrier()
+        if self.tb is not None:
+            self.tb.close()
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def test(self, save_png: bool = True, max_png: int = 16) -> dict:
+        """Distributed eval on the fixed-seed synthetic val split (SURVEY
+        §3.2): MAE + F-measure with cross-rank reduction, PNG saves."""
+        cfg = self.cfg
+        loader, _ = create_loader(cfg, "val", self.distributed,
+                                  self.rank, self.world)
+        self.model.eval()
+        mae = MAEMeter()
+        prf = PRFMeter()
+        n_png = 0
+        pre_dir = os.path.join(self.paths["pre"], "synthetic_val")
+        if self.rank == 0 and save_png:
+            os.makedirs(pre_dir, exist_ok=True)
+        for batch in loader:
+            if len(batch) == 4:
+                img, depth, mask, names = batch
+                depth = self._to_device(depth, cast=True)
+            else:
+                img, mask, names = batch
+                depth = None
+            img = self._to_device(img, cast=True)
+            mask = self._to_device(mask, cast=False)
+            logits = self.model(img, depth) if depth is not None else self.model(img)
+            pred = torch.sigmoid(logits.float())
+            mae.update(pred, mask)
+            prf.update(pred, mask)
+            if self.rank == 0 and save_png and n_png < max_png:
+                for b in range(min(pred.shape[0], max_png - n_png)):
+                    write_gray_png(os.path.join(pre_dir, f"{names[b]}.png"),
+                                   pred[b, 0].cpu().numpy())
+                    n_png += 1
+        res = {"mae": mae.compute(self.distributed),
+               **{k: v for k, v in prf.compute(self.distributed).items()
+                  if k in ("maxf", "meanf")}}
+        if self.rank == 0:
+            rec = ResultsRecorder(self.paths["root"])
+            rec.add("synthetic_val", res)
+            rec.save()
+            self.log.info(f"eval: MAE {res['mae']:.4f} maxF {res['maxf']:.4f} "
+                          f"meanF {res['meanf']:.4f}")
+        self._barrier()
+        self.model.train()
+        return res
+
+    # ------------------------------------------------------------------
+    def _barrier(self) -> None:
+        if self.distributed and dist.is_initialized():
+            dist.barrier()

@@ -1,0 +1,95 @@
+"""Fused BatchNorm(+residual add)(+ReLU) — SURVEY §2.3 K3 + K8 epilogue.
+
+GPU path: hand-written HIP kernels (csrc/bnrelu.hip):
+  * ``bn_stats``      one-pass per-channel sum / sumsq reduction (fp32)
+  * ``bn_apply``      normalize + scale + (add residual) + (ReLU), one pass
+  * ``bn_bwd_reduce`` per-channel sum(g), sum(g*xhat) with g = dy·relu-mask
+  * ``bn_bwd_apply``  dx (+ optional residual grad) in one pass
+
+SyncBN is the same op with a process group: the per-channel
+stats (and backward reduction terms) are all-reduced once per call as a single
+coalesced [2C+1] buffer — no per-layer small-message storm beyond one
+all-reduce per BN layer, and the kernel path is identical.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from . import reference
+from ._backend import ext
+
+_CL = torch.channels_last
+
+
+def _cl(x: torch.Tensor) -> torch.Tensor:
+    return x.contiguous(memory_format=_CL)
+
+
+class _BNActFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training,
+                momentum, eps, relu, residual, process_group):
+        x = _cl(x)
+        if residual is not None:
+            residual = _cl(residual)
+        C = x.shape[1]
+        n_local = x.numel() // C
+        if training:
+            stats = ext().bn_stats(x)                      # fp32 [2, C]
+            n_total = float(n_local)
+            if process_group is not None and dist.is_initialized() \
+                    and dist.get_world_size(process_group) > 1:
+                packed = torch.cat([stats.view(-1),
+                                    stats.new_tensor([n_total])])
+                dist.all_reduce(packed, group=process_group)
+                stats = packed[:2 * C].view(2, C)
+                n_total = packed[2 * C].item()
+            mean = stats[0] / n_total
+            var = stats[1] / n_total - mean * mean
+            with torch.no_grad():
+                unbiased = var * (n_total / max(n_total - 1.0, 1.0))
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+        else:
+            mean = running_mean
+            var = running_var
+            n_total = 0.0                                   # eval marker
+        invstd = torch.rsqrt(var + eps)
+        y = ext().bn_apply(x, mean, invstd, gamma, beta, relu,
+                           residual if residual is not None else torch.Tensor())
+        ctx.save_for_backward(x, y, mean, invstd, gamma)
+        ctx.meta = (relu, residual is not None, n_total, process_group)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y, mean, invstd, gamma = ctx.saved_tensors
+        relu, has_res, n_total, pg = ctx.meta
+        dy = _cl(dy)
+        sums = ext().bn_bwd_reduce(dy, y, x, mean, invstd, relu)  # fp32 [2, C]
+        if pg is not None and dist.is_initialized() and dist.get_world_size(pg) > 1:
+            dist.all_reduce(sums, group=pg)
+        dx, dres = ext().bn_bwd_apply(dy, y, x, mean, invstd, gamma, sums,
+                                      n_total, relu, has_res)
+        dgamma = sums[1].to(gamma.dtype)
+        dbeta = sums[0].to(gamma.dtype)
+        return (dx, dgamma, dbeta, None, None, None, None, None, None,
+                dres if has_res else None, None)
+
+
+def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+           running_mean: torch.Tensor, running_var: torch.Tensor,
+           training: bool, momentum: float = 0.1, eps: float = 1e-5,
+           relu: bool = True, residual: Optional[torch.Tensor] = None,
+           process_group=None) -> torch.Tensor:
+    if x.is_cuda:
+        return _BNActFn.apply(x, gamma, beta, running_mean, running_var,
+                              training, momentum, eps, relu, residual,
+                              process_group)
+    return reference.bn_act(x, gamma, beta, running_mean, running_var,
+                            training, momentum, eps, relu, residual,
+                            process_group)

@@ -1,0 +1,132 @@
+"""SodSGD — SGD-with-momentum + fp32 master weights, fused multi-tensor HIP
+kernel on GPU (one launch sweeps every parameter chunk;
+the bf16->fp32 grad upcast, 1/world grad average, loss-scale unscale, weight
+decay, momentum and the fp32->bf16 param write-back all fold into it).
+
+"sgd_trick" parity (reference config): no weight decay on bias / BN params —
+callers pass two param groups via ``split_decay_groups``.
+
+CPU path: per-param reference math (ops.reference.sgd_step).
+"""
+
+from __future__ import annotations
+
+from typing import Iterable
+
+import torch
+
+from . import reference
+from ._backend import ext
+
+_CHUNK = 1 << 21  # elements per kernel block-chunk
+
+
+def split_decay_groups(model: torch.nn.Module):
+    """Return (decay, no_decay) parameter lists: biases and BN gamma/beta
+    (ndim <= 1) get no weight decay."""
+    decay, no_decay = [], []
+    for name, p in model.named_parameters():
+        if not p.requires_grad:
+            continue
+        (no_decay if p.ndim <= 1 else decay).append(p)
+    return decay, no_decay
+
+
+class SodSGD:
+    """Not a torch.optim.Optimizer subclass on purpose: the fused step needs
+    a per-step (lr, grad_scale) and owns its flat table layout."""
+
+    def __init__(self, param_groups, momentum: float = 0.9,
+                 weight_decay: float = 5e-4, nesterov: bool = False):
+        # param_groups: list of dicts {params: [...], weight_decay: float}
+        self.groups = []
+        for g in param_groups:
+            params = [p for p in g["params"] if p.requires_grad]
+            self.groups.append({"params": params,
+                                "weight_decay": float(g.get("weight_decay", weight_decay))})
+        self.momentum = momentum
+        self.nesterov = nesterov
+        self.state: dict[torch.nn.Parameter, dict] = {}
+        for g in self.groups:
+            for p in g["params"]:
+                self.state[p] = {
+                    "master": p.detach().to(torch.float32).clone(),
+                    "momentum_buffer": torch.zeros_like(p, dtype=torch.float32),
+                }
+        self._tables = None  # lazily built device pointer tables (GPU)
+
+    # -- fused GPU tables ---------------------------------------------------
+    def _build_tables(self):
+        dev = None
+        ptrs, metas = [], []
+        for g in self.groups:
+            wd = g["weight_decay"]
+            for p in g["params"]:
+                assert p.grad is not None, "grads must be pre-assigned (reducer bucket views)"
+                st = self.state[p]
+                dev = p.device
+                ptrs.append((p.data_ptr(), p.grad.data_ptr(),
+                             st["master"].data_ptr(),
+                             st["momentum_buffer"].data_ptr()))
+                metas.append((p.numel(), wd))
+        ptr_t = torch.tensor(ptrs, dtype=torch.int64, device=dev)        # [T,4]
+        numel_t = torch.tensor([m[0] for m in metas], dtype=torch.int64, device=dev)
+        wd_t = torch.tensor([m[1] for m in metas], dtype=torch.float32, device=dev)
+        chunks = []
+        for ti, (n, _) in enumerate(metas):
+            for off in range(0, n, _CHUNK):
+                chunks.append((ti, off))
+        chunk_t = torch.tensor(chunks, dtype=torch.int64, device=dev)    # [K,2]
+        self._tables = (ptr_t, numel_t, wd_t, chunk_t)
+
+    @torch.no_grad()
+    def step(self, lr: float, grad_scale: float = 1.0):
+        first = next((p for g in self.groups for p in g["params"]), None)
+        if first is None:
+            return
+        if first.is_cuda:
+            if self._tables is None:
+                self._build_tables()
+            ptr_t, numel_t, wd_t, chunk_t = self._tables
+            # dtype code: 0=fp32 1=bf16 2=fp16 (uniform across params)
+            dt = {torch.float32: 0, torch.bfloat16: 1, torch.float16: 2}[first.dtype]
+            ext().sgd_fused(ptr_t, numel_t, wd_t, chunk_t, dt, float(lr),
+                            self.momentum, float(grad_scale), self.nesterov)
+        else:
+            for g in self.groups:
+                for p in g["params"]:
+                    if p.grad is None:
+                        continue
+                    st = self.state[p]
+                    reference.sgd_step(p.data, p.grad, st["master"],
+                                       st["momentum_buffer"], lr, self.momentum,
+                                       g["weight_decay"], grad_scale, self.nesterov)
+
+    # -- checkpointing (bitwise resume of optimizer state) -------------------
+    def state_dict(self) -> dict:
+        flat = []
+        for g in self.groups:
+            for p in g["params"]:
+                st = self.state[p]
+                flat.append({"master": st["master"],
+                             "momentum_buffer": st["momentum_buffer"]})
+        return {"momentum": self.momentum, "nesterov": self.nesterov,
+                "param_state": flat,
+                "group_wd": [g["weight_decay"] for g in self.groups]}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.momentum = sd["momentum"]
+        self.nesterov = sd["nesterov"]
+        it = iter(sd["param_state"])
+        for g in self.groups:
+            for p in g["params"]:
+                st = next(it)
+                self.state[p]["master"].copy_(st["master"])
+                self.state[p]["momentum_buffer"].copy_(st["momentum_buffer"])
+        for g, wd in zip(self.groups, sd["group_wd"]):
+            g["weight_decay"] = wd
+        self._tables = None
+
+    def params(self) -> Iterable[torch.nn.Parameter]:
+        for g in self.groups:
+            yield from g["params"]

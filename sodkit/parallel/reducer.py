@@ -1,0 +1,168 @@
+"""GradReducer — the hand-built data-parallel gradient engine (replaces
+torch.nn.parallel.DistributedDataParallel).
+
+Design, MI355X-first:
+  * Parameters are grouped in REVERSE registration order (≈ backward arrival
+    order) into flat buckets (default 25 MB, sized for RCCL over xGMI — with
+    ~2 ms of total gradient traffic per step on a fully connected 8-GPU node,
+    overlap correctness matters more than bucket heroics).
+  * Each param's ``.grad`` IS a view into its bucket slot — autograd
+    accumulates straight into the flat buffer; no copy, no regather
+    (gradient_as_bucket_view by construction).
+  * ``register_post_accumulate_grad_hook`` counts arrivals; when a bucket is
+    complete its all-reduce launches immediately on a SIDE HIP stream (the
+    comm stream waits on an event recorded on the compute stream), so
+    communication overlaps the rest of backward.
+  * ``finalize()`` makes the compute stream wait for every in-flight
+    collective.  The 1/world_size average is NOT applied here — it folds
+    into the fused SGD's grad_scale (one fewer pass over the grads).
+  * On CPU (gloo) the same engine runs with the stream layer inert — one
+    code path, the [BJ:7]/multi-process-CPU tests exercise exactly the logic
+    the GPU runs.
+
+Init-time C2 broadcast of rank-0 parameters is provided (`broadcast_params`).
+An ASSERT mode (SODKIT_COMM_ASSERT=1) checksums buckets around each
+all-reduce (SURVEY §5.2 race detection).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+
+def _world(pg) -> int:
+    if dist.is_initialized():
+        return dist.get_world_size(pg)
+    return 1
+
+
+class _Bucket:
+    __slots__ = ("params", "flat", "arrived", "work", "ready_event", "index")
+
+    def __init__(self, index: int):
+        self.index = index
+        self.params: list[torch.nn.Parameter] = []
+        self.flat: Optional[torch.Tensor] = None
+        self.arrived = 0
+        self.work = None
+        self.ready_event = None
+
+
+class GradReducer:
+    def __init__(self, model: torch.nn.Module, process_group=None,
+                 bucket_mb: int = 25, overlap: bool = True):
+        self.pg = process_group
+        self.overlap = overlap and torch.cuda.is_available()
+        self.assert_mode = os.environ.get("SODKIT_COMM_ASSERT", "0") == "1"
+        params = [p for p in model.parameters() if p.requires_grad]
+        params.reverse()  # reverse registration order ≈ backward arrival order
+        self._params = params
+
+        # --- bucket assignment ---
+        bucket_bytes = bucket_mb * (1 << 20)
+        self.buckets: list[_Bucket] = []
+        cur = _Bucket(0)
+        cur_bytes = 0
+        for p in params:
+            nbytes = p.numel() * p.element_size()
+            if cur.params and cur_bytes + nbytes > bucket_bytes:
+                self.buckets.append(cur)
+                cur = _Bucket(len(self.buckets))
+                cur_bytes = 0
+            cur.params.append(p)
+            cur_bytes += nbytes
+        if cur.params:
+            self.buckets.append(cur)
+
+        # --- flat storage + grad views ---
+        self._p2bucket: dict[torch.nn.Parameter, _Bucket] = {}
+        for b in self.buckets:
+            total = sum(p.numel() for p in b.params)
+            dev = b.params[0].device
+            dt = b.params[0].dtype
+            b.flat = torch.zeros(total, device=dev, dtype=dt)
+            off = 0
+            for p in b.params:
+                p.grad = b.flat[off:off + p.numel()].view_as(p)
+                off += p.numel()
+                self._p2bucket[p] = b
+
+        self.comm_stream = (torch.cuda.Stream()
+                            if self.overlap and _world(self.pg) > 1 else None)
+        self._hooks = []
+        for p in params:
+            self._hooks.append(
+                p.register_post_accumulate_grad_hook(self._on_grad_ready))
+
+    # ------------------------------------------------------------------
+    def broadcast_params(self, model: torch.nn.Module) -> None:
+        """C2: rank-0 params + buffers to all (belt-and-braces on top of the
+        identical-seed init)."""
+        if _world(self.pg) <= 1:
+            return
+        for t in list(model.parameters()) + list(model.buffers()):
+            dist.broadcast(t.data, src=0, group=self.pg)
+
+    # ------------------------------------------------------------------
+    def _on_grad_ready(self, p: torch.nn.Parameter) -> None:
+        b = self._p2bucket[p]
+        b.arrived += 1
+        if b.arrived == len(b.params):
+            self._launch(b)
+
+    def _launch(self, b: _Bucket) -> None:
+        if _world(self.pg) <= 1:
+            return
+        if self.assert_mode:
+            before = b.flat.float().sum().item()
+        if self.comm_stream is not None:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            with torch.cuda.stream(self.comm_stream):
+                self.comm_stream.wait_event(ev)
+                b.work = dist.all_reduce(b.flat, group=self.pg, async_op=True)
+        else:
+            b.work = dist.all_reduce(b.flat, group=self.pg, async_op=True)
+        if self.assert_mode:
+            if b.work is not None:
+                b.work.wait()
+                b.work = None
+            after = b.flat.float().sum().item()
+            w = _world(self.pg)
+            if not (abs(after) <= abs(before) * w * 4 + 1e3):  # sanity bound
+                raise RuntimeError(
+                    f"comm assert: bucket {b.index} sum {before} -> {after} "
+                    f"not plausible for a {w}-way sum")
+
+    # ------------------------------------------------------------------
+    def finalize(self) -> None:
+        """Call after loss.backward(): wait for every in-flight collective
+        (and launch any straggler buckets defensively)."""
+        for b in self.buckets:
+            if b.arrived != len(b.params) and _world(self.pg) > 1:
+                # a param got no grad this step (unused path) — its slot is
+                # zero, all-reduce anyway so ranks agree
+                self._launch(b)
+            if b.work is not None:
+                b.work.wait()  # current stream waits on NCCL/gloo completion
+                b.work = None
+            b.arrived = 0
+
+    def zero_grad(self) -> None:
+        for b in self.buckets:
+            b.flat.zero_()
+            b.arrived = 0
+
+    @property
+    def grad_scale(self) -> float:
+        """Fold this into the optimizer's grad_scale (× any loss-scale)."""
+        return 1.0 / _world(self.pg)
+
+    def detach(self) -> None:
+        for h in self._hooks:
+            h.remove()
+        self._hooks.clear()

@@ -1,0 +1,124 @@
+#!/usr/bin/env python3
+"""sodkit benchmark — the driver contract (BASELINE.json metric):
+
+    imgs/sec (whole node), MINet-R50 352x352 DDP, synthetic data, random init.
+
+    python bench.py --gpus N --steps K --warmup W
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One full optimization step per iteration (fwd + hybrid loss + bwd +
+bucketed RCCL all-reduce + fused SGD).  W untimed warmup steps, then K timed
+steps bracketed by barrier + torch.cuda.synchronize() on both sides; the
+step time is the MAX over ranks; rank 0 prints ONE JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--model", type=str, default="minet_res50")
+    ap.add_argument("--size", type=int, default=352)
+    ap.add_argument("--batch", type=int, default=32, help="per-GPU batch")
+    ap.add_argument("--dtype", type=str, default="bf16")
+    args = ap.parse_args()
+
+    from sodkit.config import make_config
+    from sodkit.engine import Solver
+    from sodkit.parallel import get_rank, get_world_size
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    cfg = make_config(
+        "minet_r50_352",
+        model=args.model, size=args.size, batch_size=args.batch,
+        dtype=args.dtype, num_workers=0, output_root="/tmp/sodkit_bench",
+        exp_name=f"bench_{args.model}", log_interval=10**9,
+    )
+    solver = Solver(cfg)
+    solver.total_iters = 10**9  # near-constant LR over the bench window
+    rank, world = get_rank(), get_world_size()
+    dev = solver.device
+    is_gpu = dev.type == "cuda"
+
+    # pre-generated on-device batches (cycle 4; rank-dependent data)
+    from sodkit.data import SyntheticSOD
+    with_depth = args.model.startswith("hdfnet")
+    ds = SyntheticSOD("train", args.size, 4 * args.batch,
+                      seed=1234 + rank, with_depth=with_depth)
+    batches = []
+    for b in range(4):
+        parts = [ds[b * args.batch + i] for i in range(args.batch)]
+        img = torch.stack([p[0] for p in parts])
+        mask = torch.stack([p[-2] for p in parts])
+        depth = torch.stack([p[1] for p in parts]) if with_depth else None
+        img = solver._to_device(img, cast=True)
+        mask = solver._to_device(mask, cast=False)
+        if depth is not None:
+            depth = solver._to_device(depth, cast=True)
+        batches.append((img, mask, depth))
+
+    def barrier_sync():
+        if dist.is_initialized() and world > 1:
+            dist.barrier()
+        if is_gpu:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        solver.step(*batches[i % 4][:2], batches[i % 4][2])
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        solver.step(*batches[i % 4][:2], batches[i % 4][2])
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX step time over ranks
+    if dist.is_initialized() and world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=dev if is_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    imgs_per_sec = args.batch * world * args.steps / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "imgs_per_sec",
+            "value": imgs_per_sec,
+            "unit": "imgs/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world,
+                "seq_len": args.size,
+                "parallelism": f"dp{world}",
+            },
+        }), flush=True)
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -1,0 +1,157 @@
+// sodkit — bilinear resize (+fused add) fwd and atomic-free gather bwd for
+// CDNA4 (the decoder top-down path and `upsample_add`).
+// NHWC; index math matches torch's align_corners=False upsample exactly
+// (area_pixel source index with max(0, .) clamp).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace sodkit {
+
+SOD_DEVINL void src_index(int o, float scale, int in_size, int &i0, int &i1,
+                          float &l) {
+    float src = fmaxf(0.f, ((float)o + 0.5f) * scale - 0.5f);
+    i0 = (int)src;
+    if (i0 > in_size - 1) i0 = in_size - 1;
+    i1 = i0 + (i0 < in_size - 1 ? 1 : 0);
+    l = src - (float)i0;
+    if (l < 0.f) l = 0.f;
+    if (l > 1.f) l = 1.f;
+}
+
+template <typename T, bool ADD>
+__global__ void bilinear_fwd_kernel(const T *__restrict__ x,
+                                    const T *__restrict__ add,
+                                    T *__restrict__ y, int N, int C, int Hi,
+                                    int Wi, int Ho, int Wo, float sh, float sw) {
+    long total = (long)N * Ho * Wo * C;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < total; i += stride) {
+        int c = (int)(i % C);
+        long pix = i / C;
+        int ow = (int)(pix % Wo);
+        long t2 = pix / Wo;
+        int oh = (int)(t2 % Ho);
+        int n = (int)(t2 / Ho);
+        int h0, h1, w0, w1;
+        float lh, lw;
+        src_index(oh, sh, Hi, h0, h1, lh);
+        src_index(ow, sw, Wi, w0, w1, lw);
+        const T *base = x + ((long)n * Hi) * Wi * C + c;
+        float v00 = ElemTraits<T>::to_f32(base[((long)h0 * Wi + w0) * C]);
+        float v01 = ElemTraits<T>::to_f32(base[((long)h0 * Wi + w1) * C]);
+        float v10 = ElemTraits<T>::to_f32(base[((long)h1 * Wi + w0) * C]);
+        float v11 = ElemTraits<T>::to_f32(base[((long)h1 * Wi + w1) * C]);
+        float v = (1 - lh) * ((1 - lw) * v00 + lw * v01)
+                + lh * ((1 - lw) * v10 + lw * v11);
+        if (ADD) v += ElemTraits<T>::to_f32(add[i]);
+        y[i] = ElemTraits<T>::from_f32(v);
+    }
+}
+
+// Deterministic gather backward: each input pixel sums the output pixels
+// whose interpolation stencil touches it (bounded range; ~2x upsampling in
+// the decoder means <= 4 taps per dim).
+template <typename T>
+__global__ void bilinear_bwd_kernel(const T *__restrict__ dy,
+                                    T *__restrict__ dx, int N, int C, int Hi,
+                                    int Wi, int Ho, int Wo, float sh, float sw) {
+    long total = (long)N * Hi * Wi * C;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < total; i += stride) {
+        int c = (int)(i % C);
+        long pix = i / C;
+        int iw = (int)(pix % Wi);
+        long t2 = pix / Wi;
+        int ih = (int)(t2 % Hi);
+        int n = (int)(t2 / Hi);
+        // conservative output ranges whose source index can hit ih/iw
+        int oh_lo = max(0, (int)floorf(((float)ih - 1.f + 0.5f) / sh - 0.5f) - 1);
+        int oh_hi = min(Ho - 1, (int)ceilf(((float)ih + 1.f + 0.5f) / sh - 0.5f) + 1);
+        int ow_lo = max(0, (int)floorf(((float)iw - 1.f + 0.5f) / sw - 0.5f) - 1);
+        int ow_hi = min(Wo - 1, (int)ceilf(((float)iw + 1.f + 0.5f) / sw - 0.5f) + 1);
+        float acc = 0.f;
+        const T *dybase = dy + (long)n * Ho * Wo * C + c;
+        for (int oh = oh_lo; oh <= oh_hi; ++oh) {
+            int h0, h1;
+            float lh;
+            src_index(oh, sh, Hi, h0, h1, lh);
+            float wh = 0.f;
+            if (h0 == ih) wh += 1.f - lh;
+            if (h1 == ih) wh += lh;
+            if (wh == 0.f) continue;
+            for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+                int w0, w1;
+                float lw;
+                src_index(ow, sw, Wi, w0, w1, lw);
+                float ww = 0.f;
+                if (w0 == iw) ww += 1.f - lw;
+                if (w1 == iw) ww += lw;
+                if (ww == 0.f) continue;
+                acc += wh * ww
+                     * ElemTraits<T>::to_f32(dybase[((long)oh * Wo + ow) * C]);
+            }
+        }
+        dx[i] = ElemTraits<T>::from_f32(acc);
+    }
+}
+
+}  // namespace sodkit
+
+#define DISPATCH_T2(TENSOR, NAME, ...)                                         \
+    AT_DISPATCH_SWITCH((TENSOR).scalar_type(), NAME,                           \
+        AT_DISPATCH_CASE(at::kBFloat16, [&] { using T = __hip_bfloat16; __VA_ARGS__(); }) \
+        AT_DISPATCH_CASE(at::kHalf,     [&] { using T = __half; __VA_ARGS__(); })          \
+        AT_DISPATCH_CASE(at::kFloat,    [&] { using T = float; __VA_ARGS__(); }))
+
+torch::Tensor bilinear_fwd(torch::Tensor x, int64_t Ho, int64_t Wo,
+                           torch::Tensor add) {
+    TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+    int N = x.size(0), C = x.size(1), Hi = x.size(2), Wi = x.size(3);
+    bool has_add = add.defined() && add.numel() > 0;
+    auto y = torch::empty({N, C, Ho, Wo},
+                          x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    float sh = (float)Hi / (float)Ho, sw = (float)Wi / (float)Wo;
+    long total = y.numel();
+    int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+    auto stream = at::hip::getCurrentHIPStream();
+    DISPATCH_T2(x, "bilinear_fwd", [&] {
+        if (has_add)
+            hipLaunchKernelGGL((sodkit::bilinear_fwd_kernel<T, true>), dim3(blocks),
+                               dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const T *>(x.data_ptr()),
+                               reinterpret_cast<const T *>(add.data_ptr()),
+                               reinterpret_cast<T *>(y.data_ptr()),
+                               N, C, Hi, Wi, (int)Ho, (int)Wo, sh, sw);
+        else
+            hipLaunchKernelGGL((sodkit::bilinear_fwd_kernel<T, false>), dim3(blocks),
+                               dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const T *>(x.data_ptr()), nullptr,
+                               reinterpret_cast<T *>(y.data_ptr()),
+                               N, C, Hi, Wi, (int)Ho, (int)Wo, sh, sw);
+    });
+    return y;
+}
+
+torch::Tensor bilinear_bwd(torch::Tensor dy, int64_t Hi, int64_t Wi) {
+    TORCH_CHECK(dy.is_cuda() && dy.dim() == 4);
+    int N = dy.size(0), C = dy.size(1), Ho = dy.size(2), Wo = dy.size(3);
+    auto dx = torch::empty({N, C, Hi, Wi},
+                           dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    float sh = (float)Hi / (float)Ho, sw = (float)Wi / (float)Wo;
+    long total = dx.numel();
+    int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+    auto stream = at::hip::getCurrentHIPStream();
+    DISPATCH_T2(dy, "bilinear_bwd", [&] {
+        hipLaunchKernelGGL((sodkit::bilinear_bwd_kernel<T>), dim3(blocks),
+                           dim3(256), 0, stream.stream(),
+                           reinterpret_cast<const T *>(dy.data_ptr()),
+                           reinterpret_cast<T *>(dx.data_ptr()),
+                           N, C, (int)Hi, (int)Wi, Ho, Wo, sh, sw);
+    });
+    return dx;
+}

@@ -1,0 +1,53 @@
+// sodkit._C — python bindings for the CDNA4 kernel suite.
+#include <torch/extension.h>
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                         int64_t stride, int64_t pad, int64_t dil);
+torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w_t, int64_t H,
+                           int64_t W, int64_t stride, int64_t pad, int64_t dil);
+torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
+                           int64_t S, int64_t stride, int64_t pad, int64_t dil);
+torch::Tensor channel_sum(torch::Tensor x);
+
+torch::Tensor bn_stats(torch::Tensor x);
+torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd,
+                       torch::Tensor gamma, torch::Tensor beta, bool relu,
+                       torch::Tensor residual);
+torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                            torch::Tensor mean, torch::Tensor invstd, bool relu);
+std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor dy, torch::Tensor y,
+                                        torch::Tensor x, torch::Tensor mean,
+                                        torch::Tensor invstd, torch::Tensor gamma,
+                                        torch::Tensor sums, double n_total,
+                                        bool relu, bool need_dres);
+
+torch::Tensor bilinear_fwd(torch::Tensor x, int64_t Ho, int64_t Wo,
+                           torch::Tensor add);
+torch::Tensor bilinear_bwd(torch::Tensor dy, int64_t Hi, int64_t Wi);
+
+std::vector<torch::Tensor> loss_fwd(torch::Tensor z, torch::Tensor t,
+                                    bool need_ssim);
+torch::Tensor loss_bwd(torch::Tensor p, torch::Tensor t, torch::Tensor img_sums,
+                       torch::Tensor stats, double wb, double wi, double ws,
+                       torch::Tensor go);
+
+void sgd_fused(torch::Tensor ptrs, torch::Tensor numels, torch::Tensor wds,
+               torch::Tensor chunks, int64_t dtype_code, double lr, double mu,
+               double gscale, bool nesterov);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.doc() = "sodkit CDNA4 (gfx950/MI355X) kernels";
+    m.def("conv2d_fwd", &conv2d_fwd);
+    m.def("conv2d_dgrad", &conv2d_dgrad);
+    m.def("conv2d_wgrad", &conv2d_wgrad);
+    m.def("channel_sum", &channel_sum);
+    m.def("bn_stats", &bn_stats);
+    m.def("bn_apply", &bn_apply);
+    m.def("bn_bwd_reduce", &bn_bwd_reduce);
+    m.def("bn_bwd_apply", &bn_bwd_apply);
+    m.def("bilinear_fwd", &bilinear_fwd);
+    m.def("bilinear_bwd", &bilinear_bwd);
+    m.def("loss_fwd", &loss_fwd);
+    m.def("loss_bwd", &loss_bwd);
+    m.def("sgd_fused", &sgd_fused);
+}

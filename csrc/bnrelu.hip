@@ -1,0 +1,231 @@
+// sodkit — fused BatchNorm(+residual)(+ReLU) kernels for CDNA4 (SURVEY §2.3
+// K3 + K8 epilogue).  NHWC layout; fp32 stats/params, bf16/fp16/fp32 data.
+// Split into stats-reduce / apply (and bwd-reduce / bwd-apply) so SyncBN is
+// one coalesced all-reduce between the halves with an identical kernel path.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace sodkit {
+
+// ---- per-channel sum + sumsq -> out[2][C] (fp32, atomics across blocks) ----
+template <typename T>
+__global__ void bn_stats_kernel(const T *__restrict__ x, float *__restrict__ out,
+                                long M, int C) {
+    int c = blockIdx.y * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float s = 0.f, ss = 0.f;
+    for (long m = blockIdx.x; m < M; m += gridDim.x) {
+        float v = ElemTraits<T>::to_f32(x[m * C + c]);
+        s += v;
+        ss += v * v;
+    }
+    atomicAdd(&out[c], s);
+    atomicAdd(&out[C + c], ss);
+}
+
+// ---- y = relu((x - mean) * invstd * gamma + beta + residual) ----
+template <typename T, bool RELU, bool RES>
+__global__ void bn_apply_kernel(const T *__restrict__ x,
+                                const T *__restrict__ res,
+                                const float *__restrict__ mean,
+                                const float *__restrict__ invstd,
+                                const float *__restrict__ gamma,
+                                const float *__restrict__ beta,
+                                T *__restrict__ y, long total, int C) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < total; i += stride) {
+        int c = (int)(i % C);
+        float sc = invstd[c] * gamma[c];
+        float sh = beta[c] - mean[c] * sc;
+        float v = ElemTraits<T>::to_f32(x[i]) * sc + sh;
+        if (RES) v += ElemTraits<T>::to_f32(res[i]);
+        if (RELU) v = fmaxf(v, 0.f);
+        y[i] = ElemTraits<T>::from_f32(v);
+    }
+}
+
+// ---- backward reduce: g = dy * relumask(y); out[0][c]=sum g,
+//      out[1][c]=sum g*xhat ----
+template <typename T, bool RELU>
+__global__ void bn_bwd_reduce_kernel(const T *__restrict__ dy,
+                                     const T *__restrict__ y,
+                                     const T *__restrict__ x,
+                                     const float *__restrict__ mean,
+                                     const float *__restrict__ invstd,
+                                     float *__restrict__ out, long M, int C) {
+    int c = blockIdx.y * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float mu = mean[c], is = invstd[c];
+    float sg = 0.f, sgx = 0.f;
+    for (long m = blockIdx.x; m < M; m += gridDim.x) {
+        long i = m * C + c;
+        float g = ElemTraits<T>::to_f32(dy[i]);
+        if (RELU && ElemTraits<T>::to_f32(y[i]) <= 0.f) g = 0.f;
+        float xh = (ElemTraits<T>::to_f32(x[i]) - mu) * is;
+        sg += g;
+        sgx += g * xh;
+    }
+    atomicAdd(&out[c], sg);
+    atomicAdd(&out[C + c], sgx);
+}
+
+// ---- backward apply:
+//   train: dx = gamma*invstd * (g - (sum_g + xhat*sum_gxhat)/n)
+//   eval (inv_n == 0): dx = gamma*invstd * g
+//   optional residual grad dres = g ----
+template <typename T, bool RELU, bool NEED_DRES>
+__global__ void bn_bwd_apply_kernel(const T *__restrict__ dy,
+                                    const T *__restrict__ y,
+                                    const T *__restrict__ x,
+                                    const float *__restrict__ mean,
+                                    const float *__restrict__ invstd,
+                                    const float *__restrict__ gamma,
+                                    const float *__restrict__ sums,
+                                    float inv_n, T *__restrict__ dx,
+                                    T *__restrict__ dres, long total, int C) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < total; i += stride) {
+        int c = (int)(i % C);
+        float g = ElemTraits<T>::to_f32(dy[i]);
+        if (RELU && ElemTraits<T>::to_f32(y[i]) <= 0.f) g = 0.f;
+        if (NEED_DRES) dres[i] = ElemTraits<T>::from_f32(g);
+        float is = invstd[c];
+        float xh = (ElemTraits<T>::to_f32(x[i]) - mean[c]) * is;
+        float v = g - (sums[c] + xh * sums[C + c]) * inv_n;
+        dx[i] = ElemTraits<T>::from_f32(gamma[c] * is * v);
+    }
+}
+
+}  // namespace sodkit
+
+// ===========================================================================
+// Host glue
+// ===========================================================================
+
+namespace {
+
+inline dim3 stats_grid(long M, int C) {
+    return dim3(std::min<long>(M, 512), (C + 255) / 256);
+}
+
+inline int ew_blocks(long total) {
+    return (int)std::min<long>((total + 255) / 256, 4096);
+}
+
+#define DISPATCH_T(TENSOR, NAME, ...)                                          \
+    AT_DISPATCH_SWITCH((TENSOR).scalar_type(), NAME,                           \
+        AT_DISPATCH_CASE(at::kBFloat16, [&] { using T = __hip_bfloat16; __VA_ARGS__(); }) \
+        AT_DISPATCH_CASE(at::kHalf,     [&] { using T = __half; __VA_ARGS__(); })          \
+        AT_DISPATCH_CASE(at::kFloat,    [&] { using T = float; __VA_ARGS__(); }))
+
+}  // namespace
+
+torch::Tensor bn_stats(torch::Tensor x) {
+    TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+    int C = x.size(1);
+    long M = x.numel() / C;
+    auto out = torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    DISPATCH_T(x, "bn_stats", [&] {
+        hipLaunchKernelGGL((sodkit::bn_stats_kernel<T>), stats_grid(M, C),
+                           dim3(256), 0, stream.stream(),
+                           reinterpret_cast<const T *>(x.data_ptr()),
+                           out.data_ptr<float>(), M, C);
+    });
+    return out;
+}
+
+torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd,
+                       torch::Tensor gamma, torch::Tensor beta, bool relu,
+                       torch::Tensor residual) {
+    TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+    int C = x.size(1);
+    long total = x.numel();
+    auto y = torch::empty_like(x);
+    auto stream = at::hip::getCurrentHIPStream();
+    bool has_res = residual.defined() && residual.numel() > 0;
+    auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
+    auto gammaf = gamma.to(torch::kFloat32).contiguous();
+    auto betaf = beta.to(torch::kFloat32).contiguous();
+    DISPATCH_T(x, "bn_apply", [&] {
+        const T *resp = has_res ? reinterpret_cast<const T *>(residual.data_ptr())
+                                : nullptr;
+        auto launch = [&](auto rl, auto rs) {
+            hipLaunchKernelGGL(
+                (sodkit::bn_apply_kernel<T, decltype(rl)::value, decltype(rs)::value>),
+                dim3(ew_blocks(total)), dim3(256), 0, stream.stream(),
+                reinterpret_cast<const T *>(x.data_ptr()), resp,
+                meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
+                reinterpret_cast<T *>(y.data_ptr()), total, C);
+        };
+        using tt = std::true_type; using ff = std::false_type;
+        if (relu) { if (has_res) launch(tt{}, tt{}); else launch(tt{}, ff{}); }
+        else      { if (has_res) launch(ff{}, tt{}); else launch(ff{}, ff{}); }
+    });
+    return y;
+}
+
+torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                            torch::Tensor mean, torch::Tensor invstd, bool relu) {
+    int C = x.size(1);
+    long M = x.numel() / C;
+    auto out = torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
+    DISPATCH_T(x, "bn_bwd_reduce", [&] {
+        auto launch = [&](auto rl) {
+            hipLaunchKernelGGL((sodkit::bn_bwd_reduce_kernel<T, decltype(rl)::value>),
+                               stats_grid(M, C), dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const T *>(dy.data_ptr()),
+                               reinterpret_cast<const T *>(y.data_ptr()),
+                               reinterpret_cast<const T *>(x.data_ptr()),
+                               meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                               out.data_ptr<float>(), M, C);
+        };
+        if (relu) launch(std::true_type{}); else launch(std::false_type{});
+    });
+    return out;
+}
+
+std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor dy, torch::Tensor y,
+                                        torch::Tensor x, torch::Tensor mean,
+                                        torch::Tensor invstd, torch::Tensor gamma,
+                                        torch::Tensor sums, double n_total,
+                                        bool relu, bool need_dres) {
+    int C = x.size(1);
+    long total = x.numel();
+    auto dx = torch::empty_like(x);
+    torch::Tensor dres;
+    if (need_dres) dres = torch::empty_like(x);
+    float inv_n = n_total > 0 ? (float)(1.0 / n_total) : 0.f;
+    auto stream = at::hip::getCurrentHIPStream();
+    auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
+    auto gammaf = gamma.to(torch::kFloat32).contiguous();
+    auto sumsf = sums.contiguous();
+    DISPATCH_T(x, "bn_bwd_apply", [&] {
+        auto launch = [&](auto rl, auto dr) {
+            hipLaunchKernelGGL(
+                (sodkit::bn_bwd_apply_kernel<T, decltype(rl)::value, decltype(dr)::value>),
+                dim3(ew_blocks(total)), dim3(256), 0, stream.stream(),
+                reinterpret_cast<const T *>(dy.data_ptr()),
+                reinterpret_cast<const T *>(y.data_ptr()),
+                reinterpret_cast<const T *>(x.data_ptr()),
+                meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                gammaf.data_ptr<float>(), sumsf.data_ptr<float>(), inv_n,
+                reinterpret_cast<T *>(dx.data_ptr()),
+                need_dres ? reinterpret_cast<T *>(dres.data_ptr()) : nullptr,
+                total, C);
+        };
+        using tt = std::true_type; using ff = std::false_type;
+        if (relu) { if (need_dres) launch(tt{}, tt{}); else launch(tt{}, ff{}); }
+        else      { if (need_dres) launch(ff{}, tt{}); else launch(ff{}, ff{}); }
+    });
+    if (!need_dres) dres = torch::Tensor();
+    return {dx, dres};
+}

@@ -1,0 +1,609 @@
+// sodkit — MFMA implicit-GEMM convolution family for CDNA4 (gfx950).
+//
+// Covers SURVEY §2.3 K1 (3x3), K2 (1x1), K6 (7x7 stem), K10 (dilated),
+// K13 (Res2Net splits are just more K1 calls): forward, dgrad and split-K
+// wgrad, all dtypes bf16/fp16, fp32 accumulation.
+//
+// Layouts (chosen for MI355X, not inherited):
+//   activations  NHWC            (torch channels_last physical layout)
+//   weights      [Cout][R][S][Cin]  == GEMM A-operand [Cout][K] row-major
+//   dgrad weight [Cin][R][S][Cout]  (small permuted copy made by the caller)
+//
+// GEMM mapping (fwd):  D[cout][m] = sum_k W[cout][k] * im2col(x)[k][m]
+//   A-operand (W) stages coalesced from global, no transpose.
+//   B-operand (X) is the im2col gather: for NHWC, 8 consecutive k are 8
+//   consecutive cin of one (r,s) tap -> 16-byte vector gathers when Cin%8==0.
+//   D-fragment: each lane holds 4 *consecutive couts* of one output pixel ->
+//   packs to one 8-byte NHWC store.  (Operand roles were chosen exactly so
+//   the epilogue store is the coalesced one.)
+//
+// Structure per K-step (BK=64): reg-staged global->LDS with XOR-swizzled
+// tiles (bank-conflict-free ds_read_b128 fragment reads, CDNA4 guide T2),
+// 2x2 waves x (FCx FM) 16x16x32 MFMA fragments, two barriers per step.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace sodkit {
+
+// swizzled byte offset inside a [rows][BK] 16-bit tile, row pitch BK*2 bytes
+template <int BK>
+SOD_DEVINL int swz(int row, int kbyte) {
+    return row * (BK * 2) + (kbyte ^ ((row & 7) << 4));
+}
+
+// ---------------------------------------------------------------------------
+// Gather address computation.  MODE 0 = fwd (gather x), MODE 1 = dgrad
+// (gather dy through the transposed-conv index map).
+// Returns element offset into the source, or -1 if out of range / padding.
+// ---------------------------------------------------------------------------
+template <int MODE>
+SOD_DEVINL long gather_off(int n, int p, int q, int r, int s, int c,
+                           int Hs, int Ws, int Cs,
+                           int stride, int pad, int dil) {
+    if (MODE == 0) {
+        int ih = p * stride + r * dil - pad;
+        int iw = q * stride + s * dil - pad;
+        if ((unsigned)ih >= (unsigned)Hs || (unsigned)iw >= (unsigned)Ws)
+            return -1;
+        return (((long)n * Hs + ih) * Ws + iw) * Cs + c;
+    } else {
+        int hh = p + pad - r * dil;
+        int ww = q + pad - s * dil;
+        if (hh < 0 || ww < 0 || hh % stride || ww % stride) return -1;
+        int oh = hh / stride, ow = ww / stride;
+        if (oh >= Hs || ow >= Ws) return -1;
+        return (((long)n * Hs + oh) * Ws + ow) * Cs + c;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// fwd / dgrad kernel.
+//   A_w  : [Ncols? no — Ncout][K_total] row-major ("weights")
+//   B_src: NHWC source for the gather (x for fwd, dy for dgrad)
+//   out  : [M_total][Ncout] (NHWC output flattened over pixels)
+// ---------------------------------------------------------------------------
+template <typename T, int MODE, int BCOUT, int BM>
+__global__ __launch_bounds__(256)
+void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
+                  const float *__restrict__ bias, T *__restrict__ out,
+                  int Ns, int Hs, int Ws, int Cs,   // gather-source dims
+                  int P, int Q,                     // output spatial
+                  int Ncout,                        // output channels
+                  int R, int S, int stride, int pad, int dil,
+                  int M_total, int K_total, int tiles_m) {
+    constexpr int BK = 64;
+    constexpr int FC = BCOUT / 32;   // a-fragments per wave (2x2 wave grid)
+    constexpr int FM = BM / 32;      // b-fragments per wave
+
+    __shared__ __align__(16) char lds[(BCOUT + BM) * BK * 2];
+    char *w_lds = lds;                         // [BCOUT][BK]
+    char *x_lds = lds + BCOUT * BK * 2;        // [BM][BK]
+
+    const int bid = blockIdx.x;
+    const int tile_m = (bid % tiles_m) * BM;
+    const int tile_c = (bid / tiles_m) * BCOUT;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wv = t >> 6;                  // 4 waves: 2x2
+    const int wv_c = wv >> 1, wv_m = wv & 1;
+
+    const bool vec_ok = (Cs % 8 == 0);
+    const bool wvec_ok = (K_total % 8 == 0);
+
+    // --- per-thread staging geometry (fixed across K-steps) ---
+    // W chunks: BCOUT*BK/8 8-elem chunks, 256 threads
+    constexpr int WCH = BCOUT * BK / 8 / 256;       // chunks per thread
+    constexpr int XCH = BM * BK / 8 / 256;
+    // chunk ch covers (row = ch / (BK/8), kv = ch % (BK/8))
+    int xm[XCH], xn[XCH], xp[XCH], xq[XCH];
+#pragma unroll
+    for (int i = 0; i < XCH; ++i) {
+        int ch = t + i * 256;
+        int m = tile_m + ch / (BK / 8);
+        xm[i] = m;
+        int mm = m < M_total ? m : 0;
+        xn[i] = mm / (P * Q);
+        int pq = mm % (P * Q);
+        xp[i] = pq / Q;
+        xq[i] = pq % Q;
+    }
+
+    f32x4 acc[FC][FM];
+#pragma unroll
+    for (int a = 0; a < FC; ++a)
+#pragma unroll
+        for (int b = 0; b < FM; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int k0 = 0; k0 < K_total; k0 += BK) {
+        // ---- stage W tile: [BCOUT][BK], coalesced rows from A_w ----
+#pragma unroll
+        for (int i = 0; i < WCH; ++i) {
+            int ch = t + i * 256;
+            int row = ch / (BK / 8);
+            int k = k0 + (ch % (BK / 8)) * 8;
+            int cout = tile_c + row;
+            s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (cout < Ncout) {
+                const T *src = A_w + (long)cout * K_total + k;
+                if (wvec_ok && k + 7 < K_total) {
+                    v = *reinterpret_cast<const s16x8 *>(src);
+                } else {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e)
+                        if (k + e < K_total)
+                            v[e] = reinterpret_cast<const short *>(src)[e];
+                }
+            }
+            *reinterpret_cast<s16x8 *>(w_lds + swz<BK>(row, (ch % (BK / 8)) * 16)) = v;
+        }
+        // ---- stage X tile: [BM][BK], im2col gather ----
+#pragma unroll
+        for (int i = 0; i < XCH; ++i) {
+            int ch = t + i * 256;
+            int row = ch / (BK / 8);
+            int k = k0 + (ch % (BK / 8)) * 8;
+            s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (xm[i] < M_total && k < K_total) {
+                if (vec_ok && (k % 8 == 0) && (k / Cs == (k + 7) / Cs)) {
+                    int rs = k / Cs, c = k % Cs;
+                    int r = rs / S, s = rs % S;
+                    long off = gather_off<MODE>(xn[i], xp[i], xq[i], r, s, c,
+                                                Hs, Ws, Cs, stride, pad, dil);
+                    if (off >= 0)
+                        v = *reinterpret_cast<const s16x8 *>(B_src + off);
+                } else {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        int ke = k + e;
+                        if (ke < K_total) {
+                            int rs = ke / Cs, c = ke % Cs;
+                            int r = rs / S, s = rs % S;
+                            long off = gather_off<MODE>(xn[i], xp[i], xq[i], r, s, c,
+                                                        Hs, Ws, Cs, stride, pad, dil);
+                            if (off >= 0)
+                                v[e] = reinterpret_cast<const short *>(B_src)[off];
+                        }
+                    }
+                }
+            }
+            *reinterpret_cast<s16x8 *>(x_lds + swz<BK>(row, (ch % (BK / 8)) * 16)) = v;
+        }
+        __syncthreads();
+
+        // ---- MFMA over the K-step ----
+#pragma unroll
+        for (int kk = 0; kk < BK / 32; ++kk) {
+            int kbyte = kk * 64 + (lane >> 4) * 16;
+            s16x8 a[FC], b[FM];
+#pragma unroll
+            for (int fc = 0; fc < FC; ++fc) {
+                int row = wv_c * (BCOUT / 2) + fc * 16 + (lane & 15);
+                a[fc] = *reinterpret_cast<const s16x8 *>(w_lds + swz<BK>(row, kbyte));
+            }
+#pragma unroll
+            for (int fm = 0; fm < FM; ++fm) {
+                int row = wv_m * (BM / 2) + fm * 16 + (lane & 15);
+                b[fm] = *reinterpret_cast<const s16x8 *>(x_lds + swz<BK>(row, kbyte));
+            }
+#pragma unroll
+            for (int fc = 0; fc < FC; ++fc)
+#pragma unroll
+                for (int fm = 0; fm < FM; ++fm)
+                    acc[fc][fm] = MfmaOp<T>::mma(a[fc], b[fm], acc[fc][fm]);
+        }
+        __syncthreads();
+    }
+
+    // ---- epilogue: D[cout][m] -> NHWC out[m][cout], 8B packed stores ----
+#pragma unroll
+    for (int fc = 0; fc < FC; ++fc) {
+        int cout = tile_c + wv_c * (BCOUT / 2) + fc * 16 + (lane >> 4) * 4;
+        float b4[4] = {0.f, 0.f, 0.f, 0.f};
+        if (bias != nullptr) {
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+                if (cout + e < Ncout) b4[e] = bias[cout + e];
+        }
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm) {
+            int m = tile_m + wv_m * (BM / 2) + fm * 16 + (lane & 15);
+            if (m >= M_total) continue;
+            f32x4 v = acc[fc][fm];
+            if (bias != nullptr) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e) v[e] += b4[e];
+            }
+            T *dst = out + (long)m * Ncout + cout;
+            // vector store needs 8B alignment: holds iff Ncout % 4 == 0
+            if ((Ncout % 4 == 0) && cout + 3 < Ncout) {
+                *reinterpret_cast<s16x4 *>(dst) = pack4<T>(v);
+            } else {
+#pragma unroll
+                for (int e = 0; e < 4; ++e)
+                    if (cout + e < Ncout)
+                        dst[e] = ElemTraits<T>::from_f32(v[e]);
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad kernel: dw[cout][(r,s,cin)] = sum_m dy[m][cout] * im2col(x)[m][(r,s,cin)]
+// Split-K over m; fp32 partials [splitk][Cout][Kw].
+//   dy^T tile is built by an in-LDS transpose (vector global read, scalar
+//   ds writes); the x tile gathers 8 consecutive output-pixels per lane
+//   (m-contiguous in LDS -> single b128 write).
+// ---------------------------------------------------------------------------
+template <typename T, int BCOUT, int BJ>
+__global__ __launch_bounds__(256)
+void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
+                  float *__restrict__ partial,
+                  int N, int H, int W, int C,     // x dims
+                  int P, int Q, int Cout,         // dy dims
+                  int R, int S, int stride, int pad, int dil,
+                  int Kw,                          // = R*S*C (dw inner size)
+                  int M_total,                     // = N*P*Q
+                  int m_per_split, int tiles_c) {
+    constexpr int BKM = 64;   // m's per K-step
+    constexpr int FC = BCOUT / 32;
+    constexpr int FJ = BJ / 32;
+
+    __shared__ __align__(16) char lds[(BCOUT + BJ) * BKM * 2];
+    char *dyT_lds = lds;                      // [BCOUT][BKM]
+    char *xT_lds = lds + BCOUT * BKM * 2;     // [BJ][BKM]
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wv = t >> 6;
+    const int wv_c = wv >> 1, wv_j = wv & 1;
+
+    const int tile_c = (blockIdx.x % tiles_c) * BCOUT;
+    const int tile_j = (blockIdx.x / tiles_c) * BJ;
+    const int z = blockIdx.y;
+    const int m_begin = z * m_per_split;
+    const int m_end = min(m_begin + m_per_split, M_total);
+
+    const bool dyvec_ok = (Cout % 8 == 0);
+
+    // per-thread x-chunk geometry: chunk ch -> (j = ch % BJ, mv8 = ch / BJ)
+    constexpr int XCH = BJ * BKM / 8 / 256;
+    int xr[XCH], xs[XCH], xc[XCH], xj[XCH];
+#pragma unroll
+    for (int i = 0; i < XCH; ++i) {
+        int ch = t + i * 256;
+        int j = tile_j + (ch % BJ);
+        xj[i] = j;
+        int jj = j < Kw ? j : 0;
+        int rs = jj / C;
+        xc[i] = jj % C;
+        xr[i] = rs / S;
+        xs[i] = rs % S;
+    }
+    constexpr int DCH = BCOUT * BKM / 8 / 256;
+
+    f32x4 acc[FC][FJ];
+#pragma unroll
+    for (int a = 0; a < FC; ++a)
+#pragma unroll
+        for (int b = 0; b < FJ; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
+        // ---- stage dy^T tile: read dy[m][cout..+7], write transposed ----
+#pragma unroll
+        for (int i = 0; i < DCH; ++i) {
+            int ch = t + i * 256;
+            int m_local = ch / (BCOUT / 8);
+            int cv = (ch % (BCOUT / 8)) * 8;
+            int m = m0 + m_local;
+            int cout = tile_c + cv;
+            s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (m < m_end && cout < Cout) {
+                const T *src = dy + (long)m * Cout + cout;
+                if (dyvec_ok && cout + 7 < Cout) {
+                    v = *reinterpret_cast<const s16x8 *>(src);
+                } else {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e)
+                        if (cout + e < Cout)
+                            v[e] = reinterpret_cast<const short *>(src)[e];
+                }
+            }
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                int row = cv + e;
+                *reinterpret_cast<short *>(dyT_lds + swz<BKM>(row, m_local * 2)) = v[e];
+            }
+        }
+        // ---- stage x^T tile: 8 m-contiguous gathers per chunk ----
+#pragma unroll
+        for (int i = 0; i < XCH; ++i) {
+            int ch = t + i * 256;
+            int j_local = ch % BJ;
+            int mv8 = ch / BJ;
+            s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (xj[i] < Kw) {
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    int m = m0 + mv8 * 8 + e;
+                    if (m < m_end) {
+                        int n = m / (P * Q);
+                        int pq = m % (P * Q);
+                        int p = pq / Q, q = pq % Q;
+                        long off = gather_off<0>(n, p, q, xr[i], xs[i], xc[i],
+                                                 H, W, C, stride, pad, dil);
+                        if (off >= 0)
+                            v[e] = reinterpret_cast<const short *>(x)[off];
+                    }
+                }
+            }
+            *reinterpret_cast<s16x8 *>(xT_lds + swz<BKM>(j_local, mv8 * 16)) = v;
+        }
+        __syncthreads();
+
+#pragma unroll
+        for (int kk = 0; kk < BKM / 32; ++kk) {
+            int kbyte = kk * 64 + (lane >> 4) * 16;
+            s16x8 a[FC], b[FJ];
+#pragma unroll
+            for (int fc = 0; fc < FC; ++fc) {
+                int row = wv_c * (BCOUT / 2) + fc * 16 + (lane & 15);
+                a[fc] = *reinterpret_cast<const s16x8 *>(dyT_lds + swz<BKM>(row, kbyte));
+            }
+#pragma unroll
+            for (int fj = 0; fj < FJ; ++fj) {
+                int row = wv_j * (BJ / 2) + fj * 16 + (lane & 15);
+                b[fj] = *reinterpret_cast<const s16x8 *>(xT_lds + swz<BKM>(row, kbyte));
+            }
+#pragma unroll
+            for (int fc = 0; fc < FC; ++fc)
+#pragma unroll
+                for (int fj = 0; fj < FJ; ++fj)
+                    acc[fc][fj] = MfmaOp<T>::mma(a[fc], b[fj], acc[fc][fj]);
+        }
+        __syncthreads();
+    }
+
+    // ---- store fp32 partial [z][Cout][Kw] ----
+    float *dst_base = partial + (long)z * Cout * Kw;
+#pragma unroll
+    for (int fc = 0; fc < FC; ++fc) {
+        int cout = tile_c + wv_c * (BCOUT / 2) + fc * 16 + (lane >> 4) * 4;
+#pragma unroll
+        for (int fj = 0; fj < FJ; ++fj) {
+            int j = tile_j + wv_j * (BJ / 2) + fj * 16 + (lane & 15);
+            if (j >= Kw) continue;
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+                if (cout + e < Cout)
+                    dst_base[(long)(cout + e) * Kw + j] = acc[fc][fj][e];
+        }
+    }
+}
+
+template <typename T>
+__global__ void splitk_reduce_kernel(const float *__restrict__ partial,
+                                     T *__restrict__ dw, long n_elems,
+                                     int splitk) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < n_elems; i += stride) {
+        float s = 0.f;
+        for (int z = 0; z < splitk; ++z) s += partial[(long)z * n_elems + i];
+        dw[i] = ElemTraits<T>::from_f32(s);
+    }
+}
+
+// channel sum of an NHWC tensor -> fp32 [C]  (conv bias grad; SURVEY K-misc)
+template <typename T>
+__global__ void channel_sum_kernel(const T *__restrict__ x, float *__restrict__ out,
+                                   long M, int C) {
+    // one block per channel-chunk; grid-stride over rows
+    int c = blockIdx.y * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float s = 0.f;
+    for (long m = blockIdx.x; m < M; m += gridDim.x)
+        s += ElemTraits<T>::to_f32(x[m * C + c]);
+    atomicAdd(&out[c], s);
+}
+
+}  // namespace sodkit
+
+// ===========================================================================
+// Host glue
+// ===========================================================================
+
+#define CHECK_GPU(x) TORCH_CHECK((x).is_cuda(), #x " must be on GPU")
+
+static inline int out_dim(int in, int k, int stride, int pad, int dil) {
+    return (in + 2 * pad - dil * (k - 1) - 1) / stride + 1;
+}
+
+namespace {
+
+template <typename scalar_t, int MODE>
+void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
+                  const torch::Tensor &bias, torch::Tensor &out,
+                  int Ns, int Hs, int Ws, int Cs, int P, int Q, int Ncout,
+                  int R, int S, int stride, int pad, int dil,
+                  long M_total, int K_total) {
+    using T = scalar_t;
+    auto stream = at::hip::getCurrentHIPStream();
+    const float *bias_ptr = bias.defined() && bias.numel()
+                                ? bias.data_ptr<float>() : nullptr;
+    auto launch = [&](auto bc_tag, auto bm_tag) {
+        constexpr int BC = decltype(bc_tag)::value;
+        constexpr int BMv = decltype(bm_tag)::value;
+        int tiles_m = (M_total + BMv - 1) / BMv;
+        int tiles_c = (Ncout + BC - 1) / BC;
+        dim3 grid(tiles_m * tiles_c);
+        hipLaunchKernelGGL((sodkit::igemm_kernel<T, MODE, BC, BMv>), grid,
+                           dim3(256), 0, stream.stream(),
+                           reinterpret_cast<const T *>(src.data_ptr()),
+                           reinterpret_cast<const T *>(w.data_ptr()), bias_ptr,
+                           reinterpret_cast<T *>(out.data_ptr()),
+                           Ns, Hs, Ws, Cs, P, Q, Ncout, R, S, stride, pad, dil,
+                           (int)M_total, K_total, tiles_m);
+    };
+    using c64 = std::integral_constant<int, 64>;
+    using c128 = std::integral_constant<int, 128>;
+    bool small_c = Ncout <= 64;
+    long tiles_m128 = (M_total + 127) / 128;
+    bool small_m = tiles_m128 * ((Ncout + 127) / 128) < 192;
+    if (small_c) {
+        if (small_m) launch(c64{}, c64{});
+        else launch(c64{}, c128{});
+    } else {
+        if (small_m) launch(c128{}, c64{});
+        else launch(c128{}, c128{});
+    }
+}
+
+}  // namespace
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                         int64_t stride, int64_t pad, int64_t dil) {
+    CHECK_GPU(x); CHECK_GPU(w);
+    TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
+    int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) ||
+                (C == 1 && x.is_contiguous()),
+                "conv2d_fwd needs channels_last input");
+    int Cout = w.size(0), R = w.size(1), S = w.size(2);
+    TORCH_CHECK(w.size(3) == C, "weight Cin mismatch");
+    TORCH_CHECK(w.is_contiguous());
+    int P = out_dim(H, R, stride, pad, dil);
+    int Q = out_dim(W_, S, stride, pad, dil);
+    long M = (long)N * P * Q;
+    int K = R * S * C;
+    auto out = torch::empty({N, Cout, P, Q},
+                            x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    torch::Tensor bias32;
+    if (bias.defined() && bias.numel()) bias32 = bias.to(torch::kFloat32);
+    AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_fwd",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            launch_igemm<__hip_bfloat16, 0>(x, w, bias32, out, N, H, W_, C,
+                                            P, Q, Cout, R, S, stride, pad, dil, M, K);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            launch_igemm<__half, 0>(x, w, bias32, out, N, H, W_, C,
+                                    P, Q, Cout, R, S, stride, pad, dil, M, K);
+        }));
+    return out;
+}
+
+torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w_t, int64_t H,
+                           int64_t W_, int64_t stride, int64_t pad, int64_t dil) {
+    CHECK_GPU(dy); CHECK_GPU(w_t);
+    int N = dy.size(0), Cout = dy.size(1), P = dy.size(2), Q = dy.size(3);
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) ||
+                (Cout == 1 && dy.is_contiguous()),
+                "conv2d_dgrad needs channels_last dy");
+    // w_t: [Cin][R][S][Cout]
+    int Cin = w_t.size(0), R = w_t.size(1), S = w_t.size(2);
+    TORCH_CHECK(w_t.size(3) == Cout && w_t.is_contiguous());
+    long M = (long)N * H * W_;
+    int K = R * S * Cout;
+    auto dx = torch::empty({N, Cin, H, W_},
+                           dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    torch::Tensor nobias;
+    AT_DISPATCH_SWITCH(dy.scalar_type(), "conv2d_dgrad",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            launch_igemm<__hip_bfloat16, 1>(dy, w_t, nobias, dx, N, P, Q, Cout,
+                                            H, W_, Cin, R, S, stride, pad, dil, M, K);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            launch_igemm<__half, 1>(dy, w_t, nobias, dx, N, P, Q, Cout,
+                                    H, W_, Cin, R, S, stride, pad, dil, M, K);
+        }));
+    return dx;
+}
+
+torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
+                           int64_t S, int64_t stride, int64_t pad, int64_t dil) {
+    CHECK_GPU(x); CHECK_GPU(dy);
+    int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    int Cout = dy.size(1), P = dy.size(2), Q = dy.size(3);
+    TORCH_CHECK(dy.size(0) == N);
+    long M = (long)N * P * Q;
+    int Kw = R * S * C;
+    auto stream = at::hip::getCurrentHIPStream();
+
+    auto run = [&](auto t_tag, auto bc_tag) -> torch::Tensor {
+        using T = typename decltype(t_tag)::type;
+        constexpr int BC = decltype(bc_tag)::value;
+        constexpr int BJ = 128;
+        int tiles_c = (Cout + BC - 1) / BC;
+        int tiles_j = (Kw + BJ - 1) / BJ;
+        int tiles = tiles_c * tiles_j;
+        int splitk = std::max(1, std::min(64, 768 / std::max(tiles, 1)));
+        long m_per_split = ((M + splitk - 1) / splitk + 63) / 64 * 64;
+        splitk = (int)((M + m_per_split - 1) / m_per_split);
+        auto partial = torch::empty({splitk, (long)Cout * Kw},
+                                    x.options().dtype(torch::kFloat32));
+        dim3 grid(tiles, splitk);
+        hipLaunchKernelGGL((sodkit::wgrad_kernel<T, BC, BJ>), grid, dim3(256),
+                           0, stream.stream(),
+                           reinterpret_cast<const T *>(x.data_ptr()),
+                           reinterpret_cast<const T *>(dy.data_ptr()),
+                           partial.data_ptr<float>(),
+                           N, H, W_, C, P, Q, Cout, (int)R, (int)S,
+                           (int)stride, (int)pad, (int)dil, Kw, (int)M,
+                           (int)m_per_split, tiles_c);
+        auto dw = torch::empty({Cout, R, S, C}, x.options());
+        long n_elems = (long)Cout * Kw;
+        int blocks = std::min<long>((n_elems + 255) / 256, 2048);
+        hipLaunchKernelGGL((sodkit::splitk_reduce_kernel<T>), dim3(blocks),
+                           dim3(256), 0, stream.stream(),
+                           partial.data_ptr<float>(),
+                           reinterpret_cast<T *>(dw.data_ptr()), n_elems, splitk);
+        return dw;
+    };
+
+    struct BF { using type = __hip_bfloat16; };
+    struct FP { using type = __half; };
+    using c64 = std::integral_constant<int, 64>;
+    using c128 = std::integral_constant<int, 128>;
+    torch::Tensor dw;
+    bool small_c = Cout <= 64;
+    AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_wgrad",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            dw = small_c ? run(BF{}, c64{}) : run(BF{}, c128{});
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            dw = small_c ? run(FP{}, c64{}) : run(FP{}, c128{});
+        }));
+    return dw;
+}
+
+torch::Tensor channel_sum(torch::Tensor x) {
+    CHECK_GPU(x);
+    int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    long M = (long)N * H * W_;
+    auto out = torch::zeros({C}, x.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    dim3 grid(std::min<long>((M + 255) / 256, 1024), (C + 255) / 256);
+    AT_DISPATCH_SWITCH(x.scalar_type(), "channel_sum",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            hipLaunchKernelGGL((sodkit::channel_sum_kernel<__hip_bfloat16>),
+                               grid, dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const __hip_bfloat16 *>(x.data_ptr()),
+                               out.data_ptr<float>(), M, C);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            hipLaunchKernelGGL((sodkit::channel_sum_kernel<__half>),
+                               grid, dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const __half *>(x.data_ptr()),
+                               out.data_ptr<float>(), M, C);
+        })
+        AT_DISPATCH_CASE(at::kFloat, [&] {
+            hipLaunchKernelGGL((sodkit::channel_sum_kernel<float>),
+                               grid, dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const float *>(x.data_ptr()),
+                               out.data_ptr<float>(), M, C);
+        }));
+    return out;
+}

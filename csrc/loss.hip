@@ -1,0 +1,371 @@
+// sodkit — fused hybrid pixel-wise loss (sigmoid + BCE + soft-IoU + SSIM)
+// for CDNA4 — SURVEY §2.3 K5, [BJ:5]: 11x11 gaussian SSIM window staged in
+// LDS, analytic backward (3 coefficient center-maps + one gaussian-gather
+// pass), all in fp32.
+//
+// Forward:
+//   k1 pointwise: p = sigmoid(z); BCE sum; per-image (sum p*t, sum p, sum t)
+//   k2 windowed : per-center SSIM from gaussian-weighted window stats with
+//                 edge renormalization (window mass wsum = rowmass*colmass,
+//                 separable, computed analytically); stores the 5 stat maps
+//                 (mu_p, mu_t, s_pp, s_tt, s_pt) for backward; SSIM sum.
+// Backward (d loss / d z):
+//   k3 pointwise: coefficient maps A1..A5 from the stats, with the
+//                 -w_ssim*go/Ntot/wsum_i factor folded in at the centers
+//   k4 windowed : gp(j) = corr(A1) + 2 p_j corr(A2) - 2 corr(A3)
+//                        + t_j corr(A4) - corr(A5)   (5 gaussian gathers)
+//                 dz = go*wb*(p-t)/Ntot + [go*wi*iou_coef + gp] * p(1-p)
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace sodkit {
+
+constexpr int WIN = 11;
+constexpr int HALF = WIN / 2;
+constexpr float SSIM_C1 = 0.01f * 0.01f;
+constexpr float SSIM_C2 = 0.03f * 0.03f;
+
+SOD_DEVINL void gauss_table(float g[WIN]) {
+    float s = 0.f;
+#pragma unroll
+    for (int i = 0; i < WIN; ++i) {
+        g[i] = __expf(-((i - HALF) * (i - HALF)) / (2.f * 1.5f * 1.5f));
+        s += g[i];
+    }
+#pragma unroll
+    for (int i = 0; i < WIN; ++i) g[i] /= s;
+}
+
+SOD_DEVINL float axis_mass(const float g[WIN], int center, int size) {
+    float m = 0.f;
+#pragma unroll
+    for (int d = 0; d < WIN; ++d) {
+        int idx = center + d - HALF;
+        if (idx >= 0 && idx < size) m += g[d];
+    }
+    return m;
+}
+
+// ---- k1: pointwise sigmoid / BCE / IoU sums.  grid = (chunks, N) ----
+template <typename T>
+__global__ void loss_pointwise_kernel(const T *__restrict__ z,
+                                      const float *__restrict__ t,
+                                      float *__restrict__ p,
+                                      float *__restrict__ bce_sum,
+                                      float *__restrict__ img_sums, long HW) {
+    __shared__ float scratch[16];
+    int n = blockIdx.y;
+    const T *zb = z + (long)n * HW;
+    const float *tb = t + (long)n * HW;
+    float *pb = p + (long)n * HW;
+    float bce = 0.f, spt = 0.f, sp = 0.f, st = 0.f;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < HW;
+         i += (long)gridDim.x * blockDim.x) {
+        float zv = ElemTraits<T>::to_f32(zb[i]);
+        float tv = tb[i];
+        float pv = 1.f / (1.f + __expf(-zv));
+        pb[i] = pv;
+        // stable BCE-with-logits: max(z,0) - z*t + log1p(exp(-|z|))
+        bce += fmaxf(zv, 0.f) - zv * tv + log1pf(__expf(-fabsf(zv)));
+        spt += pv * tv;
+        sp += pv;
+        st += tv;
+    }
+    float r;
+    r = block_reduce_sum<4>(bce, scratch);
+    if (threadIdx.x == 0) atomicAdd(bce_sum, r);
+    __syncthreads();
+    r = block_reduce_sum<4>(spt, scratch);
+    if (threadIdx.x == 0) atomicAdd(&img_sums[n * 3 + 0], r);
+    __syncthreads();
+    r = block_reduce_sum<4>(sp, scratch);
+    if (threadIdx.x == 0) atomicAdd(&img_sums[n * 3 + 1], r);
+    __syncthreads();
+    r = block_reduce_sum<4>(st, scratch);
+    if (threadIdx.x == 0) atomicAdd(&img_sums[n * 3 + 2], r);
+}
+
+constexpr int TILE = 16;
+constexpr int PATCH = TILE + 2 * HALF;  // 26
+
+// ---- k2: SSIM stats + value.  grid = (Wtiles, Htiles, N), block 16x16 ----
+__global__ void ssim_stats_kernel(const float *__restrict__ p,
+                                  const float *__restrict__ t,
+                                  float *__restrict__ stats,  // [5][N*HW]
+                                  float *__restrict__ ssim_sum,
+                                  int H, int W) {
+    __shared__ float pp[PATCH][PATCH];
+    __shared__ float tt[PATCH][PATCH];
+    __shared__ float scratch[16];
+    float g[WIN];
+    gauss_table(g);
+    int n = blockIdx.z;
+    long HW = (long)H * W;
+    const float *pb = p + n * HW;
+    const float *tb = t + n * HW;
+    int h00 = blockIdx.y * TILE - HALF;
+    int w00 = blockIdx.x * TILE - HALF;
+    int tid = threadIdx.y * TILE + threadIdx.x;
+    for (int i = tid; i < PATCH * PATCH; i += TILE * TILE) {
+        int dh = i / PATCH, dw = i % PATCH;
+        int h = h00 + dh, w = w00 + dw;
+        bool ok = (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
+        pp[dh][dw] = ok ? pb[(long)h * W + w] : 0.f;
+        tt[dh][dw] = ok ? tb[(long)h * W + w] : 0.f;
+    }
+    __syncthreads();
+    int oh = blockIdx.y * TILE + threadIdx.y;
+    int ow = blockIdx.x * TILE + threadIdx.x;
+    float val = 0.f;
+    if (oh < H && ow < W) {
+        float wsum = axis_mass(g, oh, H) * axis_mass(g, ow, W);
+        float sp = 0.f, st_ = 0.f, spp = 0.f, stt = 0.f, spt = 0.f;
+#pragma unroll
+        for (int dh = 0; dh < WIN; ++dh) {
+            float rp = 0.f, rt = 0.f, rpp = 0.f, rtt = 0.f, rpt = 0.f;
+#pragma unroll
+            for (int dw = 0; dw < WIN; ++dw) {
+                float pv = pp[threadIdx.y + dh][threadIdx.x + dw];
+                float tv = tt[threadIdx.y + dh][threadIdx.x + dw];
+                float gw = g[dw];
+                rp += gw * pv;
+                rt += gw * tv;
+                rpp += gw * pv * pv;
+                rtt += gw * tv * tv;
+                rpt += gw * pv * tv;
+            }
+            float gh = g[dh];
+            sp += gh * rp; st_ += gh * rt; spp += gh * rpp;
+            stt += gh * rtt; spt += gh * rpt;
+        }
+        float inv = 1.f / wsum;
+        float mu_p = sp * inv, mu_t = st_ * inv;
+        float s_pp = spp * inv - mu_p * mu_p;
+        float s_tt = stt * inv - mu_t * mu_t;
+        float s_pt = spt * inv - mu_p * mu_t;
+        float N1 = 2.f * mu_p * mu_t + SSIM_C1;
+        float N2 = 2.f * s_pt + SSIM_C2;
+        float D1 = mu_p * mu_p + mu_t * mu_t + SSIM_C1;
+        float D2 = s_pp + s_tt + SSIM_C2;
+        val = (N1 * N2) / (D1 * D2);
+        long idx = (long)n * HW + (long)oh * W + ow;
+        long total = (long)gridDim.z * HW;
+        stats[0 * total + idx] = mu_p;
+        stats[1 * total + idx] = mu_t;
+        stats[2 * total + idx] = s_pp;
+        stats[3 * total + idx] = s_tt;
+        stats[4 * total + idx] = s_pt;
+    }
+    __syncthreads();
+    float r = block_reduce_sum<4>(val, scratch);
+    if (tid == 0) atomicAdd(ssim_sum, r);
+}
+
+// ---- k3: coefficient center-maps for the SSIM backward ----
+__global__ void ssim_coef_kernel(const float *__restrict__ stats,
+                                 float *__restrict__ coef,  // [5][N*HW]
+                                 float scale,               // -ws*go/Ntot
+                                 int N, int H, int W) {
+    float g[WIN];
+    gauss_table(g);
+    long HW = (long)H * W;
+    long total = (long)N * HW;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += (long)gridDim.x * blockDim.x) {
+        long rem = i % HW;
+        int oh = (int)(rem / W), ow = (int)(rem % W);
+        float mu_p = stats[0 * total + i];
+        float mu_t = stats[1 * total + i];
+        float s_pp = stats[2 * total + i];
+        float s_tt = stats[3 * total + i];
+        float s_pt = stats[4 * total + i];
+        float N1 = 2.f * mu_p * mu_t + SSIM_C1;
+        float N2 = 2.f * s_pt + SSIM_C2;
+        float D1 = mu_p * mu_p + mu_t * mu_t + SSIM_C1;
+        float D2 = s_pp + s_tt + SSIM_C2;
+        float den = D1 * D2;
+        float f = (N1 * N2) / den;
+        // partials in (mu_p, sigma_pp, sigma_pt) holding the others fixed
+        float d_mu = (2.f * mu_t * N2) / den - f * (2.f * mu_p) / D1;
+        float d_spp = -f / D2;
+        float d_spt = 2.f * N1 / den;
+        float wsum = axis_mass(g, oh, H) * axis_mass(g, ow, W);
+        float s = scale / wsum;
+        coef[0 * total + i] = s * d_mu;             // A1
+        coef[1 * total + i] = s * d_spp;            // A2
+        coef[2 * total + i] = s * d_spp * mu_p;     // A3
+        coef[3 * total + i] = s * d_spt;            // A4
+        coef[4 * total + i] = s * d_spt * mu_t;     // A5
+    }
+}
+
+// ---- k4: gather + combine -> dz (fp32) ----
+template <bool HAS_SSIM>
+__global__ void loss_bwd_kernel(const float *__restrict__ p,
+                                const float *__restrict__ t,
+                                const float *__restrict__ img_sums,
+                                const float *__restrict__ coef,
+                                const float *__restrict__ go,
+                                float *__restrict__ dz,
+                                float wb, float wi,
+                                int Nimg, int H, int W) {
+    __shared__ float patch[5][PATCH][PATCH];
+    float g[WIN];
+    gauss_table(g);
+    int n = blockIdx.z;
+    long HW = (long)H * W;
+    long total = (long)Nimg * HW;
+    float gov = go[0];
+    int h00 = blockIdx.y * TILE - HALF;
+    int w00 = blockIdx.x * TILE - HALF;
+    int tid = threadIdx.y * TILE + threadIdx.x;
+    if (HAS_SSIM) {
+        for (int i = tid; i < PATCH * PATCH; i += TILE * TILE) {
+            int dh = i / PATCH, dw = i % PATCH;
+            int h = h00 + dh, w = w00 + dw;
+            bool ok = (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
+            long idx = (long)n * HW + (long)h * W + w;
+#pragma unroll
+            for (int m = 0; m < 5; ++m)
+                patch[m][dh][dw] = ok ? coef[m * total + idx] : 0.f;
+        }
+        __syncthreads();
+    }
+    int oh = blockIdx.y * TILE + threadIdx.y;
+    int ow = blockIdx.x * TILE + threadIdx.x;
+    if (oh >= H || ow >= W) return;
+    long idx = (long)n * HW + (long)oh * W + ow;
+    float pv = p[idx], tv = t[idx];
+    float c1 = 0.f, c2 = 0.f, c3 = 0.f, c4 = 0.f, c5 = 0.f;
+    if (HAS_SSIM) {
+#pragma unroll
+        for (int dh = 0; dh < WIN; ++dh) {
+            float r1 = 0.f, r2 = 0.f, r3 = 0.f, r4 = 0.f, r5 = 0.f;
+#pragma unroll
+            for (int dw = 0; dw < WIN; ++dw) {
+                float gw = g[dw];
+                r1 += gw * patch[0][threadIdx.y + dh][threadIdx.x + dw];
+                r2 += gw * patch[1][threadIdx.y + dh][threadIdx.x + dw];
+                r3 += gw * patch[2][threadIdx.y + dh][threadIdx.x + dw];
+                r4 += gw * patch[3][threadIdx.y + dh][threadIdx.x + dw];
+                r5 += gw * patch[4][threadIdx.y + dh][threadIdx.x + dw];
+            }
+            float gh = g[dh];
+            c1 += gh * r1; c2 += gh * r2; c3 += gh * r3; c4 += gh * r4; c5 += gh * r5;
+        }
+    }
+    // gp carries the -ws/Ntot scale (folded at the centers); go multiplies in
+    float gp = c1 + 2.f * pv * c2 - 2.f * c3 + tv * c4 - c5;
+    float inv_ntot = 1.f / (float)total;
+    float dz_v = gov * wb * (pv - tv) * inv_ntot;
+    float iou_coef = 0.f;
+    if (wi != 0.f) {
+        float I = img_sums[n * 3 + 0];
+        float U = img_sums[n * 3 + 1] + img_sums[n * 3 + 2] - I;
+        float u1 = U + 1.f;
+        // d[(1/Nimg) sum (1 - (I+1)/(U+1))]/dp = ((I+1) - t*(I+U+2))/(U+1)^2/Nimg
+        iou_coef = ((I + 1.f) - tv * (I + U + 2.f)) / (u1 * u1) / (float)Nimg;
+    }
+    float sig_d = pv * (1.f - pv);
+    dz_v += (gov * wi * iou_coef + gov * gp) * sig_d;
+    dz[idx] = dz_v;
+}
+
+}  // namespace sodkit
+
+// ===========================================================================
+// Host glue
+// ===========================================================================
+
+std::vector<torch::Tensor> loss_fwd(torch::Tensor z, torch::Tensor t,
+                                    bool need_ssim) {
+    TORCH_CHECK(z.is_cuda() && z.dim() == 4 && z.size(1) == 1);
+    int N = z.size(0), H = z.size(2), W = z.size(3);
+    long HW = (long)H * W;
+    auto opts = z.options().dtype(torch::kFloat32);
+    auto p = torch::empty({N, 1, H, W}, opts);
+    auto bce_sum = torch::zeros({1}, opts);
+    auto img_sums = torch::zeros({N, 3}, opts);
+    auto ssim_sum = torch::zeros({1}, opts);
+    torch::Tensor stats;
+    auto tf = t.to(torch::kFloat32).contiguous();
+    auto stream = at::hip::getCurrentHIPStream();
+    dim3 grid1((unsigned)std::min<long>((HW + 255) / 256, 256), N);
+    AT_DISPATCH_SWITCH(z.scalar_type(), "loss_fwd",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            hipLaunchKernelGGL((sodkit::loss_pointwise_kernel<__hip_bfloat16>),
+                               grid1, dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const __hip_bfloat16 *>(z.data_ptr()),
+                               tf.data_ptr<float>(), p.data_ptr<float>(),
+                               bce_sum.data_ptr<float>(), img_sums.data_ptr<float>(), HW);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            hipLaunchKernelGGL((sodkit::loss_pointwise_kernel<__half>),
+                               grid1, dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const __half *>(z.data_ptr()),
+                               tf.data_ptr<float>(), p.data_ptr<float>(),
+                               bce_sum.data_ptr<float>(), img_sums.data_ptr<float>(), HW);
+        })
+        AT_DISPATCH_CASE(at::kFloat, [&] {
+            hipLaunchKernelGGL((sodkit::loss_pointwise_kernel<float>),
+                               grid1, dim3(256), 0, stream.stream(),
+                               z.data_ptr<float>(),
+                               tf.data_ptr<float>(), p.data_ptr<float>(),
+                               bce_sum.data_ptr<float>(), img_sums.data_ptr<float>(), HW);
+        }));
+    if (need_ssim) {
+        stats = torch::empty({5, (long)N * H * W}, opts);
+        dim3 grid2((W + 15) / 16, (H + 15) / 16, N);
+        hipLaunchKernelGGL(sodkit::ssim_stats_kernel, grid2, dim3(16, 16), 0,
+                           stream.stream(), p.data_ptr<float>(),
+                           tf.data_ptr<float>(), stats.data_ptr<float>(),
+                           ssim_sum.data_ptr<float>(), H, W);
+    } else {
+        stats = torch::empty({0}, opts);
+    }
+    return {bce_sum, img_sums, ssim_sum, p, stats};
+}
+
+torch::Tensor loss_bwd(torch::Tensor p, torch::Tensor t, torch::Tensor img_sums,
+                       torch::Tensor stats, double wb, double wi, double ws,
+                       torch::Tensor go) {
+    int N = p.size(0), H = p.size(2), W = p.size(3);
+    long total = (long)N * H * W;
+    auto opts = p.options();
+    auto dz = torch::empty({N, 1, H, W}, opts);
+    auto stream = at::hip::getCurrentHIPStream();
+    bool has_ssim = ws != 0.0 && stats.numel() > 0;
+    torch::Tensor coef;
+    if (has_ssim) {
+        coef = torch::empty({5, total}, opts);
+        // NOTE: go is folded on the python side into `scale` only via device
+        // tensor in k4; here fold -ws/Ntot (go applied in k4 via... see below)
+        // => we fold go in k4 for bce/iou but coef needs go too; read it via
+        //    a separate tiny kernel would sync. Instead fold go=1 here and
+        //    multiply gp by go in k4?  gp is linear in coef, so multiply the
+        //    whole gp by go in k4.  scale = -ws/Ntot.
+        float scale = (float)(-ws / (double)total);
+        int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+        hipLaunchKernelGGL(sodkit::ssim_coef_kernel, dim3(blocks), dim3(256), 0,
+                           stream.stream(), stats.data_ptr<float>(),
+                           coef.data_ptr<float>(), scale, N, H, W);
+    }
+    dim3 grid4((W + 15) / 16, (H + 15) / 16, N);
+    auto tf = t.contiguous();
+    if (has_ssim)
+        hipLaunchKernelGGL(sodkit::loss_bwd_kernel<true>, grid4, dim3(16, 16),
+                           0, stream.stream(), p.data_ptr<float>(),
+                           tf.data_ptr<float>(), img_sums.data_ptr<float>(),
+                           coef.data_ptr<float>(), go.data_ptr<float>(),
+                           dz.data_ptr<float>(), (float)wb, (float)wi, N, H, W);
+    else
+        hipLaunchKernelGGL(sodkit::loss_bwd_kernel<false>, grid4, dim3(16, 16),
+                           0, stream.stream(), p.data_ptr<float>(),
+                           tf.data_ptr<float>(), img_sums.data_ptr<float>(),
+                           nullptr, go.data_ptr<float>(),
+                           dz.data_ptr<float>(), (float)wb, (float)wi, N, H, W);
+    return dz;
+}

@@ -1,0 +1,66 @@
+// sodkit — fused multi-tensor SGD for CDNA4.
+// One launch sweeps every parameter chunk: bf16/fp16 grad upcast, grad scale
+// (1/world * 1/loss_scale folded), weight decay, momentum, fp32 master
+// update, and the fp32->bf16/fp16 param write-back — all in a single pass.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace sodkit {
+
+constexpr int SGD_CHUNK = 16384;
+
+// ptrs  [T][4]: param, grad, master, momentum  (device addresses)
+// chunks[K][2]: tensor index, element offset
+template <typename T, bool NESTEROV>
+__global__ void sgd_fused_kernel(const long *__restrict__ ptrs,
+                                 const long *__restrict__ numels,
+                                 const float *__restrict__ wds,
+                                 const long *__restrict__ chunks,
+                                 float lr, float mu, float gscale) {
+    long ti = chunks[blockIdx.x * 2 + 0];
+    long off = chunks[blockIdx.x * 2 + 1];
+    T *param = reinterpret_cast<T *>(ptrs[ti * 4 + 0]);
+    const T *grad = reinterpret_cast<const T *>(ptrs[ti * 4 + 1]);
+    float *master = reinterpret_cast<float *>(ptrs[ti * 4 + 2]);
+    float *mom = reinterpret_cast<float *>(ptrs[ti * 4 + 3]);
+    float wd = wds[ti];
+    long end = min(off + (long)SGD_CHUNK, numels[ti]);
+    for (long i = off + threadIdx.x; i < end; i += blockDim.x) {
+        float g = ElemTraits<T>::to_f32(grad[i]) * gscale;
+        float w = master[i];
+        g += wd * w;
+        float m = mu * mom[i] + g;
+        mom[i] = m;
+        float upd = NESTEROV ? (g + mu * m) : m;
+        w -= lr * upd;
+        master[i] = w;
+        param[i] = ElemTraits<T>::from_f32(w);
+    }
+}
+
+}  // namespace sodkit
+
+void sgd_fused(torch::Tensor ptrs, torch::Tensor numels, torch::Tensor wds,
+               torch::Tensor chunks, int64_t dtype_code, double lr, double mu,
+               double gscale, bool nesterov) {
+    int K = chunks.size(0);
+    auto stream = at::hip::getCurrentHIPStream();
+    auto launch = [&](auto t_tag, auto nv) {
+        using T = typename decltype(t_tag)::type;
+        hipLaunchKernelGGL((sodkit::sgd_fused_kernel<T, decltype(nv)::value>),
+                           dim3(K), dim3(256), 0, stream.stream(),
+                           ptrs.data_ptr<long>(), numels.data_ptr<long>(),
+                           wds.data_ptr<float>(), chunks.data_ptr<long>(),
+                           (float)lr, (float)mu, (float)gscale);
+    };
+    struct F32 { using type = float; };
+    struct BF { using type = __hip_bfloat16; };
+    struct F16 { using type = __half; };
+    using tt = std::true_type; using ff = std::false_type;
+    if (dtype_code == 0) { nesterov ? launch(F32{}, tt{}) : launch(F32{}, ff{}); }
+    else if (dtype_code == 1) { nesterov ? launch(BF{}, tt{}) : launch(BF{}, ff{}); }
+    else { nesterov ? launch(F16{}, tt{}) : launch(F16{}, ff{}); }
+}

@@ -1,0 +1,38 @@
+"""Build the in-tree HIP extension sodkit._C for gfx950 (MI355X).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The .so lands inside sodkit/ so it travels with the repo snapshot to GPU
+boxes (no JIT cache dependence)."""
+
+import os
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+SRC = [
+    "csrc/bind.cpp",
+    "csrc/conv_igemm.hip",
+    "csrc/bnrelu.hip",
+    "csrc/bilinear.hip",
+    "csrc/loss.hip",
+    "csrc/sgd.hip",
+]
+
+setup(
+    name="sodkit-kernels",
+    ext_modules=[
+        CUDAExtension(
+            name="sodkit._C",
+            sources=SRC,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)

@@ -15,6 +15,8 @@ engine (`sodkit.parallel.reducer`) is a bucketed, backward-overlapped
 all-reduce over RCCL on a side HIP stream.
 """
 
+import torch as _torch  # noqa: F401  (the _C extension links against torch libs)
+
 __version__ = "0.1.0"
 
 from . import config  # noqa: F401

@@ -1,10 +1,11 @@
 """Building-block modules — parity with the reference's module/BaseBlocks.py
 (`BasicConv2d` = Conv+BN+ReLU) — built on sodkit.ops.
 
-Weights live in RSCK layout [R, S, Cin, Cout] (the implicit-GEMM B-operand
-layout the MI355X kernels consume).  BN affine params and running stats stay
-fp32 regardless of the compute dtype; `convert_precision` casts only conv
-weights/biases.
+Weights live in KRSC layout [Cout, R, S, Cin] (torch's channels-last weight
+layout): the MFMA implicit-GEMM A-operand is then W[cout][(r,s,cin)]
+row-major, so the MI355X kernels stage it into LDS coalesced with no
+transpose.  BN affine params and running stats stay fp32 regardless of the
+compute dtype; `convert_precision` casts only conv weights/biases.
 """
 
 from __future__ import annotations
@@ -24,7 +25,7 @@ class SodConv2d(nn.Module):
         super().__init__()
         self.cin, self.cout, self.k = cin, cout, k
         self.stride, self.padding, self.dilation = stride, padding, dilation
-        self.weight = nn.Parameter(torch.empty(k, k, cin, cout))
+        self.weight = nn.Parameter(torch.empty(cout, k, k, cin))
         # kaiming-normal fan_out (torchvision resnet convention)
         fan_out = k * k * cout
         with torch.no_grad():

@@ -1,9 +1,9 @@
 """Convolution op: MFMA implicit-GEMM HIP kernels on GPU, ATen on CPU.
 
-Weight layout is RSCK ``[R, S, Cin, Cout]`` — the implicit-GEMM B-operand
-layout (K-dim (r,s,c) as rows, Cout contiguous), chosen for the MI355X
-kernels rather than inherited from torch.  Activations are NHWC-physical
-(torch ``channels_last``) on GPU, plain NCHW on CPU.
+Weight layout is KRSC ``[Cout, R, S, Cin]`` (torch's channels-last weight
+layout): W[cout][(r,s,cin)] row-major IS the implicit-GEMM A-operand, staged
+into LDS with coalesced loads and no transpose.  Activations are
+NHWC-physical (torch ``channels_last``) on GPU, plain NCHW on CPU.
 
 Kernels (csrc/conv_igemm.hip): K1/K2/K6/K10/K13 of SURVEY §2.3 — one
 implicit-GEMM family covering 1x1 (pure GEMM), 3x3, 7x7 stem, strided and
@@ -44,21 +44,21 @@ class _ConvFn(torch.autograd.Function):
         dy = _cl(dy)
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            # w^T layout [R, S, Cout, Cin] for the dgrad B-operand
-            w_t = w.permute(0, 1, 3, 2).contiguous()
+            # dgrad A-operand layout [Cin, R, S, Cout] (a small permuted copy)
+            w_t = w.permute(3, 1, 2, 0).contiguous()
             dx = ext().conv2d_dgrad(dy, w_t, x.shape[2], x.shape[3],
                                     stride, padding, dilation)
         if ctx.needs_input_grad[1]:
-            dw = ext().conv2d_wgrad(x, dy, w.shape[0], w.shape[1],
+            dw = ext().conv2d_wgrad(x, dy, w.shape[1], w.shape[2],
                                     stride, padding, dilation).to(w.dtype)
         if has_bias and ctx.needs_input_grad[2]:
             db = ext().channel_sum(dy).to(w.dtype)
         return dx, dw, db, None, None, None
 
 
-def conv2d(x: torch.Tensor, w_rsck: torch.Tensor,
+def conv2d(x: torch.Tensor, w_krsc: torch.Tensor,
            bias: Optional[torch.Tensor] = None, stride: int = 1,
            padding: int = 0, dilation: int = 1) -> torch.Tensor:
     if x.is_cuda:
-        return _ConvFn.apply(x, w_rsck, bias, stride, padding, dilation)
-    return reference.conv2d(x, w_rsck, bias, stride, padding, dilation)
+        return _ConvFn.apply(x, w_krsc, bias, stride, padding, dilation)
+    return reference.conv2d(x, w_krsc, bias, stride, padding, dilation)

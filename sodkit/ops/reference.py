@@ -25,14 +25,15 @@ def _fp(x: torch.Tensor) -> torch.Tensor:
     return x if x.dtype == torch.float64 else x.float()
 
 # ---------------------------------------------------------------------------
-# Convolution.  sodkit stores conv weights in RSCK layout [R, S, Cin, Cout]
-# (the implicit-GEMM B-operand layout, K-dim (r,s,c) row-major with Cout
-# contiguous).  The reference path permutes to torch's [Cout, Cin, R, S].
+# Convolution.  sodkit stores conv weights in KRSC layout [Cout, R, S, Cin]
+# (torch's channels-last weight layout): the MFMA implicit-GEMM A-operand is
+# then W[cout][(r,s,cin)] row-major — coalesced LDS staging with no transpose.
+# The reference path permutes to torch's [Cout, Cin, R, S].
 # ---------------------------------------------------------------------------
 
-def conv2d(x: torch.Tensor, w_rsck: torch.Tensor, bias: Optional[torch.Tensor],
+def conv2d(x: torch.Tensor, w_krsc: torch.Tensor, bias: Optional[torch.Tensor],
            stride: int, padding: int, dilation: int) -> torch.Tensor:
-    w = w_rsck.permute(3, 2, 0, 1)  # [Cout, Cin, R, S]
+    w = w_krsc.permute(0, 3, 1, 2)  # [Cout, Cin, R, S]
     return F.conv2d(x, w, bias, stride=stride, padding=padding, dilation=dilation)
 
 

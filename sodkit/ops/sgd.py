@@ -18,7 +18,7 @@ import torch
 from . import reference
 from ._backend import ext
 
-_CHUNK = 1 << 21  # elements per kernel block-chunk
+_CHUNK = 16384  # elements per kernel block-chunk (must match csrc/sgd.hip)
 
 
 def split_decay_groups(model: torch.nn.Module):

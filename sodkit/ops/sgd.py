@@ -55,29 +55,31 @@ class SodSGD:
                 }
         self._tables = None  # lazily built device pointer tables (GPU)
 
-    # -- fused GPU tables ---------------------------------------------------
+    # -- fused GPU tables (one table + one launch per param dtype: conv
+    #    weights are bf16/fp16 while BN affine params stay fp32) -----------
     def _build_tables(self):
-        dev = None
-        ptrs, metas = [], []
+        groups_by_dtype: dict[torch.dtype, list] = {}
         for g in self.groups:
             wd = g["weight_decay"]
             for p in g["params"]:
                 assert p.grad is not None, "grads must be pre-assigned (reducer bucket views)"
                 st = self.state[p]
-                dev = p.device
-                ptrs.append((p.data_ptr(), p.grad.data_ptr(),
-                             st["master"].data_ptr(),
-                             st["momentum_buffer"].data_ptr()))
-                metas.append((p.numel(), wd))
-        ptr_t = torch.tensor(ptrs, dtype=torch.int64, device=dev)        # [T,4]
-        numel_t = torch.tensor([m[0] for m in metas], dtype=torch.int64, device=dev)
-        wd_t = torch.tensor([m[1] for m in metas], dtype=torch.float32, device=dev)
-        chunks = []
-        for ti, (n, _) in enumerate(metas):
-            for off in range(0, n, _CHUNK):
-                chunks.append((ti, off))
-        chunk_t = torch.tensor(chunks, dtype=torch.int64, device=dev)    # [K,2]
-        self._tables = (ptr_t, numel_t, wd_t, chunk_t)
+                groups_by_dtype.setdefault(p.dtype, []).append(
+                    ((p.data_ptr(), p.grad.data_ptr(), st["master"].data_ptr(),
+                      st["momentum_buffer"].data_ptr()), p.numel(), wd, p.device))
+        self._tables = []
+        codes = {torch.float32: 0, torch.bfloat16: 1, torch.float16: 2}
+        for dtype, entries in groups_by_dtype.items():
+            dev = entries[0][3]
+            ptr_t = torch.tensor([e[0] for e in entries], dtype=torch.int64, device=dev)
+            numel_t = torch.tensor([e[1] for e in entries], dtype=torch.int64, device=dev)
+            wd_t = torch.tensor([e[2] for e in entries], dtype=torch.float32, device=dev)
+            chunks = []
+            for ti, e in enumerate(entries):
+                for off in range(0, e[1], _CHUNK):
+                    chunks.append((ti, off))
+            chunk_t = torch.tensor(chunks, dtype=torch.int64, device=dev)
+            self._tables.append((codes[dtype], ptr_t, numel_t, wd_t, chunk_t))
 
     @torch.no_grad()
     def step(self, lr: float, grad_scale: float = 1.0):
@@ -87,11 +89,9 @@ class SodSGD:
         if first.is_cuda:
             if self._tables is None:
                 self._build_tables()
-            ptr_t, numel_t, wd_t, chunk_t = self._tables
-            # dtype code: 0=fp32 1=bf16 2=fp16 (uniform across params)
-            dt = {torch.float32: 0, torch.bfloat16: 1, torch.float16: 2}[first.dtype]
-            ext().sgd_fused(ptr_t, numel_t, wd_t, chunk_t, dt, float(lr),
-                            self.momentum, float(grad_scale), self.nesterov)
+            for dt, ptr_t, numel_t, wd_t, chunk_t in self._tables:
+                ext().sgd_fused(ptr_t, numel_t, wd_t, chunk_t, dt, float(lr),
+                                self.momentum, float(grad_scale), self.nesterov)
         else:
             for g in self.groups:
                 for p in g["params"]:

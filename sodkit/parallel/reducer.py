@@ -62,21 +62,27 @@ class GradReducer:
         params.reverse()  # reverse registration order ≈ backward arrival order
         self._params = params
 
-        # --- bucket assignment ---
+        # --- bucket assignment (per dtype: bf16 conv weights and fp32 BN
+        # params flatten into separate buckets; arrival order within each
+        # dtype group still tracks backward order) ---
         bucket_bytes = bucket_mb * (1 << 20)
         self.buckets: list[_Bucket] = []
-        cur = _Bucket(0)
-        cur_bytes = 0
+        by_dtype: dict[torch.dtype, list[torch.nn.Parameter]] = {}
         for p in params:
-            nbytes = p.numel() * p.element_size()
-            if cur.params and cur_bytes + nbytes > bucket_bytes:
+            by_dtype.setdefault(p.dtype, []).append(p)
+        for plist in by_dtype.values():
+            cur = _Bucket(len(self.buckets))
+            cur_bytes = 0
+            for p in plist:
+                nbytes = p.numel() * p.element_size()
+                if cur.params and cur_bytes + nbytes > bucket_bytes:
+                    self.buckets.append(cur)
+                    cur = _Bucket(len(self.buckets))
+                    cur_bytes = 0
+                cur.params.append(p)
+                cur_bytes += nbytes
+            if cur.params:
                 self.buckets.append(cur)
-                cur = _Bucket(len(self.buckets))
-                cur_bytes = 0
-            cur.params.append(p)
-            cur_bytes += nbytes
-        if cur.params:
-            self.buckets.append(cur)
 
         # --- flat storage + grad views ---
         self._p2bucket: dict[torch.nn.Parameter, _Bucket] = {}

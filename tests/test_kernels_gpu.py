@@ -177,11 +177,13 @@ def test_hybrid_loss_fwd_bwd(weights):
     z0 = (torch.randn(2, 1, 40, 40, device="cuda") * 2).to(torch.bfloat16)
     t = (torch.rand(2, 1, 40, 40, device="cuda") > 0.5).float()
 
-    z = _cl(z0).requires_grad_()
+    # note: for C==1 channels_last aliases the standard layout, so detach
+    # copies keep the two graphs separate
+    z = _cl(z0).detach().clone().requires_grad_()
     loss = ops.hybrid_loss(z, t, wb, wi, ws)
     loss.backward()
 
-    zr = z0.float().requires_grad_()
+    zr = z0.detach().float().clone().requires_grad_()
     loss_r = ops.reference.hybrid_loss(zr, t, wb, wi, ws)
     loss_r.backward()
     assert abs(loss.item() - loss_r.item()) < 2e-3 + 1e-2 * abs(loss_r.item())

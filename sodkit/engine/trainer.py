@@ -52,9 +52,10 @@ class Solver:
         seed_everything(cfg.seed)
         model = build_model(cfg.model)
         if self.is_gpu:
+            # NOTE: no model.to(memory_format=channels_last) — sodkit weights
+            # are KRSC-contiguous by construction and activations are made
+            # channels_last at the input / by the kernels themselves.
             model = model.to(self.device)
-            if cfg.channels_last:
-                model = model.to(memory_format=torch.channels_last)
             if self.dtype != torch.float32:
                 convert_precision(model, self.dtype)
         if cfg.sync_bn and self.distributed:

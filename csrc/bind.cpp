@@ -10,6 +10,10 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
 torch::Tensor channel_sum(torch::Tensor x);
 
 torch::Tensor bn_stats(torch::Tensor x);
+std::vector<torch::Tensor> bn_finalize(torch::Tensor stats, double n,
+                                       double momentum, double eps,
+                                       torch::Tensor running_mean,
+                                       torch::Tensor running_var, bool training);
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd,
                        torch::Tensor gamma, torch::Tensor beta, bool relu,
                        torch::Tensor residual);
@@ -42,6 +46,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv2d_wgrad", &conv2d_wgrad);
     m.def("channel_sum", &channel_sum);
     m.def("bn_stats", &bn_stats);
+    m.def("bn_finalize", &bn_finalize);
     m.def("bn_apply", &bn_apply);
     m.def("bn_bwd_reduce", &bn_bwd_reduce);
     m.def("bn_bwd_apply", &bn_bwd_apply);

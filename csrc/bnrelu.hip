@@ -11,6 +11,53 @@
 namespace sodkit {
 
 // ---- per-channel sum + sumsq -> out[2][C] (fp32, atomics across blocks) ----
+// Vector path (C % 8 == 0): each thread owns an 8-channel chunk; the 256
+// threads cover min(C/8, 256) chunks x rows_pb rows per iteration, so a
+// whole iteration's loads are one contiguous coalesced span.  Cross-row
+// partials combine in LDS, one atomicAdd set per block.
+template <typename T>
+__global__ void bn_stats_vec_kernel(const T *__restrict__ x,
+                                    float *__restrict__ out, long M, int C) {
+    __shared__ float red[256 * 16];
+    int C8 = C >> 3;
+    int cc0 = blockIdx.y * 256;
+    int nch = min(C8 - cc0, 256);
+    int rows_pb = 256 / nch;
+    int lane_c = threadIdx.x % nch;
+    int row_off = threadIdx.x / nch;
+    float acc[16];
+#pragma unroll
+    for (int e = 0; e < 16; ++e) acc[e] = 0.f;
+    if (row_off < rows_pb) {
+        const T *base = x + (long)(cc0 + lane_c) * 8;
+        for (long m = (long)blockIdx.x * rows_pb + row_off; m < M;
+             m += (long)gridDim.x * rows_pb) {
+            s16x8 v = *reinterpret_cast<const s16x8 *>(base + m * C);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                float f = s16_as_f32<T>(v[e]);
+                acc[e] += f;
+                acc[8 + e] += f * f;
+            }
+        }
+    }
+#pragma unroll
+    for (int e = 0; e < 16; ++e) red[threadIdx.x * 16 + e] = acc[e];
+    __syncthreads();
+    if (row_off == 0) {
+        for (int r = 1; r < rows_pb; ++r)
+#pragma unroll
+            for (int e = 0; e < 16; ++e)
+                acc[e] += red[(r * nch + lane_c) * 16 + e];
+        int c0 = (cc0 + lane_c) * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            atomicAdd(&out[c0 + e], acc[e]);
+            atomicAdd(&out[C + c0 + e], acc[8 + e]);
+        }
+    }
+}
+
 template <typename T>
 __global__ void bn_stats_kernel(const T *__restrict__ x, float *__restrict__ out,
                                 long M, int C) {
@@ -24,6 +71,32 @@ __global__ void bn_stats_kernel(const T *__restrict__ x, float *__restrict__ out
     }
     atomicAdd(&out[c], s);
     atomicAdd(&out[C + c], ss);
+}
+
+// ---- stats -> (mean, invstd) + running-stat update, one tiny kernel
+// (replaces ~8 ATen [C]-sized ops per BN layer per step) ----
+__global__ void bn_finalize_kernel(const float *__restrict__ stats, float n,
+                                   float momentum, float eps,
+                                   float *__restrict__ running_mean,
+                                   float *__restrict__ running_var,
+                                   float *__restrict__ mean_out,
+                                   float *__restrict__ invstd_out, int C,
+                                   bool training) {
+    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float mean, var;
+    if (training) {
+        mean = stats[c] / n;
+        var = stats[C + c] / n - mean * mean;
+        float unbiased = var * (n / fmaxf(n - 1.f, 1.f));
+        running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+        running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    } else {
+        mean = running_mean[c];
+        var = running_var[c];
+    }
+    mean_out[c] = mean;
+    invstd_out[c] = rsqrtf(var + eps);
 }
 
 // ---- y = relu((x - mean) * invstd * gamma + beta + residual) ----
@@ -50,6 +123,61 @@ __global__ void bn_apply_kernel(const T *__restrict__ x,
 
 // ---- backward reduce: g = dy * relumask(y); out[0][c]=sum g,
 //      out[1][c]=sum g*xhat ----
+template <typename T, bool RELU>
+__global__ void bn_bwd_reduce_vec_kernel(const T *__restrict__ dy,
+                                         const T *__restrict__ y,
+                                         const T *__restrict__ x,
+                                         const float *__restrict__ mean,
+                                         const float *__restrict__ invstd,
+                                         float *__restrict__ out, long M, int C) {
+    __shared__ float red[256 * 16];
+    int C8 = C >> 3;
+    int cc0 = blockIdx.y * 256;
+    int nch = min(C8 - cc0, 256);
+    int rows_pb = 256 / nch;
+    int lane_c = threadIdx.x % nch;
+    int row_off = threadIdx.x / nch;
+    float acc[16];
+#pragma unroll
+    for (int e = 0; e < 16; ++e) acc[e] = 0.f;
+    int c0 = (cc0 + lane_c) * 8;
+    float mu[8], is[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) { mu[e] = mean[c0 + e]; is[e] = invstd[c0 + e]; }
+    if (row_off < rows_pb) {
+        long off0 = (long)c0;
+        for (long m = (long)blockIdx.x * rows_pb + row_off; m < M;
+             m += (long)gridDim.x * rows_pb) {
+            long i = m * C + off0;
+            s16x8 vdy = *reinterpret_cast<const s16x8 *>(dy + i);
+            s16x8 vy, vx = *reinterpret_cast<const s16x8 *>(x + i);
+            if (RELU) vy = *reinterpret_cast<const s16x8 *>(y + i);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                float g = s16_as_f32<T>(vdy[e]);
+                if (RELU && s16_as_f32<T>(vy[e]) <= 0.f) g = 0.f;
+                float xh = (s16_as_f32<T>(vx[e]) - mu[e]) * is[e];
+                acc[e] += g;
+                acc[8 + e] += g * xh;
+            }
+        }
+    }
+#pragma unroll
+    for (int e = 0; e < 16; ++e) red[threadIdx.x * 16 + e] = acc[e];
+    __syncthreads();
+    if (row_off == 0) {
+        for (int r = 1; r < rows_pb; ++r)
+#pragma unroll
+            for (int e = 0; e < 16; ++e)
+                acc[e] += red[(r * nch + lane_c) * 16 + e];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            atomicAdd(&out[c0 + e], acc[e]);
+            atomicAdd(&out[C + c0 + e], acc[8 + e]);
+        }
+    }
+}
+
 template <typename T, bool RELU>
 __global__ void bn_bwd_reduce_kernel(const T *__restrict__ dy,
                                      const T *__restrict__ y,
@@ -125,6 +253,14 @@ inline int ew_blocks(long total) {
 
 }  // namespace
 
+static inline dim3 vec_stats_grid(long M, int C) {
+    int C8 = C / 8;
+    int nch = std::min(C8, 256);
+    int rows_pb = 256 / nch;
+    long xblocks = std::min<long>((M + rows_pb - 1) / rows_pb, 1024);
+    return dim3((unsigned)xblocks, (C8 + 255) / 256);
+}
+
 torch::Tensor bn_stats(torch::Tensor x) {
     TORCH_CHECK(x.is_cuda() && x.dim() == 4);
     int C = x.size(1);
@@ -132,12 +268,38 @@ torch::Tensor bn_stats(torch::Tensor x) {
     auto out = torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
     DISPATCH_T(x, "bn_stats", [&] {
-        hipLaunchKernelGGL((sodkit::bn_stats_kernel<T>), stats_grid(M, C),
-                           dim3(256), 0, stream.stream(),
-                           reinterpret_cast<const T *>(x.data_ptr()),
-                           out.data_ptr<float>(), M, C);
+        if (C % 8 == 0 && sizeof(T) == 2)
+            hipLaunchKernelGGL((sodkit::bn_stats_vec_kernel<T>),
+                               vec_stats_grid(M, C), dim3(256), 0,
+                               stream.stream(),
+                               reinterpret_cast<const T *>(x.data_ptr()),
+                               out.data_ptr<float>(), M, C);
+        else
+            hipLaunchKernelGGL((sodkit::bn_stats_kernel<T>), stats_grid(M, C),
+                               dim3(256), 0, stream.stream(),
+                               reinterpret_cast<const T *>(x.data_ptr()),
+                               out.data_ptr<float>(), M, C);
     });
     return out;
+}
+
+std::vector<torch::Tensor> bn_finalize(torch::Tensor stats, double n,
+                                       double momentum, double eps,
+                                       torch::Tensor running_mean,
+                                       torch::Tensor running_var,
+                                       bool training) {
+    int C = running_mean.size(0);
+    auto mean = torch::empty({C}, running_mean.options());
+    auto invstd = torch::empty({C}, running_mean.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(sodkit::bn_finalize_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream.stream(),
+                       training ? stats.data_ptr<float>() : nullptr, (float)n,
+                       (float)momentum, (float)eps,
+                       running_mean.data_ptr<float>(),
+                       running_var.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), C, training);
+    return {mean, invstd};
 }
 
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd,
@@ -180,13 +342,24 @@ torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
     auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
     DISPATCH_T(x, "bn_bwd_reduce", [&] {
         auto launch = [&](auto rl) {
-            hipLaunchKernelGGL((sodkit::bn_bwd_reduce_kernel<T, decltype(rl)::value>),
-                               stats_grid(M, C), dim3(256), 0, stream.stream(),
-                               reinterpret_cast<const T *>(dy.data_ptr()),
-                               reinterpret_cast<const T *>(y.data_ptr()),
-                               reinterpret_cast<const T *>(x.data_ptr()),
-                               meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
-                               out.data_ptr<float>(), M, C);
+            if (C % 8 == 0 && sizeof(T) == 2)
+                hipLaunchKernelGGL(
+                    (sodkit::bn_bwd_reduce_vec_kernel<T, decltype(rl)::value>),
+                    vec_stats_grid(M, C), dim3(256), 0, stream.stream(),
+                    reinterpret_cast<const T *>(dy.data_ptr()),
+                    reinterpret_cast<const T *>(y.data_ptr()),
+                    reinterpret_cast<const T *>(x.data_ptr()),
+                    meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                    out.data_ptr<float>(), M, C);
+            else
+                hipLaunchKernelGGL(
+                    (sodkit::bn_bwd_reduce_kernel<T, decltype(rl)::value>),
+                    stats_grid(M, C), dim3(256), 0, stream.stream(),
+                    reinterpret_cast<const T *>(dy.data_ptr()),
+                    reinterpret_cast<const T *>(y.data_ptr()),
+                    reinterpret_cast<const T *>(x.data_ptr()),
+                    meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                    out.data_ptr<float>(), M, C);
         };
         if (relu) launch(std::true_type{}); else launch(std::false_type{});
     });

@@ -54,6 +54,14 @@ template <> struct MfmaOp<__half> {
     }
 };
 
+// Reinterpret one 16-bit lane of an s16 vector as T and convert to f32.
+template <typename T>
+SOD_DEVINL float s16_as_f32(short bits) {
+    T t;
+    *reinterpret_cast<short *>(&t) = bits;
+    return ElemTraits<T>::to_f32(t);
+}
+
 // Pack 4 f32 accumulator values into 4 consecutive 16-bit elems.
 template <typename T>
 SOD_DEVINL s16x4 pack4(const f32x4 &v) {

@@ -268,14 +268,19 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     const int m_end = min(m_begin + m_per_split, M_total);
 
     const bool dyvec_ok = (Cout % 8 == 0);
+    const bool xvec_ok = (C % 8 == 0);
 
-    // per-thread x-chunk geometry: chunk ch -> (j = ch % BJ, mv8 = ch / BJ)
+    // per-thread x-chunk geometry: chunk ch -> (jv8 = ch % (BJ/8),
+    // m_local = ch / (BJ/8)); 8 consecutive j = 8 consecutive cin of one
+    // (r,s) tap when C%8==0 -> one 16B gather + 8 LDS transpose writes.
     constexpr int XCH = BJ * BKM / 8 / 256;
-    int xr[XCH], xs[XCH], xc[XCH], xj[XCH];
+    int xr[XCH], xs[XCH], xc[XCH], xj[XCH], xml[XCH];
 #pragma unroll
     for (int i = 0; i < XCH; ++i) {
         int ch = t + i * 256;
-        int j = tile_j + (ch % BJ);
+        int jv8 = ch % (BJ / 8);
+        xml[i] = ch / (BJ / 8);
+        int j = tile_j + jv8 * 8;
         xj[i] = j;
         int jj = j < Kw ? j : 0;
         int rs = jj / C;
@@ -318,29 +323,40 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                 *reinterpret_cast<short *>(dyT_lds + swz<BKM>(row, m_local * 2)) = v[e];
             }
         }
-        // ---- stage x^T tile: 8 m-contiguous gathers per chunk ----
+        // ---- stage x^T tile: one 16B cin-gather per chunk, transposed
+        //      into [j][m] via 8 LDS u16 writes ----
 #pragma unroll
         for (int i = 0; i < XCH; ++i) {
-            int ch = t + i * 256;
-            int j_local = ch % BJ;
-            int mv8 = ch / BJ;
+            int m = m0 + xml[i];
+            int j_local0 = (xj[i] - tile_j);
             s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (xj[i] < Kw) {
+            if (xj[i] < Kw && m < m_end) {
+                int n = m / (P * Q);
+                int pq = m % (P * Q);
+                int p = pq / Q, q = pq % Q;
+                if (xvec_ok) {
+                    long off = gather_off<0>(n, p, q, xr[i], xs[i], xc[i],
+                                             H, W, C, stride, pad, dil);
+                    if (off >= 0)
+                        v = *reinterpret_cast<const s16x8 *>(x + off);
+                } else {
 #pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    int m = m0 + mv8 * 8 + e;
-                    if (m < m_end) {
-                        int n = m / (P * Q);
-                        int pq = m % (P * Q);
-                        int p = pq / Q, q = pq % Q;
-                        long off = gather_off<0>(n, p, q, xr[i], xs[i], xc[i],
-                                                 H, W, C, stride, pad, dil);
-                        if (off >= 0)
-                            v[e] = reinterpret_cast<const short *>(x)[off];
+                    for (int e = 0; e < 8; ++e) {
+                        int j = xj[i] + e;
+                        if (j < Kw) {
+                            int rs = j / C, c = j % C;
+                            long off = gather_off<0>(n, p, q, rs / S, rs % S, c,
+                                                     H, W, C, stride, pad, dil);
+                            if (off >= 0)
+                                v[e] = reinterpret_cast<const short *>(x)[off];
+                        }
                     }
                 }
             }
-            *reinterpret_cast<s16x8 *>(xT_lds + swz<BKM>(j_local, mv8 * 16)) = v;
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                *reinterpret_cast<short *>(
+                    xT_lds + swz<BKM>(j_local0 + e, xml[i] * 2)) = v[e];
         }
         __syncthreads();
 
@@ -539,7 +555,9 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
         int tiles_c = (Cout + BC - 1) / BC;
         int tiles_j = (Kw + BJ - 1) / BJ;
         int tiles = tiles_c * tiles_j;
-        int splitk = std::max(1, std::min(64, 768 / std::max(tiles, 1)));
+        // fill the chip: ~2 blocks per CU (256 CUs) regardless of tile count
+        int splitk = std::max(1, std::min(512, 512 / std::max(tiles, 1)));
+        splitk = (int)std::min<long>(splitk, (M + 63) / 64);
         long m_per_split = ((M + splitk - 1) / splitk + 63) / 64 * 64;
         splitk = (int)((M + m_per_split - 1) / m_per_split);
         auto partial = torch::empty({splitk, (long)Cout * Kw},

@@ -40,11 +40,14 @@ class AIM(nn.Module):
 
     def forward(self, low: Optional[torch.Tensor], mid: torch.Tensor,
                 high: Optional[torch.Tensor]) -> torch.Tensor:
+        # every join is fused into the producing op (no standalone adds):
+        # the neighbor branches land in the BN residual epilogue / the
+        # bilinear fused-add.
         y = self.conv_m(mid)
         if self.conv_l is not None and low is not None:
-            y = y + self.conv_l(low)
+            y = self.conv_l(low, residual=y)
         if self.conv_h is not None and high is not None:
-            y = y + bilinear2d(self.conv_h(high), mid.shape[2], mid.shape[3])
+            y = bilinear2d(self.conv_h(high), mid.shape[2], mid.shape[3], add=y)
         # fused residual join: ReLU(BN(conv(y)) + mid)
         return self.fuse_bn(self.fuse_conv(y), residual=mid, relu=True)
 

@@ -43,22 +43,16 @@ class _BNActFn(torch.autograd.Function):
             n_total = float(n_local)
             if process_group is not None and dist.is_initialized() \
                     and dist.get_world_size(process_group) > 1:
-                packed = torch.cat([stats.view(-1),
-                                    stats.new_tensor([n_total])])
-                dist.all_reduce(packed, group=process_group)
-                stats = packed[:2 * C].view(2, C)
-                n_total = packed[2 * C].item()
-            mean = stats[0] / n_total
-            var = stats[1] / n_total - mean * mean
-            with torch.no_grad():
-                unbiased = var * (n_total / max(n_total - 1.0, 1.0))
-                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
-                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+                dist.all_reduce(stats, group=process_group)
+                # equal per-rank shapes by construction (DistributedSampler
+                # with drop_last) -> no count exchange, no host sync
+                n_total = float(n_local) * dist.get_world_size(process_group)
+            mean, invstd = ext().bn_finalize(stats, n_total, momentum, eps,
+                                             running_mean, running_var, True)
         else:
-            mean = running_mean
-            var = running_var
             n_total = 0.0                                   # eval marker
-        invstd = torch.rsqrt(var + eps)
+            mean, invstd = ext().bn_finalize(running_mean, 0.0, momentum, eps,
+                                             running_mean, running_var, False)
         y = ext().bn_apply(x, mean, invstd, gamma, beta, relu,
                            residual if residual is not None else torch.Tensor())
         ctx.save_for_backward(x, y, mean, invstd, gamma)

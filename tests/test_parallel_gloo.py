@@ -5,6 +5,7 @@ DDP-correctness oracle (2-rank global-batch-B == 1-rank batch-B)."""
 
 import os
 import socket
+import tempfile
 
 import pytest
 import torch
@@ -24,12 +25,19 @@ def _free_port() -> int:
     return port
 
 
-def _init(rank: int, world: int, port: int):
-    os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(port)
+def _init(rank: int, world: int, store_file: str):
+    # FileStore rendezvous: no TCP port races between tests
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
-    dist.init_process_group("gloo", rank=rank, world_size=world)
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"file://{store_file}")
+
+
+def _store_file() -> str:
+    fd, path = tempfile.mkstemp(prefix="sodkit_pg_")
+    os.close(fd)
+    os.unlink(path)
+    return path
 
 
 def _reducer_worker(rank, world, port, q):
@@ -122,7 +130,7 @@ def _minet_worker(rank, world, port, q):
 
 @pytest.mark.timeout(300)
 def test_reducer_allreduces_grads():
-    port = _free_port()
+    port = _store_file()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     ps = [ctx.Process(target=_reducer_worker, args=(r, 2, port, q)) for r in range(2)]
@@ -139,7 +147,7 @@ def test_reducer_allreduces_grads():
 
 @pytest.mark.timeout(600)
 def test_ddp_oracle_matches_single_process():
-    port = _free_port()
+    port = _store_file()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     ps = [ctx.Process(target=_oracle_worker, args=(r, 2, port, q)) for r in range(2)]
@@ -178,7 +186,7 @@ def test_ddp_oracle_matches_single_process():
 
 @pytest.mark.timeout(600)
 def test_minet_ranks_stay_identical():
-    port = _free_port()
+    port = _store_file()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     ps = [ctx.Process(target=_minet_worker, args=(r, 2, port, q)) for r in range(2)]

@@ -30,8 +30,24 @@ __global__ void bn_stats_vec_kernel(const T *__restrict__ x,
     for (int e = 0; e < 16; ++e) acc[e] = 0.f;
     if (row_off < rows_pb) {
         const T *base = x + (long)(cc0 + lane_c) * 8;
-        for (long m = (long)blockIdx.x * rows_pb + row_off; m < M;
-             m += (long)gridDim.x * rows_pb) {
+        long mstep = (long)gridDim.x * rows_pb;
+        long m = (long)blockIdx.x * rows_pb + row_off;
+        // 4-deep unroll: 4 independent loads in flight per wave (the
+        // dependent single-load chain leaves HBM ~8x under-subscribed)
+        for (; m + 3 * mstep < M; m += 4 * mstep) {
+            s16x8 v0 = *reinterpret_cast<const s16x8 *>(base + m * C);
+            s16x8 v1 = *reinterpret_cast<const s16x8 *>(base + (m + mstep) * C);
+            s16x8 v2 = *reinterpret_cast<const s16x8 *>(base + (m + 2 * mstep) * C);
+            s16x8 v3 = *reinterpret_cast<const s16x8 *>(base + (m + 3 * mstep) * C);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                float f0 = s16_as_f32<T>(v0[e]), f1 = s16_as_f32<T>(v1[e]);
+                float f2 = s16_as_f32<T>(v2[e]), f3 = s16_as_f32<T>(v3[e]);
+                acc[e] += (f0 + f1) + (f2 + f3);
+                acc[8 + e] += (f0 * f0 + f1 * f1) + (f2 * f2 + f3 * f3);
+            }
+        }
+        for (; m < M; m += mstep) {
             s16x8 v = *reinterpret_cast<const s16x8 *>(base + m * C);
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
@@ -146,8 +162,31 @@ __global__ void bn_bwd_reduce_vec_kernel(const T *__restrict__ dy,
     for (int e = 0; e < 8; ++e) { mu[e] = mean[c0 + e]; is[e] = invstd[c0 + e]; }
     if (row_off < rows_pb) {
         long off0 = (long)c0;
-        for (long m = (long)blockIdx.x * rows_pb + row_off; m < M;
-             m += (long)gridDim.x * rows_pb) {
+        long mstep = (long)gridDim.x * rows_pb;
+        long m = (long)blockIdx.x * rows_pb + row_off;
+        for (; m + mstep < M; m += 2 * mstep) {
+            long i0 = m * C + off0, i1 = (m + mstep) * C + off0;
+            s16x8 vdy0 = *reinterpret_cast<const s16x8 *>(dy + i0);
+            s16x8 vdy1 = *reinterpret_cast<const s16x8 *>(dy + i1);
+            s16x8 vx0 = *reinterpret_cast<const s16x8 *>(x + i0);
+            s16x8 vx1 = *reinterpret_cast<const s16x8 *>(x + i1);
+            s16x8 vy0, vy1;
+            if (RELU) {
+                vy0 = *reinterpret_cast<const s16x8 *>(y + i0);
+                vy1 = *reinterpret_cast<const s16x8 *>(y + i1);
+            }
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                float g0 = s16_as_f32<T>(vdy0[e]), g1 = s16_as_f32<T>(vdy1[e]);
+                if (RELU && s16_as_f32<T>(vy0[e]) <= 0.f) g0 = 0.f;
+                if (RELU && s16_as_f32<T>(vy1[e]) <= 0.f) g1 = 0.f;
+                float xh0 = (s16_as_f32<T>(vx0[e]) - mu[e]) * is[e];
+                float xh1 = (s16_as_f32<T>(vx1[e]) - mu[e]) * is[e];
+                acc[e] += g0 + g1;
+                acc[8 + e] += g0 * xh0 + g1 * xh1;
+            }
+        }
+        for (; m < M; m += mstep) {
             long i = m * C + off0;
             s16x8 vdy = *reinterpret_cast<const s16x8 *>(dy + i);
             s16x8 vy, vx = *reinterpret_cast<const s16x8 *>(x + i);
@@ -257,7 +296,7 @@ static inline dim3 vec_stats_grid(long M, int C) {
     int C8 = C / 8;
     int nch = std::min(C8, 256);
     int rows_pb = 256 / nch;
-    long xblocks = std::min<long>((M + rows_pb - 1) / rows_pb, 1024);
+    long xblocks = std::min<long>((M + rows_pb - 1) / rows_pb, 2048);
     return dim3((unsigned)xblocks, (C8 + 255) / 256);
 }
 

@@ -66,12 +66,25 @@ __global__ void bn_stats_vec_kernel(const T *__restrict__ x,
             for (int e = 0; e < 16; ++e)
                 acc[e] += red[(r * nch + lane_c) * 16 + e];
         int c0 = (cc0 + lane_c) * 8;
+        // per-block partial [2C] slab: plain stores; a second kernel reduces
+        // (same-address atomicAdd serializes ~linearly in block count)
+        float *slab = out + (long)blockIdx.x * 2 * C;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-            atomicAdd(&out[c0 + e], acc[e]);
-            atomicAdd(&out[C + c0 + e], acc[8 + e]);
+            slab[c0 + e] = acc[e];
+            slab[C + c0 + e] = acc[8 + e];
         }
     }
+}
+
+// partial[nb][R] -> out[R] (deterministic fixed-order sum)
+__global__ void slab_reduce_kernel(const float *__restrict__ partial,
+                                   float *__restrict__ out, int nb, long R) {
+    long r = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (r >= R) return;
+    float s = 0.f;
+    for (int b = 0; b < nb; ++b) s += partial[(long)b * R + r];
+    out[r] = s;
 }
 
 template <typename T>
@@ -209,10 +222,11 @@ __global__ void bn_bwd_reduce_vec_kernel(const T *__restrict__ dy,
 #pragma unroll
             for (int e = 0; e < 16; ++e)
                 acc[e] += red[(r * nch + lane_c) * 16 + e];
+        float *slab = out + (long)blockIdx.x * 2 * C;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-            atomicAdd(&out[c0 + e], acc[e]);
-            atomicAdd(&out[C + c0 + e], acc[8 + e]);
+            slab[c0 + e] = acc[e];
+            slab[C + c0 + e] = acc[8 + e];
         }
     }
 }
@@ -296,7 +310,7 @@ static inline dim3 vec_stats_grid(long M, int C) {
     int C8 = C / 8;
     int nch = std::min(C8, 256);
     int rows_pb = 256 / nch;
-    long xblocks = std::min<long>((M + rows_pb - 1) / rows_pb, 2048);
+    long xblocks = std::min<long>((M + rows_pb - 1) / rows_pb, 512);
     return dim3((unsigned)xblocks, (C8 + 255) / 256);
 }
 
@@ -307,13 +321,22 @@ torch::Tensor bn_stats(torch::Tensor x) {
     auto out = torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
     DISPATCH_T(x, "bn_stats", [&] {
-        if (C % 8 == 0 && sizeof(T) == 2)
-            hipLaunchKernelGGL((sodkit::bn_stats_vec_kernel<T>),
-                               vec_stats_grid(M, C), dim3(256), 0,
-                               stream.stream(),
+        if (C % 8 == 0 && sizeof(T) == 2) {
+            dim3 g = vec_stats_grid(M, C);
+            // y-chunks (C8 > 256) share x-indexed slabs: disjoint channel
+            // ranges, no collision
+            int nb = g.x;
+            auto partial = torch::empty({nb, 2L * C},
+                                        x.options().dtype(torch::kFloat32));
+            hipLaunchKernelGGL((sodkit::bn_stats_vec_kernel<T>), g, dim3(256),
+                               0, stream.stream(),
                                reinterpret_cast<const T *>(x.data_ptr()),
-                               out.data_ptr<float>(), M, C);
-        else
+                               partial.data_ptr<float>(), M, C);
+            hipLaunchKernelGGL(sodkit::slab_reduce_kernel,
+                               dim3((2 * C + 255) / 256), dim3(256), 0,
+                               stream.stream(), partial.data_ptr<float>(),
+                               out.data_ptr<float>(), nb, 2L * C);
+        } else
             hipLaunchKernelGGL((sodkit::bn_stats_kernel<T>), stats_grid(M, C),
                                dim3(256), 0, stream.stream(),
                                reinterpret_cast<const T *>(x.data_ptr()),
@@ -381,16 +404,24 @@ torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
     auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
     DISPATCH_T(x, "bn_bwd_reduce", [&] {
         auto launch = [&](auto rl) {
-            if (C % 8 == 0 && sizeof(T) == 2)
+            if (C % 8 == 0 && sizeof(T) == 2) {
+                dim3 g = vec_stats_grid(M, C);
+                int nb = g.x;
+                auto partial = torch::empty({nb, 2L * C},
+                                            x.options().dtype(torch::kFloat32));
                 hipLaunchKernelGGL(
                     (sodkit::bn_bwd_reduce_vec_kernel<T, decltype(rl)::value>),
-                    vec_stats_grid(M, C), dim3(256), 0, stream.stream(),
+                    g, dim3(256), 0, stream.stream(),
                     reinterpret_cast<const T *>(dy.data_ptr()),
                     reinterpret_cast<const T *>(y.data_ptr()),
                     reinterpret_cast<const T *>(x.data_ptr()),
                     meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
-                    out.data_ptr<float>(), M, C);
-            else
+                    partial.data_ptr<float>(), M, C);
+                hipLaunchKernelGGL(sodkit::slab_reduce_kernel,
+                                   dim3((2 * C + 255) / 256), dim3(256), 0,
+                                   stream.stream(), partial.data_ptr<float>(),
+                                   out.data_ptr<float>(), nb, 2L * C);
+            } else
                 hipLaunchKernelGGL(
                     (sodkit::bn_bwd_reduce_kernel<T, decltype(rl)::value>),
                     stats_grid(M, C), dim3(256), 0, stream.stream(),

@@ -77,14 +77,19 @@ __global__ void bn_stats_vec_kernel(const T *__restrict__ x,
     }
 }
 
-// partial[nb][R] -> out[R] (deterministic fixed-order sum)
+// partial[nb][R] -> out[R] (deterministic fixed-order sum; 8 independent
+// accumulators keep 8 loads in flight — a serial chain is latency-bound)
 __global__ void slab_reduce_kernel(const float *__restrict__ partial,
                                    float *__restrict__ out, int nb, long R) {
     long r = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (r >= R) return;
-    float s = 0.f;
-    for (int b = 0; b < nb; ++b) s += partial[(long)b * R + r];
-    out[r] = s;
+    float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    int b = 0;
+    for (; b + 8 <= nb; b += 8)
+#pragma unroll
+        for (int u = 0; u < 8; ++u) a[u] += partial[(long)(b + u) * R + r];
+    for (; b < nb; ++b) a[0] += partial[(long)b * R + r];
+    out[r] = ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
 }
 
 template <typename T>
@@ -318,7 +323,10 @@ torch::Tensor bn_stats(torch::Tensor x) {
     TORCH_CHECK(x.is_cuda() && x.dim() == 4);
     int C = x.size(1);
     long M = x.numel() / C;
-    auto out = torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
+    bool vec = (C % 8 == 0) && x.element_size() == 2;
+    // the vec path's slab_reduce overwrites out fully -> empty, not zeros
+    auto out = vec ? torch::empty({2, C}, x.options().dtype(torch::kFloat32))
+                   : torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
     DISPATCH_T(x, "bn_stats", [&] {
         if (C % 8 == 0 && sizeof(T) == 2) {
@@ -399,7 +407,9 @@ torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
                             torch::Tensor mean, torch::Tensor invstd, bool relu) {
     int C = x.size(1);
     long M = x.numel() / C;
-    auto out = torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
+    bool vec = (C % 8 == 0) && x.element_size() == 2;
+    auto out = vec ? torch::empty({2, C}, x.options().dtype(torch::kFloat32))
+                   : torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
     auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
     DISPATCH_T(x, "bn_bwd_reduce", [&] {

@@ -407,9 +407,16 @@ __global__ void splitk_reduce_kernel(const float *__restrict__ partial,
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
     for (; i < n_elems; i += stride) {
-        float s = 0.f;
-        for (int z = 0; z < splitk; ++z) s += partial[(long)z * n_elems + i];
-        dw[i] = ElemTraits<T>::from_f32(s);
+        // 8 independent accumulators: keep loads in flight over large splitk
+        float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+        int z = 0;
+        for (; z + 8 <= splitk; z += 8)
+#pragma unroll
+            for (int u = 0; u < 8; ++u)
+                a[u] += partial[(long)(z + u) * n_elems + i];
+        for (; z < splitk; ++z) a[0] += partial[(long)z * n_elems + i];
+        dw[i] = ElemTraits<T>::from_f32(((a[0] + a[1]) + (a[2] + a[3]))
+                                        + ((a[4] + a[5]) + (a[6] + a[7])));
     }
 }
 

@@ -133,7 +133,37 @@ __global__ void bn_finalize_kernel(const float *__restrict__ stats, float n,
     invstd_out[c] = rsqrtf(var + eps);
 }
 
-// ---- y = relu((x - mean) * invstd * gamma + beta + residual) ----
+// ---- vectorized apply: 8 channels per thread (C % 8 == 0) ----
+template <typename T, bool RELU, bool RES>
+__global__ void bn_apply_vec_kernel(const T *__restrict__ x,
+                                    const T *__restrict__ res,
+                                    const float *__restrict__ mean,
+                                    const float *__restrict__ invstd,
+                                    const float *__restrict__ gamma,
+                                    const float *__restrict__ beta,
+                                    T *__restrict__ y, long total8, int C8) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < total8; i += stride) {
+        int c0 = (int)(i % C8) * 8;
+        s16x8 vx = *reinterpret_cast<const s16x8 *>(x + i * 8);
+        s16x8 vr;
+        if (RES) vr = *reinterpret_cast<const s16x8 *>(res + i * 8);
+        s16x8 vo;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            float sc = invstd[c0 + e] * gamma[c0 + e];
+            float sh = beta[c0 + e] - mean[c0 + e] * sc;
+            float v = s16_as_f32<T>(vx[e]) * sc + sh;
+            if (RES) v += s16_as_f32<T>(vr[e]);
+            if (RELU) v = fmaxf(v, 0.f);
+            T tv = ElemTraits<T>::from_f32(v);
+            vo[e] = *reinterpret_cast<const short *>(&tv);
+        }
+        *reinterpret_cast<s16x8 *>(y + i * 8) = vo;
+    }
+}
+
 template <typename T, bool RELU, bool RES>
 __global__ void bn_apply_kernel(const T *__restrict__ x,
                                 const T *__restrict__ res,
@@ -257,6 +287,48 @@ __global__ void bn_bwd_reduce_kernel(const T *__restrict__ dy,
     }
     atomicAdd(&out[c], sg);
     atomicAdd(&out[C + c], sgx);
+}
+
+// ---- vectorized backward apply (C % 8 == 0) ----
+template <typename T, bool RELU, bool NEED_DRES>
+__global__ void bn_bwd_apply_vec_kernel(const T *__restrict__ dy,
+                                        const T *__restrict__ y,
+                                        const T *__restrict__ x,
+                                        const float *__restrict__ mean,
+                                        const float *__restrict__ invstd,
+                                        const float *__restrict__ gamma,
+                                        const float *__restrict__ sums,
+                                        float inv_n, T *__restrict__ dx,
+                                        T *__restrict__ dres, long total8,
+                                        int C8) {
+    int C = C8 * 8;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < total8; i += stride) {
+        int c0 = (int)(i % C8) * 8;
+        s16x8 vdy = *reinterpret_cast<const s16x8 *>(dy + i * 8);
+        s16x8 vx = *reinterpret_cast<const s16x8 *>(x + i * 8);
+        s16x8 vy;
+        if (RELU) vy = *reinterpret_cast<const s16x8 *>(y + i * 8);
+        s16x8 vo, vdr;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            int c = c0 + e;
+            float g = s16_as_f32<T>(vdy[e]);
+            if (RELU && s16_as_f32<T>(vy[e]) <= 0.f) g = 0.f;
+            if (NEED_DRES) {
+                T tg = ElemTraits<T>::from_f32(g);
+                vdr[e] = *reinterpret_cast<const short *>(&tg);
+            }
+            float is = invstd[c];
+            float xh = (s16_as_f32<T>(vx[e]) - mean[c]) * is;
+            float v = g - (sums[c] + xh * sums[C + c]) * inv_n;
+            T tv = ElemTraits<T>::from_f32(gamma[c] * is * v);
+            vo[e] = *reinterpret_cast<const short *>(&tv);
+        }
+        *reinterpret_cast<s16x8 *>(dx + i * 8) = vo;
+        if (NEED_DRES) *reinterpret_cast<s16x8 *>(dres + i * 8) = vdr;
+    }
 }
 
 // ---- backward apply:
@@ -388,13 +460,23 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd
         const T *resp = has_res ? reinterpret_cast<const T *>(residual.data_ptr())
                                 : nullptr;
         auto launch = [&](auto rl, auto rs) {
-            hipLaunchKernelGGL(
-                (sodkit::bn_apply_kernel<T, decltype(rl)::value, decltype(rs)::value>),
-                dim3(ew_blocks(total)), dim3(256), 0, stream.stream(),
-                reinterpret_cast<const T *>(x.data_ptr()), resp,
-                meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
-                gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
-                reinterpret_cast<T *>(y.data_ptr()), total, C);
+            if (C % 8 == 0 && sizeof(T) == 2) {
+                long total8 = total / 8;
+                hipLaunchKernelGGL(
+                    (sodkit::bn_apply_vec_kernel<T, decltype(rl)::value, decltype(rs)::value>),
+                    dim3(ew_blocks(total8)), dim3(256), 0, stream.stream(),
+                    reinterpret_cast<const T *>(x.data_ptr()), resp,
+                    meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                    gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
+                    reinterpret_cast<T *>(y.data_ptr()), total8, C / 8);
+            } else
+                hipLaunchKernelGGL(
+                    (sodkit::bn_apply_kernel<T, decltype(rl)::value, decltype(rs)::value>),
+                    dim3(ew_blocks(total)), dim3(256), 0, stream.stream(),
+                    reinterpret_cast<const T *>(x.data_ptr()), resp,
+                    meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                    gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
+                    reinterpret_cast<T *>(y.data_ptr()), total, C);
         };
         using tt = std::true_type; using ff = std::false_type;
         if (relu) { if (has_res) launch(tt{}, tt{}); else launch(tt{}, ff{}); }
@@ -463,17 +545,31 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor dy, torch::Tensor y,
     auto sumsf = sums.contiguous();
     DISPATCH_T(x, "bn_bwd_apply", [&] {
         auto launch = [&](auto rl, auto dr) {
-            hipLaunchKernelGGL(
-                (sodkit::bn_bwd_apply_kernel<T, decltype(rl)::value, decltype(dr)::value>),
-                dim3(ew_blocks(total)), dim3(256), 0, stream.stream(),
-                reinterpret_cast<const T *>(dy.data_ptr()),
-                reinterpret_cast<const T *>(y.data_ptr()),
-                reinterpret_cast<const T *>(x.data_ptr()),
-                meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
-                gammaf.data_ptr<float>(), sumsf.data_ptr<float>(), inv_n,
-                reinterpret_cast<T *>(dx.data_ptr()),
-                need_dres ? reinterpret_cast<T *>(dres.data_ptr()) : nullptr,
-                total, C);
+            if (C % 8 == 0 && sizeof(T) == 2) {
+                long total8 = total / 8;
+                hipLaunchKernelGGL(
+                    (sodkit::bn_bwd_apply_vec_kernel<T, decltype(rl)::value, decltype(dr)::value>),
+                    dim3(ew_blocks(total8)), dim3(256), 0, stream.stream(),
+                    reinterpret_cast<const T *>(dy.data_ptr()),
+                    reinterpret_cast<const T *>(y.data_ptr()),
+                    reinterpret_cast<const T *>(x.data_ptr()),
+                    meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                    gammaf.data_ptr<float>(), sumsf.data_ptr<float>(), inv_n,
+                    reinterpret_cast<T *>(dx.data_ptr()),
+                    need_dres ? reinterpret_cast<T *>(dres.data_ptr()) : nullptr,
+                    total8, C / 8);
+            } else
+                hipLaunchKernelGGL(
+                    (sodkit::bn_bwd_apply_kernel<T, decltype(rl)::value, decltype(dr)::value>),
+                    dim3(ew_blocks(total)), dim3(256), 0, stream.stream(),
+                    reinterpret_cast<const T *>(dy.data_ptr()),
+                    reinterpret_cast<const T *>(y.data_ptr()),
+                    reinterpret_cast<const T *>(x.data_ptr()),
+                    meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
+                    gammaf.data_ptr<float>(), sumsf.data_ptr<float>(), inv_n,
+                    reinterpret_cast<T *>(dx.data_ptr()),
+                    need_dres ? reinterpret_cast<T *>(dres.data_ptr()) : nullptr,
+                    total, C);
         };
         using tt = std::true_type; using ff = std::false_type;
         if (relu) { if (need_dres) launch(tt{}, tt{}); else launch(tt{}, ff{}); }

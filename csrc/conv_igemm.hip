@@ -271,15 +271,17 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     const bool xvec_ok = (C % 8 == 0);
 
     // per-thread x-chunk geometry: chunk ch -> (jv8 = ch % (BJ/8),
-    // m_local = ch / (BJ/8)); 8 consecutive j = 8 consecutive cin of one
-    // (r,s) tap when C%8==0 -> one 16B gather + 8 LDS transpose writes.
-    constexpr int XCH = BJ * BKM / 8 / 256;
+    // m_quad = ch / (BJ/8)); 8 consecutive j = 8 consecutive cin of one
+    // (r,s) tap when C%8==0 -> four 16B gathers (m..m+3), transposed in
+    // registers into 8 ds_write_b64 (the scattered-u16 transpose was
+    // write-instruction-bound).
+    constexpr int XCH = BJ * (BKM / 4) / 8 / 256;
     int xr[XCH], xs[XCH], xc[XCH], xj[XCH], xml[XCH];
 #pragma unroll
     for (int i = 0; i < XCH; ++i) {
         int ch = t + i * 256;
         int jv8 = ch % (BJ / 8);
-        xml[i] = ch / (BJ / 8);
+        xml[i] = (ch / (BJ / 8)) * 4;          // first m of the quad
         int j = tile_j + jv8 * 8;
         xj[i] = j;
         int jj = j < Kw ? j : 0;
@@ -288,7 +290,6 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
         xr[i] = rs / S;
         xs[i] = rs % S;
     }
-    constexpr int DCH = BCOUT * BKM / 8 / 256;
 
     f32x4 acc[FC][FJ];
 #pragma unroll
@@ -296,67 +297,79 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
 #pragma unroll
         for (int b = 0; b < FJ; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
+    constexpr int DNCH = BCOUT * (BKM / 4) / 8;   // dy chunks (4 m x 8 cout)
     for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
-        // ---- stage dy^T tile: read dy[m][cout..+7], write transposed ----
-#pragma unroll
-        for (int i = 0; i < DCH; ++i) {
-            int ch = t + i * 256;
-            int m_local = ch / (BCOUT / 8);
+        // ---- stage dy^T tile: 4 coalesced 16B reads dy[m..m+3][cout..+7],
+        //      register transpose, 8 ds_write_b64 ----
+        for (int ch = t; ch < DNCH; ch += 256) {
+            int m_local = (ch / (BCOUT / 8)) * 4;
             int cv = (ch % (BCOUT / 8)) * 8;
-            int m = m0 + m_local;
             int cout = tile_c + cv;
-            s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (m < m_end && cout < Cout) {
-                const T *src = dy + (long)m * Cout + cout;
-                if (dyvec_ok && cout + 7 < Cout) {
-                    v = *reinterpret_cast<const s16x8 *>(src);
-                } else {
+            s16x8 v[4];
 #pragma unroll
-                    for (int e = 0; e < 8; ++e)
-                        if (cout + e < Cout)
-                            v[e] = reinterpret_cast<const short *>(src)[e];
+            for (int u = 0; u < 4; ++u) {
+                v[u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                int m = m0 + m_local + u;
+                if (m < m_end && cout < Cout) {
+                    const T *src = dy + (long)m * Cout + cout;
+                    if (dyvec_ok && cout + 7 < Cout) {
+                        v[u] = *reinterpret_cast<const s16x8 *>(src);
+                    } else {
+#pragma unroll
+                        for (int e = 0; e < 8; ++e)
+                            if (cout + e < Cout)
+                                v[u][e] = reinterpret_cast<const short *>(src)[e];
+                    }
                 }
             }
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
-                int row = cv + e;
-                *reinterpret_cast<short *>(dyT_lds + swz<BKM>(row, m_local * 2)) = v[e];
+                s16x4 w4 = {v[0][e], v[1][e], v[2][e], v[3][e]};
+                *reinterpret_cast<s16x4 *>(
+                    dyT_lds + swz<BKM>(cv + e, m_local * 2)) = w4;
             }
         }
-        // ---- stage x^T tile: one 16B cin-gather per chunk, transposed
-        //      into [j][m] via 8 LDS u16 writes ----
+        // ---- stage x^T tile: four 16B cin-gathers (m..m+3) per chunk,
+        //      register transpose, 8 ds_write_b64 ----
 #pragma unroll
         for (int i = 0; i < XCH; ++i) {
-            int m = m0 + xml[i];
             int j_local0 = (xj[i] - tile_j);
-            s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (xj[i] < Kw && m < m_end) {
-                int n = m / (P * Q);
-                int pq = m % (P * Q);
-                int p = pq / Q, q = pq % Q;
-                if (xvec_ok) {
-                    long off = gather_off<0>(n, p, q, xr[i], xs[i], xc[i],
-                                             H, W, C, stride, pad, dil);
-                    if (off >= 0)
-                        v = *reinterpret_cast<const s16x8 *>(x + off);
-                } else {
+            s16x8 v[4];
 #pragma unroll
-                    for (int e = 0; e < 8; ++e) {
-                        int j = xj[i] + e;
-                        if (j < Kw) {
-                            int rs = j / C, c = j % C;
-                            long off = gather_off<0>(n, p, q, rs / S, rs % S, c,
-                                                     H, W, C, stride, pad, dil);
-                            if (off >= 0)
-                                v[e] = reinterpret_cast<const short *>(x)[off];
+            for (int u = 0; u < 4; ++u) {
+                v[u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                int m = m0 + xml[i] + u;
+                if (xj[i] < Kw && m < m_end) {
+                    int n = m / (P * Q);
+                    int pq = m % (P * Q);
+                    int p = pq / Q, q = pq % Q;
+                    if (xvec_ok) {
+                        long off = gather_off<0>(n, p, q, xr[i], xs[i], xc[i],
+                                                 H, W, C, stride, pad, dil);
+                        if (off >= 0)
+                            v[u] = *reinterpret_cast<const s16x8 *>(x + off);
+                    } else {
+#pragma unroll
+                        for (int e = 0; e < 8; ++e) {
+                            int j = xj[i] + e;
+                            if (j < Kw) {
+                                int rs = j / C, c = j % C;
+                                long off = gather_off<0>(n, p, q, rs / S,
+                                                         rs % S, c,
+                                                         H, W, C, stride, pad, dil);
+                                if (off >= 0)
+                                    v[u][e] = reinterpret_cast<const short *>(x)[off];
+                            }
                         }
                     }
                 }
             }
 #pragma unroll
-            for (int e = 0; e < 8; ++e)
-                *reinterpret_cast<short *>(
-                    xT_lds + swz<BKM>(j_local0 + e, xml[i] * 2)) = v[e];
+            for (int e = 0; e < 8; ++e) {
+                s16x4 w4 = {v[0][e], v[1][e], v[2][e], v[3][e]};
+                *reinterpret_cast<s16x4 *>(
+                    xT_lds + swz<BKM>(j_local0 + e, xml[i] * 2)) = w4;
+            }
         }
         __syncthreads();
 

@@ -82,7 +82,16 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     char *w_lds = lds;                         // [BCOUT][BK]
     char *x_lds = lds + BCOUT * BK * 2;        // [BM][BK]
 
-    const int bid = blockIdx.x;
+    // XCD-aware block remap (T1): the dispatcher places block b on XCD b%8;
+    // giving each XCD a contiguous tile range makes neighbor tiles (which
+    // share W panels / x rows) hit the same L2.  Bijective variant.
+    int bid = blockIdx.x;
+    {
+        int nwg = gridDim.x;
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = bid & 7, idx = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
     const int tile_m = (bid % tiles_m) * BM;
     const int tile_c = (bid / tiles_m) * BCOUT;
 

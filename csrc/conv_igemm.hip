@@ -28,10 +28,16 @@
 
 namespace sodkit {
 
-// swizzled byte offset inside a [rows][BK] 16-bit tile, row pitch BK*2 bytes
+// Swizzled byte offset inside a [rows][BK] 16-bit tile, row pitch BK*2 bytes.
+// XOR nibble = (row&7) ^ ((row>>3)&7): conflict-FREE for the b128 fragment
+// reads (16-lane groups read rows r..r+15: evens/odds each land on 8
+// distinct 16B slots) AND <=2-way for the wgrad transpose's b64 writes
+// (rows strided by 8 within a lane group, where a plain (row&7) nibble is
+// constant across the group -> 16-way conflict, measured 23% of wave
+// cycles as SQ_LDS_BANK_CONFLICT).
 template <int BK>
 SOD_DEVINL int swz(int row, int kbyte) {
-    return row * (BK * 2) + (kbyte ^ ((row & 7) << 4));
+    return row * (BK * 2) + (kbyte ^ ((((row & 7) ^ ((row >> 3) & 7))) << 4));
 }
 
 // ---------------------------------------------------------------------------

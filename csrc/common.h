@@ -92,11 +92,15 @@ SOD_DEVINL float wave_allreduce_sum(float v) {
     return v;
 }
 
-// Block-level sum into lane 0 of wave 0 (needs 16 floats of LDS scratch).
+// Block-level sum into flat-thread 0 (needs 16 floats of LDS scratch).
+// Uses the FLAT thread id so 2D/3D blocks work (a threadIdx.x-only wave id
+// is wrong for a 16x16 block: every thread would claim wave 0 and the
+// scratch writes race).
 template <int NWAVES>
 SOD_DEVINL float block_reduce_sum(float v, float *lds_scratch) {
-    int lane = threadIdx.x & 63;
-    int w = threadIdx.x >> 6;
+    int ft = (threadIdx.z * blockDim.y + threadIdx.y) * blockDim.x + threadIdx.x;
+    int lane = ft & 63;
+    int w = ft >> 6;
     v = wave_reduce_sum(v);
     if (lane == 0) lds_scratch[w] = v;
     __syncthreads();
@@ -107,7 +111,7 @@ SOD_DEVINL float block_reduce_sum(float v, float *lds_scratch) {
         for (int off = NWAVES >> 1; off > 0; off >>= 1)
             r += __shfl_down(r, off, 64);
     }
-    return r;  // valid in thread 0
+    return r;  // valid in flat-thread 0
 }
 
 SOD_DEVINL int ceil_div(int a, int b) { return (a + b - 1) / b; }

@@ -186,7 +186,9 @@ def test_hybrid_loss_fwd_bwd(weights):
     zr = z0.detach().float().clone().requires_grad_()
     loss_r = ops.reference.hybrid_loss(zr, t, wb, wi, ws)
     loss_r.backward()
-    assert abs(loss.item() - loss_r.item()) < 2e-3 + 1e-2 * abs(loss_r.item())
+    # tight: the legit bf16-vs-fp32 gap here is ~1e-4 (a loose bound once hid
+    # a block-reduce race that only corrupted the reported loss value)
+    assert abs(loss.item() - loss_r.item()) < 1e-3 + 2e-3 * abs(loss_r.item())
     close(z.grad, zr.grad, rtol=3e-2, atol=1e-4, what="loss dz")
 
 

@@ -103,7 +103,7 @@ __global__ void ssim_stats_kernel(const float *__restrict__ p,
     __shared__ float pp[PATCH][PATCH];
     __shared__ float tt[PATCH][PATCH];
     __shared__ float vmap[5][TILE][PATCH];
-    __shared__ float scratch[16];
+    __shared__ float red2[TILE * TILE];
     float g[WIN];
     gauss_table(g);
     int n = blockIdx.z;
@@ -176,9 +176,15 @@ __global__ void ssim_stats_kernel(const float *__restrict__ p,
         stats[3 * total + idx] = s_tt;
         stats[4 * total + idx] = s_pt;
     }
+    // plain LDS tree reduction (2D block)
     __syncthreads();
-    float r = block_reduce_sum<4>(val, scratch);
-    if (tid == 0) atomicAdd(ssim_sum, r);
+    red2[tid] = val;
+    __syncthreads();
+    for (int sft = TILE * TILE / 2; sft > 0; sft >>= 1) {
+        if (tid < sft) red2[tid] += red2[tid + sft];
+        __syncthreads();
+    }
+    if (tid == 0) atomicAdd(ssim_sum, red2[0]);
 }
 
 // ---- k3: coefficient center-maps for the SSIM backward ----

@@ -387,7 +387,10 @@ static inline dim3 vec_stats_grid(long M, int C) {
     int C8 = C / 8;
     int nch = std::min(C8, 256);
     int rows_pb = 256 / nch;
-    long xblocks = std::min<long>((M + rows_pb - 1) / rows_pb, 512);
+    // target >=8 loop iterations per thread: small tensors get fewer blocks
+    // (=> fewer partial slabs for the reduce pass), big ones fill the chip
+    long xblocks = std::min<long>((M + rows_pb * 8 - 1) / (rows_pb * 8), 512);
+    if (xblocks < 64) xblocks = std::min<long>((M + rows_pb - 1) / rows_pb, 64);
     return dim3((unsigned)xblocks, (C8 + 255) / 256);
 }
 

@@ -78,6 +78,12 @@ class Solver:
 
         self.criterion = HybridLoss(cfg.w_bce, cfg.w_iou, cfg.w_ssim)
         self.loss_scale = cfg.loss_scale if cfg.dtype == "fp16" else 1.0
+        # simple dynamic loss scaling for fp16 (bf16 needs none): on a
+        # non-finite grad norm the step is skipped and the scale halves;
+        # after `growth_interval` good steps it doubles.
+        self.dynamic_scale = cfg.dtype == "fp16"
+        self._good_steps = 0
+        self._growth_interval = 500
 
         self.start_epoch = 0
         self.iter_count = 0
@@ -103,6 +109,35 @@ class Solver:
         return t
 
     # ------------------------------------------------------------------
+    def _grad_norm_and_clip_scale(self) -> float:
+        """Global grad norm over the flat buckets; returns the extra scale
+        factor that implements clipping (folded into the fused SGD's
+        grad_scale — no separate pass over the grads).  Handles the fp16
+        dynamic scaler (returns 0.0 => skip this step)."""
+        cfg = self.cfg
+        if not (cfg.grad_clip > 0 or self.dynamic_scale):
+            return 1.0
+        sq = None
+        for b in self.reducer.buckets:
+            n = b.flat.float().square().sum()
+            sq = n if sq is None else sq + n
+        # norm of the SUMMED (not yet averaged/unscaled) grads
+        raw_norm = torch.sqrt(sq)
+        eff = self.reducer.grad_scale / self.loss_scale  # avg + unscale
+        norm = (raw_norm * eff).item()
+        if self.dynamic_scale and not (norm == norm and norm < float("inf")):
+            self.loss_scale = max(self.loss_scale / 2.0, 1.0)
+            self._good_steps = 0
+            return 0.0
+        if self.dynamic_scale:
+            self._good_steps += 1
+            if self._good_steps >= self._growth_interval:
+                self.loss_scale = min(self.loss_scale * 2.0, 65536.0)
+                self._good_steps = 0
+        if cfg.grad_clip > 0 and norm > cfg.grad_clip:
+            return cfg.grad_clip / (norm + 1e-6)
+        return 1.0
+
     def step(self, img: torch.Tensor, mask: torch.Tensor,
              depth: Optional[torch.Tensor] = None) -> torch.Tensor:
         """One optimization step on already-on-device tensors.  Returns the
@@ -112,8 +147,11 @@ class Solver:
         loss = self.criterion(logits, mask)
         (loss * self.loss_scale).backward() if self.loss_scale != 1.0 else loss.backward()
         self.reducer.finalize()
-        lr = self._lr_at(self.iter_count)
-        self.optimizer.step(lr, self.reducer.grad_scale / self.loss_scale)
+        clip_scale = self._grad_norm_and_clip_scale()
+        if clip_scale != 0.0:   # 0.0 == overflow: skip step, scale halved
+            lr = self._lr_at(self.iter_count)
+            self.optimizer.step(
+                lr, self.reducer.grad_scale / self.loss_scale * clip_scale)
         self.iter_count += 1
         return loss.detach()
 
@@ -132,9 +170,14 @@ class Solver:
         if self.rank == 0 and cfg.tb:
             self.tb = TBRecorder(self.paths["tb"], enabled=True)
 
+        from ..utils.profiler import profile_ctx
         self.model.train()
         loss_meter = AvgMeter()
         ips_meter = AvgMeter()
+        prof_cm = profile_ctx(cfg.profile_steps,
+                              os.path.join(self.paths["tb"], "profiler"),
+                              self.rank)
+        prof_step = prof_cm.__enter__()
         for epoch in range(self.start_epoch, cfg.epoch_num):
             if sampler is not None:
                 sampler.set_epoch(epoch)
@@ -150,6 +193,7 @@ class Solver:
                 img = self._to_device(img, cast=True)
                 mask = self._to_device(mask, cast=False)
                 loss = self.step(img, mask, depth)
+                prof_step()
                 now = time.perf_counter()
                 ips = img.shape[0] * self.world / (now - t_last)
                 t_last = now
@@ -169,6 +213,7 @@ class Solver:
                 save_checkpoint(self.paths, epoch, self.model, self.optimizer,
                                 {"loss_scale": self.loss_scale})
             self._barrier()
+        prof_cm.__exit__(None, None, None)
         if self.tb is not None:
             self.tb.close()
 

@@ -71,10 +71,22 @@ SOD_DEVINL long gather_off(int n, int p, int q, int r, int s, int c,
 //   B_src: NHWC source for the gather (x for fwd, dy for dgrad)
 //   out  : [M_total][Ncout] (NHWC output flattened over pixels)
 // ---------------------------------------------------------------------------
+// addrspace helpers for global_load_lds (async global->LDS DMA)
+typedef __attribute__((address_space(3))) void lds_void_t;
+typedef __attribute__((address_space(1))) const void gbl_void_t;
+
+SOD_DEVINL void glds16(const void *gsrc, char *lds_dst) {
+    __builtin_amdgcn_global_load_lds((const gbl_void_t *)gsrc,
+                                     (lds_void_t *)lds_dst, 16, 0, 0);
+}
+
+SOD_DEVINL int swz_nib(int row) { return ((row & 7) ^ ((row >> 3) & 7)) << 4; }
+
 template <typename T, int MODE, int BCOUT, int BM>
 __global__ __launch_bounds__(256)
 void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                   const float *__restrict__ bias, T *__restrict__ out,
+                  const T *__restrict__ zero_page,
                   int Ns, int Hs, int Ws, int Cs,   // gather-source dims
                   int P, int Q,                     // output spatial
                   int Ncout,                        // output channels
@@ -83,6 +95,8 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     constexpr int BK = 64;
     constexpr int FC = BCOUT / 32;   // a-fragments per wave (2x2 wave grid)
     constexpr int FM = BM / 32;      // b-fragments per wave
+    constexpr int GW = BCOUT * BK * 2 / 4096;  // glds insts/wave, W tile
+    constexpr int GX = BM * BK * 2 / 4096;     // glds insts/wave, X tile
 
     __shared__ __align__(16) char lds[(BCOUT + BM) * BK * 2];
     char *w_lds = lds;                         // [BCOUT][BK]
@@ -126,6 +140,34 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
         xp[i] = pq / Q;
         xq[i] = pq % Q;
     }
+    // --- glds lane geometry: the LDS image is written lane-linear, so the
+    // tile swizzle moves onto the SOURCE address (inverse-swizzle per lane;
+    // CDNA4 guide rule 21).  Usable when 16B global chunks are whole
+    // (Cs % 8 == 0 and K_total % 8 == 0); OOB/padding lanes read a zero page.
+    const bool glds_ok = vec_ok && wvec_ok;
+    int gw_row[GW], gw_kel[GW];        // W tile: row, k-elem offset in step
+    int gx_row[GX], gx_kel[GX];
+    int gx_n[GX], gx_p[GX], gx_q[GX];  // X tile: pixel geometry per row
+#pragma unroll
+    for (int i = 0; i < GW; ++i) {
+        int off = (wv * GW + i) * 1024 + lane * 16;
+        int row = off >> 7;
+        gw_row[i] = row;
+        gw_kel[i] = ((off & 127) ^ swz_nib(row)) >> 1;
+    }
+#pragma unroll
+    for (int i = 0; i < GX; ++i) {
+        int off = (wv * GX + i) * 1024 + lane * 16;
+        int row = off >> 7;
+        gx_row[i] = row;
+        gx_kel[i] = ((off & 127) ^ swz_nib(row)) >> 1;
+        int m = tile_m + row;
+        int mm = m < M_total ? m : 0;
+        gx_n[i] = mm / (P * Q);
+        int pq = mm % (P * Q);
+        gx_p[i] = pq / Q;
+        gx_q[i] = pq % Q;
+    }
 
     f32x4 acc[FC][FM];
 #pragma unroll
@@ -134,6 +176,33 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
         for (int b = 0; b < FM; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     for (int k0 = 0; k0 < K_total; k0 += BK) {
+        if (glds_ok) {
+            // ---- async global->LDS staging (16B per lane per inst) ----
+#pragma unroll
+            for (int i = 0; i < GW; ++i) {
+                int cout = tile_c + gw_row[i];
+                int k = k0 + gw_kel[i];
+                const T *src = (cout < Ncout && k < K_total)
+                                   ? A_w + (long)cout * K_total + k
+                                   : zero_page;
+                glds16(src, w_lds + (wv * GW + i) * 1024);
+            }
+#pragma unroll
+            for (int i = 0; i < GX; ++i) {
+                int k = k0 + gx_kel[i];
+                const T *src = zero_page;
+                if (tile_m + gx_row[i] < M_total && k < K_total) {
+                    int rs = k / Cs, c = k % Cs;
+                    long off = gather_off<MODE>(gx_n[i], gx_p[i], gx_q[i],
+                                                rs / S, rs % S, c,
+                                                Hs, Ws, Cs, stride, pad, dil);
+                    if (off >= 0) src = B_src + off;
+                }
+                glds16(src, x_lds + (wv * GX + i) * 1024);
+            }
+            __syncthreads();
+            goto mfma_phase;
+        }
         // ---- stage W tile: [BCOUT][BK], coalesced rows from A_w ----
 #pragma unroll
         for (int i = 0; i < WCH; ++i) {
@@ -189,6 +258,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
         }
         __syncthreads();
 
+    mfma_phase:
         // ---- MFMA over the K-step ----
 #pragma unroll
         for (int kk = 0; kk < BK / 32; ++kk) {
@@ -475,6 +545,14 @@ static inline int out_dim(int in, int k, int stride, int pad, int dil) {
 
 namespace {
 
+// 128-byte device zero page for glds OOB/padding lanes
+static void *zero_page_ptr(const torch::Tensor &like) {
+    static torch::Tensor z;
+    if (!z.defined() || z.device() != like.device())
+        z = torch::zeros({32}, like.options().dtype(torch::kFloat32));
+    return z.data_ptr();
+}
+
 template <typename scalar_t, int MODE>
 void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
                   const torch::Tensor &bias, torch::Tensor &out,
@@ -485,6 +563,7 @@ void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
     auto stream = at::hip::getCurrentHIPStream();
     const float *bias_ptr = bias.defined() && bias.numel()
                                 ? bias.data_ptr<float>() : nullptr;
+    const T *zpage = reinterpret_cast<const T *>(zero_page_ptr(src));
     auto launch = [&](auto bc_tag, auto bm_tag) {
         constexpr int BC = decltype(bc_tag)::value;
         constexpr int BMv = decltype(bm_tag)::value;
@@ -495,7 +574,7 @@ void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
                            dim3(256), 0, stream.stream(),
                            reinterpret_cast<const T *>(src.data_ptr()),
                            reinterpret_cast<const T *>(w.data_ptr()), bias_ptr,
-                           reinterpret_cast<T *>(out.data_ptr()),
+                           reinterpret_cast<T *>(out.data_ptr()), zpage,
                            Ns, Hs, Ws, Cs, P, Q, Ncout, R, S, stride, pad, dil,
                            (int)M_total, K_total, tiles_m);
     };

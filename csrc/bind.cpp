@@ -14,6 +14,10 @@ std::vector<torch::Tensor> bn_finalize(torch::Tensor stats, double n,
                                        double momentum, double eps,
                                        torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training);
+std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x, double n,
+                                             double momentum, double eps,
+                                             torch::Tensor running_mean,
+                                             torch::Tensor running_var);
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd,
                        torch::Tensor gamma, torch::Tensor beta, bool relu,
                        torch::Tensor residual);
@@ -47,6 +51,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("channel_sum", &channel_sum);
     m.def("bn_stats", &bn_stats);
     m.def("bn_finalize", &bn_finalize);
+    m.def("bn_stats_finalize", &bn_stats_finalize);
     m.def("bn_apply", &bn_apply);
     m.def("bn_bwd_reduce", &bn_bwd_reduce);
     m.def("bn_bwd_apply", &bn_bwd_apply);

@@ -92,6 +92,43 @@ __global__ void slab_reduce_kernel(const float *__restrict__ partial,
     out[r] = ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
 }
 
+// fused: reduce the per-block stat slabs AND produce mean/invstd/running
+// stats in one kernel (halves the per-BN-layer launch count on the
+// non-SyncBN forward path)
+__global__ void bn_reduce_finalize_kernel(const float *__restrict__ partial,
+                                          int nb, int C, float n,
+                                          float momentum, float eps,
+                                          float *__restrict__ running_mean,
+                                          float *__restrict__ running_var,
+                                          float *__restrict__ mean_out,
+                                          float *__restrict__ invstd_out) {
+    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    float q[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    long R = 2L * C;
+    int b = 0;
+    for (; b + 8 <= nb; b += 8)
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+            a[u] += partial[(long)(b + u) * R + c];
+            q[u] += partial[(long)(b + u) * R + C + c];
+        }
+    for (; b < nb; ++b) {
+        a[0] += partial[(long)b * R + c];
+        q[0] += partial[(long)b * R + C + c];
+    }
+    float sum = ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
+    float ssq = ((q[0] + q[1]) + (q[2] + q[3])) + ((q[4] + q[5]) + (q[6] + q[7]));
+    float mean = sum / n;
+    float var = ssq / n - mean * mean;
+    float unbiased = var * (n / fmaxf(n - 1.f, 1.f));
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    mean_out[c] = mean;
+    invstd_out[c] = rsqrtf(var + eps);
+}
+
 template <typename T>
 __global__ void bn_stats_kernel(const T *__restrict__ x, float *__restrict__ out,
                                 long M, int C) {
@@ -444,6 +481,43 @@ std::vector<torch::Tensor> bn_finalize(torch::Tensor stats, double n,
                        running_mean.data_ptr<float>(),
                        running_var.data_ptr<float>(), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), C, training);
+    return {mean, invstd};
+}
+
+std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x, double n,
+                                              double momentum, double eps,
+                                              torch::Tensor running_mean,
+                                              torch::Tensor running_var) {
+    TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+    int C = x.size(1);
+    long M = x.numel() / C;
+    auto stream = at::hip::getCurrentHIPStream();
+    auto mean = torch::empty({C}, running_mean.options());
+    auto invstd = torch::empty({C}, running_mean.options());
+    bool vec = (C % 8 == 0) && x.element_size() == 2;
+    if (!vec) {
+        // fallback: plain stats (atomics) + finalize
+        auto stats = bn_stats(x);
+        return bn_finalize(stats, n, momentum, eps, running_mean, running_var,
+                           true);
+    }
+    dim3 g = vec_stats_grid(M, C);
+    int nb = g.x;
+    auto partial = torch::empty({nb, 2L * C},
+                                x.options().dtype(torch::kFloat32));
+    DISPATCH_T(x, "bn_stats_finalize", [&] {
+        hipLaunchKernelGGL((sodkit::bn_stats_vec_kernel<T>), g, dim3(256), 0,
+                           stream.stream(),
+                           reinterpret_cast<const T *>(x.data_ptr()),
+                           partial.data_ptr<float>(), M, C);
+    });
+    hipLaunchKernelGGL(sodkit::bn_reduce_finalize_kernel,
+                       dim3((C + 255) / 256), dim3(256), 0, stream.stream(),
+                       partial.data_ptr<float>(), nb, C, (float)n,
+                       (float)momentum, (float)eps,
+                       running_mean.data_ptr<float>(),
+                       running_var.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>());
     return {mean, invstd};
 }
 

@@ -39,16 +39,22 @@ class _BNActFn(torch.autograd.Function):
         C = x.shape[1]
         n_local = x.numel() // C
         if training:
-            stats = ext().bn_stats(x)                      # fp32 [2, C]
-            n_total = float(n_local)
-            if process_group is not None and dist.is_initialized() \
-                    and dist.get_world_size(process_group) > 1:
+            sync = (process_group is not None and dist.is_initialized()
+                    and dist.get_world_size(process_group) > 1)
+            if sync:
+                stats = ext().bn_stats(x)                  # fp32 [2, C]
                 dist.all_reduce(stats, group=process_group)
                 # equal per-rank shapes by construction (DistributedSampler
                 # with drop_last) -> no count exchange, no host sync
                 n_total = float(n_local) * dist.get_world_size(process_group)
-            mean, invstd = ext().bn_finalize(stats, n_total, momentum, eps,
-                                             running_mean, running_var, True)
+                mean, invstd = ext().bn_finalize(stats, n_total, momentum,
+                                                 eps, running_mean,
+                                                 running_var, True)
+            else:
+                n_total = float(n_local)
+                mean, invstd = ext().bn_stats_finalize(x, n_total, momentum,
+                                                       eps, running_mean,
+                                                       running_var)
         else:
             n_total = 0.0                                   # eval marker
             mean, invstd = ext().bn_finalize(running_mean, 0.0, momentum, eps,

@@ -50,7 +50,9 @@ class Config:
     dtype: str = "bf16"                  # bf16 | fp16 | fp32 (compute dtype on GPU)
     loss_scale: float = 1.0              # static loss scale (fp16 only; bf16 needs none)
 
-    # loss weights (hybrid pixel-wise loss, BASELINE.json north star)
+    # loss selection + weights (hybrid = the fused BCE+IoU+SSIM kernel;
+    # cel / hel are the reference's MINet / HDFNet recipes)
+    loss: str = "hybrid"
     w_bce: float = 1.0
     w_iou: float = 1.0
     w_ssim: float = 1.0

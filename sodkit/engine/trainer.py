@@ -19,7 +19,7 @@ import torch.distributed as dist
 from .. import ops
 from ..config import Config
 from ..data import create_loader
-from ..losses import HybridLoss
+from ..losses import build_loss
 from ..metrics import MAEMeter, PRFMeter
 from ..models import build_model, convert_precision, set_syncbn
 from ..parallel import GradReducer, setup, get_rank, get_world_size
@@ -76,7 +76,7 @@ class Solver:
                                     weight_decay=cfg.weight_decay,
                                     nesterov=cfg.nesterov)
 
-        self.criterion = HybridLoss(cfg.w_bce, cfg.w_iou, cfg.w_ssim)
+        self.criterion = build_loss(cfg)
         self.loss_scale = cfg.loss_scale if cfg.dtype == "fp16" else 1.0
         # simple dynamic loss scaling for fp16 (bf16 needs none): on a
         # non-finite grad norm the step is skipped and the scale halves;

@@ -59,3 +59,27 @@ def test_meters_match_per_image_fns():
     pf.update(pred, mask)
     out = pf.compute()
     assert 0 <= out["maxf"] <= 1
+
+
+def test_cel_and_hel_losses():
+    from sodkit.losses import CELLoss, HELLoss
+    torch.manual_seed(3)
+    z = torch.randn(2, 1, 24, 24)
+    t = (torch.rand(2, 1, 24, 24) > 0.5).float()
+    for crit in (CELLoss(), HELLoss()):
+        z2 = z.clone().requires_grad_()
+        loss = crit(z2, t)
+        loss.backward()
+        assert loss.item() == loss.item() and z2.grad is not None
+    # CEL is ~0 on a perfect confident prediction
+    zgood = (t * 2 - 1) * 12.0
+    assert CELLoss(w_bce=0.0)(zgood, t).item() < 0.01
+    assert HELLoss()(zgood, t).item() < 0.02
+
+
+def test_build_loss_dispatch():
+    from sodkit.config import make_config
+    from sodkit.losses import build_loss, HybridLoss, CELLoss, HELLoss
+    assert isinstance(build_loss(make_config("cpu_plumbing")), HybridLoss)
+    assert isinstance(build_loss(make_config("cpu_plumbing", loss="cel")), CELLoss)
+    assert isinstance(build_loss(make_config("cpu_plumbing", loss="hel")), HELLoss)

@@ -98,9 +98,9 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     constexpr int GW = BCOUT * BK * 2 / 4096;  // glds insts/wave, W tile
     constexpr int GX = BM * BK * 2 / 4096;     // glds insts/wave, X tile
 
-    __shared__ __align__(16) char lds[(BCOUT + BM) * BK * 2];
-    char *w_lds = lds;                         // [BCOUT][BK]
-    char *x_lds = lds + BCOUT * BK * 2;        // [BM][BK]
+    // two buffers: the glds path overlaps tile t+1 staging with tile t MFMAs
+    constexpr int TILE_BYTES = (BCOUT + BM) * BK * 2;
+    __shared__ __align__(16) char lds[2 * TILE_BYTES];
 
     // XCD-aware block remap (T1): the dispatcher places block b on XCD b%8;
     // giving each XCD a contiguous tile range makes neighbor tiles (which
@@ -175,34 +175,36 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
 #pragma unroll
         for (int b = 0; b < FM; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    for (int k0 = 0; k0 < K_total; k0 += BK) {
-        if (glds_ok) {
-            // ---- async global->LDS staging (16B per lane per inst) ----
+    auto stage_glds = [&](int buf, int k0) {
+        char *w_lds = lds + buf * TILE_BYTES;
+        char *x_lds = w_lds + BCOUT * BK * 2;
 #pragma unroll
-            for (int i = 0; i < GW; ++i) {
-                int cout = tile_c + gw_row[i];
-                int k = k0 + gw_kel[i];
-                const T *src = (cout < Ncout && k < K_total)
-                                   ? A_w + (long)cout * K_total + k
-                                   : zero_page;
-                glds16(src, w_lds + (wv * GW + i) * 1024);
-            }
-#pragma unroll
-            for (int i = 0; i < GX; ++i) {
-                int k = k0 + gx_kel[i];
-                const T *src = zero_page;
-                if (tile_m + gx_row[i] < M_total && k < K_total) {
-                    int rs = k / Cs, c = k % Cs;
-                    long off = gather_off<MODE>(gx_n[i], gx_p[i], gx_q[i],
-                                                rs / S, rs % S, c,
-                                                Hs, Ws, Cs, stride, pad, dil);
-                    if (off >= 0) src = B_src + off;
-                }
-                glds16(src, x_lds + (wv * GX + i) * 1024);
-            }
-            __syncthreads();
-            goto mfma_phase;
+        for (int i = 0; i < GW; ++i) {
+            int cout = tile_c + gw_row[i];
+            int k = k0 + gw_kel[i];
+            const T *src = (cout < Ncout && k < K_total)
+                               ? A_w + (long)cout * K_total + k
+                               : zero_page;
+            glds16(src, w_lds + (wv * GW + i) * 1024);
         }
+#pragma unroll
+        for (int i = 0; i < GX; ++i) {
+            int k = k0 + gx_kel[i];
+            const T *src = zero_page;
+            if (tile_m + gx_row[i] < M_total && k < K_total) {
+                int rs = k / Cs, c = k % Cs;
+                long off = gather_off<MODE>(gx_n[i], gx_p[i], gx_q[i],
+                                            rs / S, rs % S, c,
+                                            Hs, Ws, Cs, stride, pad, dil);
+                if (off >= 0) src = B_src + off;
+            }
+            glds16(src, x_lds + (wv * GX + i) * 1024);
+        }
+    };
+
+    auto stage_reg = [&](int k0) {
+        char *w_lds = lds;
+        char *x_lds = lds + BCOUT * BK * 2;
         // ---- stage W tile: [BCOUT][BK], coalesced rows from A_w ----
 #pragma unroll
         for (int i = 0; i < WCH; ++i) {
@@ -256,10 +258,11 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
             }
             *reinterpret_cast<s16x8 *>(x_lds + swz<BK>(row, (ch % (BK / 8)) * 16)) = v;
         }
-        __syncthreads();
+    };
 
-    mfma_phase:
-        // ---- MFMA over the K-step ----
+    auto mfma_tile = [&](int buf) {
+        const char *w_lds = lds + buf * TILE_BYTES;
+        const char *x_lds = w_lds + BCOUT * BK * 2;
 #pragma unroll
         for (int kk = 0; kk < BK / 32; ++kk) {
             int kbyte = kk * 64 + (lane >> 4) * 16;
@@ -280,7 +283,28 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                 for (int fm = 0; fm < FM; ++fm)
                     acc[fc][fm] = MfmaOp<T>::mma(a[fc], b[fm], acc[fc][fm]);
         }
+    };
+
+    const int nsteps = (K_total + BK - 1) / BK;
+    if (glds_ok) {
+        // 2-phase double buffer: tile s+1's async staging issues BEFORE
+        // tile s's MFMAs; one barrier per tile (it also drains the DMA)
+        stage_glds(0, 0);
         __syncthreads();
+        int cur = 0;
+        for (int s = 0; s < nsteps; ++s) {
+            if (s + 1 < nsteps) stage_glds(cur ^ 1, (s + 1) * BK);
+            mfma_tile(cur);
+            __syncthreads();
+            cur ^= 1;
+        }
+    } else {
+        for (int s = 0; s < nsteps; ++s) {
+            stage_reg(s * BK);
+            __syncthreads();
+            mfma_tile(0);
+            __syncthreads();
+        }
     }
 
     // ---- epilogue: D[cout][m] -> NHWC out[m][cout], 8B packed stores ----

@@ -31,7 +31,7 @@ class Config:
     train_samples: int = 3200            # synthetic "dataset" length per epoch
     val_samples: int = 128               # fixed-seed synthetic validation split
     batch_size: int = 32                 # per-GPU
-    num_workers: int = 2
+    num_workers: int = 8
     data_seed: int = 20260913
 
     # optimization

@@ -284,3 +284,76 @@ def test_hdfnet_train_step_runs():
                      cast=False)
     loss = s.step(x, t, d).item()
     assert loss == loss and loss < 100
+
+
+@pytest.mark.parametrize("dtype", [torch.float16])
+def test_conv2d_backward_fp16(dtype):
+    from sodkit import ops
+    torch.manual_seed(9)
+    N, Cin, H, Cout, k, st, pad, dil = 2, 32, 24, 48, 3, 1, 1, 1
+    x0 = torch.randn(N, Cin, H, H, device="cuda").to(dtype)
+    w0 = (torch.randn(Cout, k, k, Cin, device="cuda") * 0.1).to(dtype)
+    x = _cl(x0).requires_grad_()
+    w = w0.clone().requires_grad_()
+    y = ops.conv2d(x, w, None, st, pad, dil)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    xr = x0.float().detach().requires_grad_()
+    wr = w0.float().permute(0, 3, 1, 2).detach().requires_grad_()
+    yr = F.conv2d(xr, wr, None, stride=st, padding=pad, dilation=dil)
+    yr.backward(gy.float())
+    close(x.grad, xr.grad, what="fp16 dgrad")
+    close(w.grad, wr.grad.permute(0, 2, 3, 1), rtol=3e-2, atol=3e-2,
+          what="fp16 wgrad")
+
+
+def test_conv2d_strided_dilated_combo():
+    from sodkit import ops
+    torch.manual_seed(10)
+    x = _cl(torch.randn(1, 16, 33, 33, device="cuda").to(torch.bfloat16))
+    w = (torch.randn(24, 3, 3, 16, device="cuda") * 0.2).to(torch.bfloat16)
+    y = ops.conv2d(x, w, None, 2, 2, 2)
+    yr = F.conv2d(x.float(), w.float().permute(0, 3, 1, 2), None,
+                  stride=2, padding=2, dilation=2)
+    assert y.shape == yr.shape
+    close(y, yr, what="stride2+dil2")
+
+
+def test_bn_odd_channels_scalar_path():
+    from sodkit import ops
+    torch.manual_seed(11)
+    C = 9  # C % 8 != 0 -> scalar kernels
+    x = _cl(torch.randn(3, C, 14, 14, device="cuda").to(torch.bfloat16)) \
+        .requires_grad_()
+    gamma = (torch.rand(C, device="cuda") + 0.5).requires_grad_()
+    beta = torch.randn(C, device="cuda").requires_grad_()
+    rm, rv = torch.zeros(C, device="cuda"), torch.ones(C, device="cuda")
+    y = ops.bn_act(x, gamma, beta, rm, rv, True, 0.1, 1e-5, True, None)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    xr = x.detach().float().clone().requires_grad_()
+    gr = gamma.detach().clone().requires_grad_()
+    br = beta.detach().clone().requires_grad_()
+    yr = ops.reference.bn_act(xr, gr, br, torch.zeros(C, device="cuda"),
+                              torch.ones(C, device="cuda"), True, 0.1, 1e-5,
+                              True, None)
+    yr.backward(gy.float())
+    close(y, yr, what="bn C=9 fwd")
+    close(x.grad, xr.grad, rtol=3e-2, atol=2e-2, what="bn C=9 dx")
+
+
+def test_bilinear_odd_channels():
+    from sodkit import ops
+    x = _cl(torch.randn(2, 3, 13, 17, device="cuda").to(torch.bfloat16))
+    y = ops.bilinear2d(x, 29, 31)
+    yr = F.interpolate(x.float(), size=(29, 31), mode="bilinear",
+                       align_corners=False)
+    close(y, yr, what="bilinear C=3")
+
+
+def test_channel_sum_matches():
+    from sodkit import _C
+    x = _cl(torch.randn(3, 24, 10, 10, device="cuda").to(torch.bfloat16))
+    got = _C.channel_sum(x)
+    want = x.float().sum(dim=[0, 2, 3])
+    close(got, want, rtol=1e-2, atol=1e-2, what="channel_sum")

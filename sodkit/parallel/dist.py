@@ -34,8 +34,11 @@ def setup(backend: str = "auto", device: str = "auto") -> torch.device:
     torchrun env: skips init and returns the device."""
     use_cuda = torch.cuda.is_available() and device != "cpu"
     if use_cuda:
-        torch.cuda.set_device(get_local_rank())
-        dev = torch.device("cuda", get_local_rank())
+        # modulo lets >1 rank share a GPU (CPU-backend integration smokes);
+        # on a real node local_rank < device_count and this is the identity
+        idx = get_local_rank() % max(torch.cuda.device_count(), 1)
+        torch.cuda.set_device(idx)
+        dev = torch.device("cuda", idx)
     else:
         dev = torch.device("cpu")
 

@@ -39,6 +39,10 @@ torch::Tensor loss_bwd(torch::Tensor p, torch::Tensor t, torch::Tensor img_sums,
                        torch::Tensor stats, double wb, double wi, double ws,
                        torch::Tensor go);
 
+torch::Tensor dynfilter_fwd(torch::Tensor x, torch::Tensor k, int64_t dil);
+std::vector<torch::Tensor> dynfilter_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor k, int64_t dil);
+
 void sgd_fused(torch::Tensor ptrs, torch::Tensor numels, torch::Tensor wds,
                torch::Tensor chunks, int64_t dtype_code, double lr, double mu,
                double gscale, bool nesterov);
@@ -59,5 +63,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bilinear_bwd", &bilinear_bwd);
     m.def("loss_fwd", &loss_fwd);
     m.def("loss_bwd", &loss_bwd);
+    m.def("dynfilter_fwd", &dynfilter_fwd);
+    m.def("dynfilter_bwd", &dynfilter_bwd);
     m.def("sgd_fused", &sgd_fused);
 }

@@ -20,6 +20,7 @@ SRC = [
     "csrc/bilinear.hip",
     "csrc/loss.hip",
     "csrc/sgd.hip",
+    "csrc/dynfilter.hip",
 ]
 
 setup(

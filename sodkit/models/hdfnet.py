@@ -7,8 +7,8 @@ generating spatially-variant kernels from depth features — SURVEY §2.1,
 DDPM: depth-derived features predict a per-pixel 3x3 kernel per dilation
 d in {1, 2, 4}; each is applied to the RGB decoder feature as a dynamic
 (spatially-variant) filter; the pyramid sum passes through Conv+BN+ReLU.
-The dynamic filtering currently composes unfold+mul (ATen); a dedicated HIP
-kernel (K10 generalization) is scheduled work.
+Dynamic filtering runs
+on the dedicated K10 HIP kernels (csrc/dynfilter.hip) on GPU.
 """
 
 from __future__ import annotations
@@ -17,18 +17,9 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import bilinear2d, upsample_add
+from ..ops import bilinear2d, dynamic_filter, upsample_add
 from .blocks import ConvBNReLU, SodConv2d
 from .backbones import resnet50, vgg16
-
-
-def dynamic_filter(x: torch.Tensor, kernels: torch.Tensor, dilation: int) -> torch.Tensor:
-    """Apply per-pixel 3x3 kernels [N,9,H,W] to x [N,C,H,W] at `dilation`."""
-    n, c, h, w = x.shape
-    cols = F.unfold(x.contiguous(), 3, dilation=dilation, padding=dilation)
-    cols = cols.view(n, c, 9, h * w)
-    k = kernels.reshape(n, 1, 9, h * w)
-    return (cols * k).sum(dim=2).view(n, c, h, w)
 
 
 class DenseTransLayer(nn.Module):

@@ -113,6 +113,20 @@ def bilinear2d(x: torch.Tensor, out_h: int, out_w: int,
 
 
 # ---------------------------------------------------------------------------
+# Dynamic (spatially-variant) 3x3 filtering — HDFNet DDPM (K10).
+# ---------------------------------------------------------------------------
+
+def dynamic_filter(x: torch.Tensor, kernels: torch.Tensor,
+                   dilation: int) -> torch.Tensor:
+    """x [N,C,H,W], kernels [N,9,H,W] -> sum_e k_e * shift_e(x), zero pad."""
+    n, c, h, w = x.shape
+    cols = F.unfold(x.contiguous(), 3, dilation=dilation, padding=dilation)
+    cols = cols.view(n, c, 9, h * w)
+    k = kernels.reshape(n, 1, 9, h * w)
+    return (cols * k).sum(dim=2).view(n, c, h, w)
+
+
+# ---------------------------------------------------------------------------
 # Hybrid pixel-wise loss: BCE-with-logits + soft IoU + SSIM (the [BJ:5]
 # generalized loss; the reference's CEL/HEL family, SURVEY §2.1 loss/).
 # All math in fp32.  Per-image IoU and SSIM, averaged over the batch.

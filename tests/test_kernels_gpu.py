@@ -357,3 +357,26 @@ def test_channel_sum_matches():
     got = _C.channel_sum(x)
     want = x.float().sum(dim=[0, 2, 3])
     close(got, want, rtol=1e-2, atol=1e-2, what="channel_sum")
+
+
+def test_dynamic_filter_fwd_bwd():
+    from sodkit import ops
+    torch.manual_seed(12)
+    N, C, H, W = 2, 64, 18, 14
+    for dil in (1, 2, 4):
+        x0 = torch.randn(N, C, H, W, device="cuda").to(torch.bfloat16)
+        k0 = torch.softmax(torch.randn(N, 9, H, W, device="cuda"), dim=1) \
+            .to(torch.bfloat16)
+        x = _cl(x0).detach().clone().requires_grad_()
+        k = _cl(k0).detach().clone().requires_grad_()
+        y = ops.dynamic_filter(x, k, dil)
+        gy = torch.randn_like(y)
+        y.backward(gy)
+
+        xr = x0.detach().float().clone().requires_grad_()
+        kr = k0.detach().float().clone().requires_grad_()
+        yr = ops.reference.dynamic_filter(xr, kr, dil)
+        yr.backward(gy.float())
+        close(y, yr, what=f"dynfilter fwd d{dil}")
+        close(x.grad, xr.grad, rtol=3e-2, atol=2e-2, what=f"dynfilter dx d{dil}")
+        close(k.grad, kr.grad, rtol=3e-2, atol=2e-2, what=f"dynfilter dk d{dil}")

@@ -55,6 +55,14 @@ SOD_DEVINL long gather_off(int n, int p, int q, int r, int s, int c,
         if ((unsigned)ih >= (unsigned)Hs || (unsigned)iw >= (unsigned)Ws)
             return -1;
         return (((long)n * Hs + ih) * Ws + iw) * Cs + c;
+    } else if (stride == 1) {
+        // fast path: most layers are stride 1 and the generic path's
+        // runtime %/÷ by stride costs real VALU in the staging loop
+        int oh = p + pad - r * dil;
+        int ow = q + pad - s * dil;
+        if ((unsigned)oh >= (unsigned)Hs || (unsigned)ow >= (unsigned)Ws)
+            return -1;
+        return (((long)n * Hs + oh) * Ws + ow) * Cs + c;
     } else {
         int hh = p + pad - r * dil;
         int ww = q + pad - s * dil;
@@ -98,9 +106,11 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     constexpr int GW = BCOUT * BK * 2 / 4096;  // glds insts/wave, W tile
     constexpr int GX = BM * BK * 2 / 4096;     // glds insts/wave, X tile
 
-    // two buffers: the glds path overlaps tile t+1 staging with tile t MFMAs
+    // dynamic LDS: 2 buffers when the glds path double-buffers (K > BK),
+    // 1 otherwise — a fixed 2x allocation halves occupancy exactly for the
+    // low-arithmetic-intensity single-K-step convs that need it most
     constexpr int TILE_BYTES = (BCOUT + BM) * BK * 2;
-    __shared__ __align__(16) char lds[2 * TILE_BYTES];
+    extern __shared__ __attribute__((aligned(16))) char lds[];
 
     // XCD-aware block remap (T1): the dispatcher places block b on XCD b%8;
     // giving each XCD a contiguous tile range makes neighbor tiles (which
@@ -148,6 +158,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     int gw_row[GW], gw_kel[GW];        // W tile: row, k-elem offset in step
     int gx_row[GX], gx_kel[GX];
     int gx_n[GX], gx_p[GX], gx_q[GX];  // X tile: pixel geometry per row
+    int gx_c[GX], gx_r[GX], gx_s[GX];  // incremental (r,s,c) per inst
 #pragma unroll
     for (int i = 0; i < GW; ++i) {
         int off = (wv * GW + i) * 1024 + lane * 16;
@@ -160,7 +171,12 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
         int off = (wv * GX + i) * 1024 + lane * 16;
         int row = off >> 7;
         gx_row[i] = row;
-        gx_kel[i] = ((off & 127) ^ swz_nib(row)) >> 1;
+        int kel = ((off & 127) ^ swz_nib(row)) >> 1;
+        gx_kel[i] = kel;
+        int rs = kel / Cs;
+        gx_c[i] = kel % Cs;
+        gx_r[i] = rs / S;
+        gx_s[i] = rs % S;
         int m = tile_m + row;
         int mm = m < M_total ? m : 0;
         gx_n[i] = mm / (P * Q);
@@ -189,16 +205,20 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
         }
 #pragma unroll
         for (int i = 0; i < GX; ++i) {
-            int k = k0 + gx_kel[i];
             const T *src = zero_page;
-            if (tile_m + gx_row[i] < M_total && k < K_total) {
-                int rs = k / Cs, c = k % Cs;
+            if (tile_m + gx_row[i] < M_total && k0 + gx_kel[i] < K_total) {
                 long off = gather_off<MODE>(gx_n[i], gx_p[i], gx_q[i],
-                                            rs / S, rs % S, c,
+                                            gx_r[i], gx_s[i], gx_c[i],
                                             Hs, Ws, Cs, stride, pad, dil);
                 if (off >= 0) src = B_src + off;
             }
             glds16(src, x_lds + (wv * GX + i) * 1024);
+            // advance (r,s,c) by BK elements for the next K-step
+            // (stage_glds is called once per k0 in increasing order)
+            int c = gx_c[i] + BK;
+            int sS = gx_s[i], rR = gx_r[i];
+            while (c >= Cs) { c -= Cs; if (++sS == S) { sS = 0; ++rR; } }
+            gx_c[i] = c; gx_s[i] = sS; gx_r[i] = rR;
         }
     };
 
@@ -386,6 +406,7 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     // write-instruction-bound).
     constexpr int XCH = BJ * (BKM / 4) / 8 / 256;
     int xr[XCH], xs[XCH], xc[XCH], xj[XCH], xml[XCH];
+    int xgn[XCH], xgp[XCH], xgq[XCH];   // incremental (n,p,q) of the quad base
 #pragma unroll
     for (int i = 0; i < XCH; ++i) {
         int ch = t + i * 256;
@@ -398,6 +419,11 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
         xc[i] = jj % C;
         xr[i] = rs / S;
         xs[i] = rs % S;
+        long m = m_begin + xml[i];
+        xgn[i] = (int)(m / ((long)P * Q));
+        int pq = (int)(m % ((long)P * Q));
+        xgp[i] = pq / Q;
+        xgq[i] = pq % Q;
     }
 
     f32x4 acc[FC][FJ];
@@ -449,9 +475,10 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                 v[u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
                 int m = m0 + xml[i] + u;
                 if (xj[i] < Kw && m < m_end) {
-                    int n = m / (P * Q);
-                    int pq = m % (P * Q);
-                    int p = pq / Q, q = pq % Q;
+                    // quad-base geometry is maintained incrementally; +u
+                    // wraps at most once (Q >= 4 for every supported shape)
+                    int n = xgn[i], p = xgp[i], q = xgq[i] + u;
+                    if (q >= Q) { q -= Q; if (++p >= P) { p = 0; ++n; } }
                     if (xvec_ok) {
                         long off = gather_off<0>(n, p, q, xr[i], xs[i], xc[i],
                                                  H, W, C, stride, pad, dil);
@@ -479,6 +506,11 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                 *reinterpret_cast<s16x4 *>(
                     xT_lds + swz<BKM>(j_local0 + e, xml[i] * 2)) = w4;
             }
+            // advance the quad base by BKM pixels for the next K-step
+            int q = xgq[i] + BKM, p = xgp[i], n = xgn[i];
+            while (q >= Q) { q -= Q; ++p; }
+            while (p >= P) { p -= P; ++n; }
+            xgq[i] = q; xgp[i] = p; xgn[i] = n;
         }
         __syncthreads();
 
@@ -594,8 +626,12 @@ void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
         int tiles_m = (M_total + BMv - 1) / BMv;
         int tiles_c = (Ncout + BC - 1) / BC;
         dim3 grid(tiles_m * tiles_c);
+        bool glds = (Cs % 8 == 0) && (K_total % 8 == 0);
+        int nsteps = (K_total + 63) / 64;
+        size_t shmem = (size_t)((glds && nsteps > 1) ? 2 : 1)
+                       * (BC + BMv) * 64 * 2;
         hipLaunchKernelGGL((sodkit::igemm_kernel<T, MODE, BC, BMv>), grid,
-                           dim3(256), 0, stream.stream(),
+                           dim3(256), shmem, stream.stream(),
                            reinterpret_cast<const T *>(src.data_ptr()),
                            reinterpret_cast<const T *>(w.data_ptr()), bias_ptr,
                            reinterpret_cast<T *>(out.data_ptr()), zpage,

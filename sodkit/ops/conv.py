@@ -60,5 +60,15 @@ def conv2d(x: torch.Tensor, w_krsc: torch.Tensor,
            bias: Optional[torch.Tensor] = None, stride: int = 1,
            padding: int = 0, dilation: int = 1) -> torch.Tensor:
     if x.is_cuda:
+        cin = x.shape[1]
+        if cin < 8 and cin % 8 != 0 and not x.requires_grad:
+            # stem-style inputs (RGB=3 / depth=1): pad channels to 8 so the
+            # kernel takes the vectorized/async-DMA path instead of the
+            # scalar gather (zero channels contribute nothing; w padding is
+            # differentiable so dw slices back automatically)
+            import torch.nn.functional as F
+            pad_c = 8 - cin
+            x = F.pad(x, (0, 0, 0, 0, 0, pad_c))
+            w_krsc = F.pad(w_krsc, (0, pad_c))
         return _ConvFn.apply(x, w_krsc, bias, stride, padding, dilation)
     return reference.conv2d(x, w_krsc, bias, stride, padding, dilation)

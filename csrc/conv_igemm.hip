@@ -37,7 +37,15 @@ namespace sodkit {
 // cycles as SQ_LDS_BANK_CONFLICT).
 template <int BK>
 SOD_DEVINL int swz(int row, int kbyte) {
-    return row * (BK * 2) + (kbyte ^ ((((row & 7) ^ ((row >> 3) & 7))) << 4));
+    if constexpr (BK == 128) {
+        // 256B row pitch: the row contributes nothing to the bank slot, so
+        // a 4-bit nibble (row&15)^((row>>4)&15) is needed — conflict-free
+        // for both the contiguous-16-row b128 reads and the stride-8-row
+        // b64 transpose writes
+        return row * 256 + (kbyte ^ ((((row & 15) ^ ((row >> 4) & 15))) << 4));
+    } else {
+        return row * (BK * 2) + (kbyte ^ ((((row & 7) ^ ((row >> 3) & 7))) << 4));
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -367,7 +375,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
 //   ds writes); the x tile gathers 8 consecutive output-pixels per lane
 //   (m-contiguous in LDS -> single b128 write).
 // ---------------------------------------------------------------------------
-template <typename T, int BCOUT, int BJ>
+template <typename T, int BCOUT, int BJ, int BKM>
 __global__ __launch_bounds__(256)
 void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                   float *__restrict__ partial,
@@ -377,7 +385,6 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                   int Kw,                          // = R*S*C (dw inner size)
                   int M_total,                     // = N*P*Q
                   int m_per_split, int tiles_c) {
-    constexpr int BKM = 64;   // m's per K-step
     constexpr int FC = BCOUT / 32;
     constexpr int FJ = BJ / 32;
 
@@ -726,18 +733,22 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
         using T = typename decltype(t_tag)::type;
         constexpr int BC = decltype(bc_tag)::value;
         constexpr int BJ = 128;
+        // the hot BCOUT=64 (decoder) variant runs BKM=128: half the
+        // barriers per MFMA at 48KB LDS (3 blocks/CU); BCOUT=128 keeps
+        // BKM=64 (64KB would drop it to 2 blocks/CU)
+        constexpr int BKM = (BC == 64) ? 128 : 64;
         int tiles_c = (Cout + BC - 1) / BC;
         int tiles_j = (Kw + BJ - 1) / BJ;
         int tiles = tiles_c * tiles_j;
         // fill the chip: ~2 blocks per CU (256 CUs) regardless of tile count
         int splitk = std::max(1, std::min(512, 512 / std::max(tiles, 1)));
-        splitk = (int)std::min<long>(splitk, (M + 63) / 64);
-        long m_per_split = ((M + splitk - 1) / splitk + 63) / 64 * 64;
+        splitk = (int)std::min<long>(splitk, (M + BKM - 1) / BKM);
+        long m_per_split = ((M + splitk - 1) / splitk + BKM - 1) / BKM * BKM;
         splitk = (int)((M + m_per_split - 1) / m_per_split);
         auto partial = torch::empty({splitk, (long)Cout * Kw},
                                     x.options().dtype(torch::kFloat32));
         dim3 grid(tiles, splitk);
-        hipLaunchKernelGGL((sodkit::wgrad_kernel<T, BC, BJ>), grid, dim3(256),
+        hipLaunchKernelGGL((sodkit::wgrad_kernel<T, BC, BJ, BKM>), grid, dim3(256),
                            0, stream.stream(),
                            reinterpret_cast<const T *>(x.data_ptr()),
                            reinterpret_cast<const T *>(dy.data_ptr()),

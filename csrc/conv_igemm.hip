@@ -99,7 +99,8 @@ SOD_DEVINL void glds16(const void *gsrc, char *lds_dst) {
 SOD_DEVINL int swz_nib(int row) { return ((row & 7) ^ ((row >> 3) & 7)) << 4; }
 
 template <typename T, int MODE, int BCOUT, int BM>
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(256, 3)   // force >= 3 waves/SIMD (171-reg
+                                       // natural allocation lands at 2)
 void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                   const float *__restrict__ bias, T *__restrict__ out,
                   const T *__restrict__ zero_page,
@@ -107,7 +108,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                   int P, int Q,                     // output spatial
                   int Ncout,                        // output channels
                   int R, int S, int stride, int pad, int dil,
-                  int M_total, int K_total, int tiles_m) {
+                  int M_total, int K_total, int tiles_m, int dbuf) {
     constexpr int BK = 64;
     constexpr int FC = BCOUT / 32;   // a-fragments per wave (2x2 wave grid)
     constexpr int FM = BM / 32;      // b-fragments per wave
@@ -145,42 +146,20 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     // W chunks: BCOUT*BK/8 8-elem chunks, 256 threads
     constexpr int WCH = BCOUT * BK / 8 / 256;       // chunks per thread
     constexpr int XCH = BM * BK / 8 / 256;
-    // chunk ch covers (row = ch / (BK/8), kv = ch % (BK/8))
-    int xm[XCH], xn[XCH], xp[XCH], xq[XCH];
-#pragma unroll
-    for (int i = 0; i < XCH; ++i) {
-        int ch = t + i * 256;
-        int m = tile_m + ch / (BK / 8);
-        xm[i] = m;
-        int mm = m < M_total ? m : 0;
-        xn[i] = mm / (P * Q);
-        int pq = mm % (P * Q);
-        xp[i] = pq / Q;
-        xq[i] = pq % Q;
-    }
     // --- glds lane geometry: the LDS image is written lane-linear, so the
     // tile swizzle moves onto the SOURCE address (inverse-swizzle per lane;
     // CDNA4 guide rule 21).  Usable when 16B global chunks are whole
     // (Cs % 8 == 0 and K_total % 8 == 0); OOB/padding lanes read a zero page.
     const bool glds_ok = vec_ok && wvec_ok;
-    int gw_row[GW], gw_kel[GW];        // W tile: row, k-elem offset in step
-    int gx_row[GX], gx_kel[GX];
+    // per-inst state kept minimal for occupancy: rows/k-offsets are cheap
+    // ALU recomputes; only the incremental pixel/tap state lives in regs
     int gx_n[GX], gx_p[GX], gx_q[GX];  // X tile: pixel geometry per row
     int gx_c[GX], gx_r[GX], gx_s[GX];  // incremental (r,s,c) per inst
-#pragma unroll
-    for (int i = 0; i < GW; ++i) {
-        int off = (wv * GW + i) * 1024 + lane * 16;
-        int row = off >> 7;
-        gw_row[i] = row;
-        gw_kel[i] = ((off & 127) ^ swz_nib(row)) >> 1;
-    }
 #pragma unroll
     for (int i = 0; i < GX; ++i) {
         int off = (wv * GX + i) * 1024 + lane * 16;
         int row = off >> 7;
-        gx_row[i] = row;
         int kel = ((off & 127) ^ swz_nib(row)) >> 1;
-        gx_kel[i] = kel;
         int rs = kel / Cs;
         gx_c[i] = kel % Cs;
         gx_r[i] = rs / S;
@@ -204,8 +183,10 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
         char *x_lds = w_lds + BCOUT * BK * 2;
 #pragma unroll
         for (int i = 0; i < GW; ++i) {
-            int cout = tile_c + gw_row[i];
-            int k = k0 + gw_kel[i];
+            int off = (wv * GW + i) * 1024 + lane * 16;
+            int row = off >> 7;
+            int cout = tile_c + row;
+            int k = k0 + (((off & 127) ^ swz_nib(row)) >> 1);
             const T *src = (cout < Ncout && k < K_total)
                                ? A_w + (long)cout * K_total + k
                                : zero_page;
@@ -213,8 +194,10 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
         }
 #pragma unroll
         for (int i = 0; i < GX; ++i) {
+            int row = ((wv * GX + i) * 1024 + lane * 16) >> 7;
             const T *src = zero_page;
-            if (tile_m + gx_row[i] < M_total && k0 + gx_kel[i] < K_total) {
+            // k < K_total  <=>  r < R (k = (r*S+s)*Cs+c, K = R*S*Cs exact)
+            if (tile_m + row < M_total && gx_r[i] < R) {
                 long off = gather_off<MODE>(gx_n[i], gx_p[i], gx_q[i],
                                             gx_r[i], gx_s[i], gx_c[i],
                                             Hs, Ws, Cs, stride, pad, dil);
@@ -255,17 +238,24 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
             *reinterpret_cast<s16x8 *>(w_lds + swz<BK>(row, (ch % (BK / 8)) * 16)) = v;
         }
         // ---- stage X tile: [BM][BK], im2col gather ----
+        // (fallback path: geometry recomputed per call so its registers
+        // never stay live alongside the glds path's state)
 #pragma unroll
         for (int i = 0; i < XCH; ++i) {
             int ch = t + i * 256;
             int row = ch / (BK / 8);
             int k = k0 + (ch % (BK / 8)) * 8;
+            int m = tile_m + row;
+            int mm = m < M_total ? m : 0;
+            int xn = mm / (P * Q);
+            int pq = mm % (P * Q);
+            int xp = pq / Q, xq = pq % Q;
             s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (xm[i] < M_total && k < K_total) {
+            if (m < M_total && k < K_total) {
                 if (vec_ok && (k % 8 == 0) && (k / Cs == (k + 7) / Cs)) {
                     int rs = k / Cs, c = k % Cs;
                     int r = rs / S, s = rs % S;
-                    long off = gather_off<MODE>(xn[i], xp[i], xq[i], r, s, c,
+                    long off = gather_off<MODE>(xn, xp, xq, r, s, c,
                                                 Hs, Ws, Cs, stride, pad, dil);
                     if (off >= 0)
                         v = *reinterpret_cast<const s16x8 *>(B_src + off);
@@ -276,7 +266,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                         if (ke < K_total) {
                             int rs = ke / Cs, c = ke % Cs;
                             int r = rs / S, s = rs % S;
-                            long off = gather_off<MODE>(xn[i], xp[i], xq[i], r, s, c,
+                            long off = gather_off<MODE>(xn, xp, xq, r, s, c,
                                                         Hs, Ws, Cs, stride, pad, dil);
                             if (off >= 0)
                                 v[e] = reinterpret_cast<const short *>(B_src)[off];
@@ -314,7 +304,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     };
 
     const int nsteps = (K_total + BK - 1) / BK;
-    if (glds_ok) {
+    if (glds_ok && dbuf) {
         // 2-phase double buffer: tile s+1's async staging issues BEFORE
         // tile s's MFMAs; one barrier per tile (it also drains the DMA)
         stage_glds(0, 0);
@@ -325,6 +315,15 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
             mfma_tile(cur);
             __syncthreads();
             cur ^= 1;
+        }
+    } else if (glds_ok) {
+        // single-buffer glds (half the LDS -> more blocks/CU; cross-block
+        // overlap instead of intra-block pipelining)
+        for (int s = 0; s < nsteps; ++s) {
+            stage_glds(0, s * BK);
+            __syncthreads();
+            mfma_tile(0);
+            __syncthreads();
         }
     } else {
         for (int s = 0; s < nsteps; ++s) {
@@ -635,15 +634,23 @@ void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
         dim3 grid(tiles_m * tiles_c);
         bool glds = (Cs % 8 == 0) && (K_total % 8 == 0);
         int nsteps = (K_total + 63) / 64;
-        size_t shmem = (size_t)((glds && nsteps > 1) ? 2 : 1)
-                       * (BC + BMv) * 64 * 2;
+        // 64KB double-buffer caps the largest variant at 2 blocks/CU; the
+        // single-buffer form trades intra-block pipelining for occupancy
+        // (policy from A/B measurement; SODKIT_IGEMM_SB=1 forces single)
+        bool dbuf = glds && nsteps > 1;
+        static const bool force_sb = []() {
+            const char *e = getenv("SODKIT_IGEMM_SB");
+            return e && e[0] == '1';
+        }();
+        if (force_sb) dbuf = false;
+        size_t shmem = (size_t)(dbuf ? 2 : 1) * (BC + BMv) * 64 * 2;
         hipLaunchKernelGGL((sodkit::igemm_kernel<T, MODE, BC, BMv>), grid,
                            dim3(256), shmem, stream.stream(),
                            reinterpret_cast<const T *>(src.data_ptr()),
                            reinterpret_cast<const T *>(w.data_ptr()), bias_ptr,
                            reinterpret_cast<T *>(out.data_ptr()), zpage,
                            Ns, Hs, Ws, Cs, P, Q, Ncout, R, S, stride, pad, dil,
-                           (int)M_total, K_total, tiles_m);
+                           (int)M_total, K_total, tiles_m, (int)dbuf);
     };
     using c64 = std::integral_constant<int, 64>;
     using c128 = std::integral_constant<int, 128>;

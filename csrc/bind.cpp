@@ -2,12 +2,14 @@
 #include <torch/extension.h>
 
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
-                         int64_t stride, int64_t pad, int64_t dil);
+                         int64_t stride, int64_t pad, int64_t dil, int64_t ld);
 torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w_t, int64_t H,
-                           int64_t W, int64_t stride, int64_t pad, int64_t dil);
+                           int64_t W, int64_t stride, int64_t pad, int64_t dil,
+                           int64_t lddy);
 torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
-                           int64_t S, int64_t stride, int64_t pad, int64_t dil);
-torch::Tensor channel_sum(torch::Tensor x);
+                           int64_t S, int64_t stride, int64_t pad, int64_t dil,
+                           int64_t ldx, int64_t lddy);
+torch::Tensor channel_sum(torch::Tensor x, int64_t ld);
 
 torch::Tensor bn_stats(torch::Tensor x);
 std::vector<torch::Tensor> bn_finalize(torch::Tensor stats, double n,

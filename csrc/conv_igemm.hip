@@ -55,14 +55,14 @@ SOD_DEVINL int swz(int row, int kbyte) {
 // ---------------------------------------------------------------------------
 template <int MODE>
 SOD_DEVINL long gather_off(int n, int p, int q, int r, int s, int c,
-                           int Hs, int Ws, int Cs,
+                           int Hs, int Ws, int Cs, int ldc,
                            int stride, int pad, int dil) {
     if (MODE == 0) {
         int ih = p * stride + r * dil - pad;
         int iw = q * stride + s * dil - pad;
         if ((unsigned)ih >= (unsigned)Hs || (unsigned)iw >= (unsigned)Ws)
             return -1;
-        return (((long)n * Hs + ih) * Ws + iw) * Cs + c;
+        return (((long)n * Hs + ih) * Ws + iw) * ldc + c;
     } else if (stride == 1) {
         // fast path: most layers are stride 1 and the generic path's
         // runtime %/÷ by stride costs real VALU in the staging loop
@@ -70,14 +70,14 @@ SOD_DEVINL long gather_off(int n, int p, int q, int r, int s, int c,
         int ow = q + pad - s * dil;
         if ((unsigned)oh >= (unsigned)Hs || (unsigned)ow >= (unsigned)Ws)
             return -1;
-        return (((long)n * Hs + oh) * Ws + ow) * Cs + c;
+        return (((long)n * Hs + oh) * Ws + ow) * ldc + c;
     } else {
         int hh = p + pad - r * dil;
         int ww = q + pad - s * dil;
         if (hh < 0 || ww < 0 || hh % stride || ww % stride) return -1;
         int oh = hh / stride, ow = ww / stride;
         if (oh >= Hs || ow >= Ws) return -1;
-        return (((long)n * Hs + oh) * Ws + ow) * Cs + c;
+        return (((long)n * Hs + oh) * Ws + ow) * ldc + c;
     }
 }
 
@@ -105,6 +105,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                   const float *__restrict__ bias, T *__restrict__ out,
                   const T *__restrict__ zero_page,
                   int Ns, int Hs, int Ws, int Cs,   // gather-source dims
+                  int ld_src,                       // pixel pitch of B_src
                   int P, int Q,                     // output spatial
                   int Ncout,                        // output channels
                   int R, int S, int stride, int pad, int dil,
@@ -139,7 +140,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     const int wv = t >> 6;                  // 4 waves: 2x2
     const int wv_c = wv >> 1, wv_m = wv & 1;
 
-    const bool vec_ok = (Cs % 8 == 0);
+    const bool vec_ok = (Cs % 8 == 0) && (ld_src % 8 == 0);
     const bool wvec_ok = (K_total % 8 == 0);
 
     // --- per-thread staging geometry (fixed across K-steps) ---
@@ -200,7 +201,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
             if (tile_m + row < M_total && gx_r[i] < R) {
                 long off = gather_off<MODE>(gx_n[i], gx_p[i], gx_q[i],
                                             gx_r[i], gx_s[i], gx_c[i],
-                                            Hs, Ws, Cs, stride, pad, dil);
+                                            Hs, Ws, Cs, ld_src, stride, pad, dil);
                 if (off >= 0) src = B_src + off;
             }
             glds16(src, x_lds + (wv * GX + i) * 1024);
@@ -256,7 +257,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                     int rs = k / Cs, c = k % Cs;
                     int r = rs / S, s = rs % S;
                     long off = gather_off<MODE>(xn, xp, xq, r, s, c,
-                                                Hs, Ws, Cs, stride, pad, dil);
+                                                Hs, Ws, Cs, ld_src, stride, pad, dil);
                     if (off >= 0)
                         v = *reinterpret_cast<const s16x8 *>(B_src + off);
                 } else {
@@ -267,7 +268,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                             int rs = ke / Cs, c = ke % Cs;
                             int r = rs / S, s = rs % S;
                             long off = gather_off<MODE>(xn, xp, xq, r, s, c,
-                                                        Hs, Ws, Cs, stride, pad, dil);
+                                                        Hs, Ws, Cs, ld_src, stride, pad, dil);
                             if (off >= 0)
                                 v[e] = reinterpret_cast<const short *>(B_src)[off];
                         }
@@ -379,6 +380,7 @@ __global__ __launch_bounds__(256)
 void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                   float *__restrict__ partial,
                   int N, int H, int W, int C,     // x dims
+                  int ldx, int lddy,              // pixel pitches (slices)
                   int P, int Q, int Cout,         // dy dims
                   int R, int S, int stride, int pad, int dil,
                   int Kw,                          // = R*S*C (dw inner size)
@@ -402,8 +404,8 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     const int m_begin = z * m_per_split;
     const int m_end = min(m_begin + m_per_split, M_total);
 
-    const bool dyvec_ok = (Cout % 8 == 0);
-    const bool xvec_ok = (C % 8 == 0);
+    const bool dyvec_ok = (Cout % 8 == 0) && (lddy % 8 == 0);
+    const bool xvec_ok = (C % 8 == 0) && (ldx % 8 == 0);
 
     // per-thread x-chunk geometry: chunk ch -> (jv8 = ch % (BJ/8),
     // m_quad = ch / (BJ/8)); 8 consecutive j = 8 consecutive cin of one
@@ -452,7 +454,7 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                 v[u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
                 int m = m0 + m_local + u;
                 if (m < m_end && cout < Cout) {
-                    const T *src = dy + (long)m * Cout + cout;
+                    const T *src = dy + (long)m * lddy + cout;
                     if (dyvec_ok && cout + 7 < Cout) {
                         v[u] = *reinterpret_cast<const s16x8 *>(src);
                     } else {
@@ -487,7 +489,7 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                     if (q >= Q) { q -= Q; if (++p >= P) { p = 0; ++n; } }
                     if (xvec_ok) {
                         long off = gather_off<0>(n, p, q, xr[i], xs[i], xc[i],
-                                                 H, W, C, stride, pad, dil);
+                                                 H, W, C, ldx, stride, pad, dil);
                         if (off >= 0)
                             v[u] = *reinterpret_cast<const s16x8 *>(x + off);
                     } else {
@@ -498,7 +500,7 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                                 int rs = j / C, c = j % C;
                                 long off = gather_off<0>(n, p, q, rs / S,
                                                          rs % S, c,
-                                                         H, W, C, stride, pad, dil);
+                                                         H, W, C, ldx, stride, pad, dil);
                                 if (off >= 0)
                                     v[u][e] = reinterpret_cast<const short *>(x)[off];
                             }
@@ -583,13 +585,13 @@ __global__ void splitk_reduce_kernel(const float *__restrict__ partial,
 // channel sum of an NHWC tensor -> fp32 [C]  (conv bias grad; SURVEY K-misc)
 template <typename T>
 __global__ void channel_sum_kernel(const T *__restrict__ x, float *__restrict__ out,
-                                   long M, int C) {
+                                   long M, int C, int ld) {
     // one block per channel-chunk; grid-stride over rows
     int c = blockIdx.y * blockDim.x + threadIdx.x;
     if (c >= C) return;
     float s = 0.f;
     for (long m = blockIdx.x; m < M; m += gridDim.x)
-        s += ElemTraits<T>::to_f32(x[m * C + c]);
+        s += ElemTraits<T>::to_f32(x[m * ld + c]);
     atomicAdd(&out[c], s);
 }
 
@@ -618,7 +620,8 @@ static void *zero_page_ptr(const torch::Tensor &like) {
 template <typename scalar_t, int MODE>
 void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
                   const torch::Tensor &bias, torch::Tensor &out,
-                  int Ns, int Hs, int Ws, int Cs, int P, int Q, int Ncout,
+                  int Ns, int Hs, int Ws, int Cs, int ld_src,
+                  int P, int Q, int Ncout,
                   int R, int S, int stride, int pad, int dil,
                   long M_total, int K_total) {
     using T = scalar_t;
@@ -649,7 +652,8 @@ void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
                            reinterpret_cast<const T *>(src.data_ptr()),
                            reinterpret_cast<const T *>(w.data_ptr()), bias_ptr,
                            reinterpret_cast<T *>(out.data_ptr()), zpage,
-                           Ns, Hs, Ws, Cs, P, Q, Ncout, R, S, stride, pad, dil,
+                           Ns, Hs, Ws, Cs, ld_src, P, Q, Ncout, R, S,
+                           stride, pad, dil,
                            (int)M_total, K_total, tiles_m, (int)dbuf);
     };
     using c64 = std::integral_constant<int, 64>;
@@ -669,13 +673,12 @@ void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
 }  // namespace
 
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
-                         int64_t stride, int64_t pad, int64_t dil) {
+                         int64_t stride, int64_t pad, int64_t dil,
+                         int64_t ld) {
     CHECK_GPU(x); CHECK_GPU(w);
     TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
     int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
-    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) ||
-                (C == 1 && x.is_contiguous()),
-                "conv2d_fwd needs channels_last input");
+    TORCH_CHECK(ld >= C, "bad channel pitch");
     int Cout = w.size(0), R = w.size(1), S = w.size(2);
     TORCH_CHECK(w.size(3) == C, "weight Cin mismatch");
     TORCH_CHECK(w.is_contiguous());
@@ -690,22 +693,22 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
     AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_fwd",
         AT_DISPATCH_CASE(at::kBFloat16, [&] {
             launch_igemm<__hip_bfloat16, 0>(x, w, bias32, out, N, H, W_, C,
-                                            P, Q, Cout, R, S, stride, pad, dil, M, K);
+                                            (int)ld, P, Q, Cout, R, S,
+                                            stride, pad, dil, M, K);
         })
         AT_DISPATCH_CASE(at::kHalf, [&] {
-            launch_igemm<__half, 0>(x, w, bias32, out, N, H, W_, C,
+            launch_igemm<__half, 0>(x, w, bias32, out, N, H, W_, C, (int)ld,
                                     P, Q, Cout, R, S, stride, pad, dil, M, K);
         }));
     return out;
 }
 
 torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w_t, int64_t H,
-                           int64_t W_, int64_t stride, int64_t pad, int64_t dil) {
+                           int64_t W_, int64_t stride, int64_t pad,
+                           int64_t dil, int64_t lddy) {
     CHECK_GPU(dy); CHECK_GPU(w_t);
     int N = dy.size(0), Cout = dy.size(1), P = dy.size(2), Q = dy.size(3);
-    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) ||
-                (Cout == 1 && dy.is_contiguous()),
-                "conv2d_dgrad needs channels_last dy");
+    TORCH_CHECK(lddy >= Cout, "bad channel pitch");
     // w_t: [Cin][R][S][Cout]
     int Cin = w_t.size(0), R = w_t.size(1), S = w_t.size(2);
     TORCH_CHECK(w_t.size(3) == Cout && w_t.is_contiguous());
@@ -717,17 +720,20 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w_t, int64_t H,
     AT_DISPATCH_SWITCH(dy.scalar_type(), "conv2d_dgrad",
         AT_DISPATCH_CASE(at::kBFloat16, [&] {
             launch_igemm<__hip_bfloat16, 1>(dy, w_t, nobias, dx, N, P, Q, Cout,
-                                            H, W_, Cin, R, S, stride, pad, dil, M, K);
+                                            (int)lddy, H, W_, Cin, R, S,
+                                            stride, pad, dil, M, K);
         })
         AT_DISPATCH_CASE(at::kHalf, [&] {
             launch_igemm<__half, 1>(dy, w_t, nobias, dx, N, P, Q, Cout,
-                                    H, W_, Cin, R, S, stride, pad, dil, M, K);
+                                    (int)lddy, H, W_, Cin, R, S,
+                                    stride, pad, dil, M, K);
         }));
     return dx;
 }
 
 torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
-                           int64_t S, int64_t stride, int64_t pad, int64_t dil) {
+                           int64_t S, int64_t stride, int64_t pad,
+                           int64_t dil, int64_t ldx, int64_t lddy) {
     CHECK_GPU(x); CHECK_GPU(dy);
     int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
     int Cout = dy.size(1), P = dy.size(2), Q = dy.size(3);
@@ -760,7 +766,8 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
                            reinterpret_cast<const T *>(x.data_ptr()),
                            reinterpret_cast<const T *>(dy.data_ptr()),
                            partial.data_ptr<float>(),
-                           N, H, W_, C, P, Q, Cout, (int)R, (int)S,
+                           N, H, W_, C, (int)ldx, (int)lddy,
+                           P, Q, Cout, (int)R, (int)S,
                            (int)stride, (int)pad, (int)dil, Kw, (int)M,
                            (int)m_per_split, tiles_c);
         auto dw = torch::empty({Cout, R, S, C}, x.options());
@@ -789,7 +796,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
     return dw;
 }
 
-torch::Tensor channel_sum(torch::Tensor x) {
+torch::Tensor channel_sum(torch::Tensor x, int64_t ld) {
     CHECK_GPU(x);
     int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
     long M = (long)N * H * W_;
@@ -801,19 +808,19 @@ torch::Tensor channel_sum(torch::Tensor x) {
             hipLaunchKernelGGL((sodkit::channel_sum_kernel<__hip_bfloat16>),
                                grid, dim3(256), 0, stream.stream(),
                                reinterpret_cast<const __hip_bfloat16 *>(x.data_ptr()),
-                               out.data_ptr<float>(), M, C);
+                               out.data_ptr<float>(), M, C, (int)ld);
         })
         AT_DISPATCH_CASE(at::kHalf, [&] {
             hipLaunchKernelGGL((sodkit::channel_sum_kernel<__half>),
                                grid, dim3(256), 0, stream.stream(),
                                reinterpret_cast<const __half *>(x.data_ptr()),
-                               out.data_ptr<float>(), M, C);
+                               out.data_ptr<float>(), M, C, (int)ld);
         })
         AT_DISPATCH_CASE(at::kFloat, [&] {
             hipLaunchKernelGGL((sodkit::channel_sum_kernel<float>),
                                grid, dim3(256), 0, stream.stream(),
                                reinterpret_cast<const float *>(x.data_ptr()),
-                               out.data_ptr<float>(), M, C);
+                               out.data_ptr<float>(), M, C, (int)ld);
         }));
     return out;
 }

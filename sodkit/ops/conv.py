@@ -23,6 +23,21 @@ from ._backend import ext
 _CL = torch.channels_last
 
 
+def _cl_ld(x: torch.Tensor):
+    """Return (tensor, channel pitch).  A channel-slice of a channels_last
+    tensor (torch.split / cat-backward narrow views) keeps NHWC pixel layout
+    with pitch = parent C; the kernels take that pitch directly, so the
+    slice costs no copy."""
+    if x.is_contiguous(memory_format=_CL) or (x.shape[1] == 1 and x.is_contiguous()):
+        return x, x.shape[1]
+    st = x.stride()
+    if (x.dim() == 4 and st[1] == 1 and st[2] == x.shape[3] * st[3]
+            and st[0] == x.shape[2] * st[2] and st[3] >= x.shape[1]):
+        return x, st[3]
+    x = x.contiguous(memory_format=_CL)
+    return x, x.shape[1]
+
+
 def _cl(x: torch.Tensor) -> torch.Tensor:
     return x.contiguous(memory_format=_CL)
 
@@ -30,29 +45,30 @@ def _cl(x: torch.Tensor) -> torch.Tensor:
 class _ConvFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, bias, stride, padding, dilation):
-        x = _cl(x)
+        x, ldx = _cl_ld(x)
         y = ext().conv2d_fwd(x, w, bias if bias is not None else torch.Tensor(),
-                             stride, padding, dilation)
+                             stride, padding, dilation, ldx)
         ctx.save_for_backward(x, w)
-        ctx.meta = (stride, padding, dilation, bias is not None)
+        ctx.meta = (stride, padding, dilation, bias is not None, ldx)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x, w = ctx.saved_tensors
-        stride, padding, dilation, has_bias = ctx.meta
-        dy = _cl(dy)
+        stride, padding, dilation, has_bias, ldx = ctx.meta
+        dy, lddy = _cl_ld(dy)
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             # dgrad A-operand layout [Cin, R, S, Cout] (a small permuted copy)
             w_t = w.permute(3, 1, 2, 0).contiguous()
             dx = ext().conv2d_dgrad(dy, w_t, x.shape[2], x.shape[3],
-                                    stride, padding, dilation)
+                                    stride, padding, dilation, lddy)
         if ctx.needs_input_grad[1]:
             dw = ext().conv2d_wgrad(x, dy, w.shape[1], w.shape[2],
-                                    stride, padding, dilation).to(w.dtype)
+                                    stride, padding, dilation, ldx,
+                                    lddy).to(w.dtype)
         if has_bias and ctx.needs_input_grad[2]:
-            db = ext().channel_sum(dy).to(w.dtype)
+            db = ext().channel_sum(dy, lddy).to(w.dtype)
         return dx, dw, db, None, None, None
 
 

@@ -49,11 +49,11 @@ def main():
         K = k * k * Cin
         flop = 2.0 * M * K * Cout
         nob = torch.Tensor()
-        t_f = bench(lambda: _C.conv2d_fwd(x, w, nob, st, pad, dil))
-        y = _C.conv2d_fwd(x, w, nob, st, pad, dil)
+        t_f = bench(lambda: _C.conv2d_fwd(x, w, nob, st, pad, dil, Cin))
+        y = _C.conv2d_fwd(x, w, nob, st, pad, dil, Cin)
         w_t = w.permute(3, 1, 2, 0).contiguous()
-        t_d = bench(lambda: _C.conv2d_dgrad(y, w_t, H, H, st, pad, dil))
-        t_w = bench(lambda: _C.conv2d_wgrad(x, y, k, k, st, pad, dil))
+        t_d = bench(lambda: _C.conv2d_dgrad(y, w_t, H, H, st, pad, dil, Cout))
+        t_w = bench(lambda: _C.conv2d_wgrad(x, y, k, k, st, pad, dil, Cin, Cout))
         print(f"{name:>12} {M:>8} {K:>6} {Cout:>6} | "
               f"{flop / t_f / 1e12:7.0f} {flop / t_d / 1e12:8.0f} "
               f"{flop / t_w / 1e12:8.0f}")

@@ -354,7 +354,7 @@ def test_bilinear_odd_channels():
 def test_channel_sum_matches():
     from sodkit import _C
     x = _cl(torch.randn(3, 24, 10, 10, device="cuda").to(torch.bfloat16))
-    got = _C.channel_sum(x)
+    got = _C.channel_sum(x, x.shape[1])
     want = x.float().sum(dim=[0, 2, 3])
     close(got, want, rtol=1e-2, atol=1e-2, what="channel_sum")
 
@@ -403,21 +403,23 @@ def test_conv_backward_on_slice_view():
     from sodkit import ops
     torch.manual_seed(14)
     xfull = _cl(torch.randn(2, 64, 16, 16, device="cuda").to(torch.bfloat16))
+    w0 = (torch.randn(16, 3, 3, 32, device="cuda") * 0.2).to(torch.bfloat16)
+    gy0 = torch.randn(2, 16, 16, 16, device="cuda").to(torch.bfloat16)
 
-    def run(make_input):
-        x = make_input().requires_grad_()
-        w = (torch.randn(16, 3, 3, 32, device="cuda") * 0.2) \
-            .to(torch.bfloat16).requires_grad_()
-        torch.manual_seed(15)
-        parts = torch.split(x, 32, dim=1)
-        y = ops.conv2d(parts[1], w, None, 1, 1, 1)
-        gy = torch.randn_like(y)
-        y.backward(gy)
-        return x.grad.clone(), w.grad.clone()
+    # A: conv on the split VIEW (ld path)
+    x = xfull.detach().clone().requires_grad_()
+    w = w0.clone().requires_grad_()
+    y = ops.conv2d(torch.split(x, 32, dim=1)[1], w, None, 1, 1, 1)
+    y.backward(_cl(gy0))
+    dx_v = x.grad[:, 32:64].clone()
+    dw_v = w.grad.clone()
 
-    dx_v, dw_v = run(lambda: xfull.detach().clone())
-    # second run forces contiguity through a wrapper that copies
-    dx_c, dw_c = run(lambda: xfull.detach().clone().contiguous(
-        memory_format=torch.channels_last))
-    close(dx_v, dx_c, rtol=1e-3, atol=1e-3, what="slice dx")
-    close(dw_v, dw_c, rtol=1e-3, atol=1e-3, what="slice dw")
+    # B: conv on an explicit contiguous copy of the slice
+    x2 = xfull[:, 32:64].detach().clone().contiguous(
+        memory_format=torch.channels_last).requires_grad_()
+    w2 = w0.clone().requires_grad_()
+    y2 = ops.conv2d(x2, w2, None, 1, 1, 1)
+    y2.backward(_cl(gy0))
+    assert torch.equal(y.detach(), y2.detach())
+    close(dx_v, x2.grad, rtol=1e-3, atol=1e-3, what="slice dx")
+    close(dw_v, w2.grad, rtol=1e-3, atol=1e-3, what="slice dw")

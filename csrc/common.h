@@ -32,7 +32,15 @@ template <> struct ElemTraits<__hip_bfloat16> {
 };
 template <> struct ElemTraits<__half> {
     SOD_DEVINL static float to_f32(__half v) { return __half2float(v); }
-    SOD_DEVINL static __half from_f32(float v) { return __float2half(v); }
+    SOD_DEVINL static __half from_f32(float v) {
+        // saturate to the fp16 range (NaN-preserving: a NaN fails both
+        // comparisons).  Early-training eval passes run activations through
+        // unconverged BN running stats; letting them become inf poisons the
+        // whole map, saturation keeps eval finite (bf16 needs none).
+        if (v > 65504.f) v = 65504.f;
+        else if (v < -65504.f) v = -65504.f;
+        return __float2half(v);
+    }
 };
 template <> struct ElemTraits<float> {
     SOD_DEVINL static float to_f32(float v) { return v; }

@@ -121,7 +121,9 @@ __global__ void bn_reduce_finalize_kernel(const float *__restrict__ partial,
     float sum = ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
     float ssq = ((q[0] + q[1]) + (q[2] + q[3])) + ((q[4] + q[5]) + (q[6] + q[7]));
     float mean = sum / n;
-    float var = ssq / n - mean * mean;
+    // two-pass variance can go (slightly) negative by cancellation at
+    // large-mean fp16 scales; rsqrt(negative) would poison the tensor
+    float var = fmaxf(ssq / n - mean * mean, 0.f);
     float unbiased = var * (n / fmaxf(n - 1.f, 1.f));
     running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
     running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
@@ -158,13 +160,13 @@ __global__ void bn_finalize_kernel(const float *__restrict__ stats, float n,
     float mean, var;
     if (training) {
         mean = stats[c] / n;
-        var = stats[C + c] / n - mean * mean;
+        var = fmaxf(stats[C + c] / n - mean * mean, 0.f);
         float unbiased = var * (n / fmaxf(n - 1.f, 1.f));
         running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
         running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
     } else {
         mean = running_mean[c];
-        var = running_var[c];
+        var = fmaxf(running_var[c], 0.f);
     }
     mean_out[c] = mean;
     invstd_out[c] = rsqrtf(var + eps);

@@ -80,14 +80,15 @@ def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
             s, ss, cnt = packed[: s.numel()], packed[s.numel():-1], packed[-1:]
         n_tot = cnt.item()
         mean = s / n_tot
-        var = ss / n_tot - mean * mean          # biased (population) variance
+        # clamp: the two-pass form can go slightly negative by cancellation
+        var = (ss / n_tot - mean * mean).clamp_min(0.0)
         with torch.no_grad():
             unbiased = var * (n_tot / max(n_tot - 1.0, 1.0))
             running_mean.mul_(1 - momentum).add_(mean.detach(), alpha=momentum)
             running_var.mul_(1 - momentum).add_(unbiased.detach(), alpha=momentum)
     else:
         mean, var = running_mean, running_var
-    invstd = torch.rsqrt(var + eps)
+    invstd = torch.rsqrt(var.clamp_min(0.0) + eps)
     y = (xf - mean.view(1, -1, 1, 1)) * (invstd * gamma).view(1, -1, 1, 1) \
         + beta.view(1, -1, 1, 1)
     if residual is not None:

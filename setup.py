@@ -13,6 +13,26 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
+
+# torch's hipify step compiles copied csrc/*_hip.hip files whose dep files
+# do not track csrc/common.h: a header edit would silently keep stale
+# objects.  Invalidate explicitly.
+def _invalidate_stale_objects():
+    import glob
+    hdr = os.path.join(ROOT, "csrc", "common.h")
+    if not os.path.exists(hdr):
+        return
+    hmt = os.path.getmtime(hdr)
+    for o in glob.glob(os.path.join(ROOT, "build", "temp*", "csrc", "*.o")):
+        if os.path.getmtime(o) < hmt:
+            os.remove(o)
+    for c in glob.glob(os.path.join(ROOT, "csrc", "*_hip.hip")):
+        if os.path.getmtime(c) < hmt:
+            os.remove(c)
+
+
+_invalidate_stale_objects()
+
 SRC = [
     "csrc/bind.cpp",
     "csrc/conv_igemm.hip",

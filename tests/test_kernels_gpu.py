@@ -423,3 +423,30 @@ def test_conv_backward_on_slice_view():
     assert torch.equal(y.detach(), y2.detach())
     close(dx_v, x2.grad, rtol=1e-3, atol=1e-3, what="slice dx")
     close(dw_v, w2.grad, rtol=1e-3, atol=1e-3, what="slice dw")
+
+
+def test_fp16_epilogue_saturates():
+    """fp32 accumulators past the fp16 range must saturate (NaN-preserving),
+    not become inf — a header-dependency build bug once shipped without this
+    and early-training fp16 eval passes emitted inf maps."""
+    from sodkit import _C
+    x = torch.ones(1, 256, 8, 8, device="cuda", dtype=torch.float16) \
+        .contiguous(memory_format=CL)
+    for sign in (1.0, -1.0):
+        w = torch.full((8, 1, 1, 256), sign * 273.4, device="cuda",
+                       dtype=torch.float16)
+        y = _C.conv2d_fwd(x, w, torch.Tensor(), 1, 0, 1, 256)
+        assert torch.isfinite(y.float()).all()
+        assert y[0, 0, 0, 0].item() == sign * 65504.0
+
+
+def test_conv_k_tail_not_multiple_of_64():
+    """K = 9*56 = 504 (K % 64 != 0): the tail K-step's zero-fill path."""
+    from sodkit import ops
+    torch.manual_seed(21)
+    x = _cl(torch.randn(2, 56, 24, 24, device="cuda").to(torch.bfloat16))
+    w = (torch.randn(56, 3, 3, 56, device="cuda") * 0.1).to(torch.bfloat16)
+    y = ops.conv2d(x, w, None, 2, 1, 1)
+    yr = F.conv2d(x.float(), w.float().permute(0, 3, 1, 2), None,
+                  stride=2, padding=1)
+    close(y, yr, what="K=504 tail")

@@ -200,11 +200,17 @@ class Solver:
                 ips_meter.update(ips)
                 if (i + 1) % cfg.log_interval == 0 or i + 1 == len(loader):
                     loss_meter.update(loss.item())
+                    comm = ""
+                    if self.reducer.stats_mode and self.distributed:
+                        st = self.reducer.pop_stats()
+                        comm = (f" comm[buckets {st['launches']} spread "
+                                f"{st['launch_spread_ms']:.1f}ms wait "
+                                f"{st['wait_ms']:.1f}ms]")
                     self.log.info(
                         f"ep {epoch} it {i + 1}/{len(loader)} "
                         f"loss {loss_meter.val:.4f} (avg {loss_meter.avg:.4f}) "
                         f"lr {self._lr_at(self.iter_count - 1):.5f} "
-                        f"imgs/s {ips_meter.avg:.1f}")
+                        f"imgs/s {ips_meter.avg:.1f}{comm}")
                     if self.tb is not None:
                         self.tb.scalar("train/loss", loss_meter.val, self.iter_count)
                         self.tb.scalar("train/lr", self._lr_at(self.iter_count - 1),

@@ -99,6 +99,9 @@ class GradReducer:
 
         self.comm_stream = (torch.cuda.Stream()
                             if self.overlap and _world(self.pg) > 1 else None)
+        # per-bucket comm-overlap stats: SODKIT_COMM_STATS=1
+        self.stats_mode = os.environ.get("SODKIT_COMM_STATS", "0") == "1"
+        self.stats = {"launches": 0, "launch_t": [], "wait_ms": 0.0}
         self._hooks = []
         for p in params:
             self._hooks.append(
@@ -123,6 +126,10 @@ class GradReducer:
     def _launch(self, b: _Bucket) -> None:
         if _world(self.pg) <= 1:
             return
+        if self.stats_mode:
+            import time
+            self.stats["launches"] += 1
+            self.stats["launch_t"].append(time.perf_counter())
         if self.assert_mode:
             before = b.flat.float().sum().item()
         if self.comm_stream is not None:
@@ -154,7 +161,13 @@ class GradReducer:
                 # zero, all-reduce anyway so ranks agree
                 self._launch(b)
             if b.work is not None:
-                b.work.wait()  # current stream waits on NCCL/gloo completion
+                if self.stats_mode:
+                    import time
+                    t0 = time.perf_counter()
+                    b.work.wait()
+                    self.stats["wait_ms"] += (time.perf_counter() - t0) * 1e3
+                else:
+                    b.work.wait()  # current stream waits for the collective
                 b.work = None
             b.arrived = 0
 
@@ -162,6 +175,17 @@ class GradReducer:
         for b in self.buckets:
             b.flat.zero_()
             b.arrived = 0
+
+    def pop_stats(self) -> dict:
+        """Comm-overlap summary since the last call (stats mode only):
+        bucket launch spread within backward + host-side wait time."""
+        st = self.stats
+        out = {"launches": st["launches"],
+               "launch_spread_ms": ((max(st["launch_t"]) - min(st["launch_t"]))
+                                    * 1e3 if len(st["launch_t"]) > 1 else 0.0),
+               "wait_ms": st["wait_ms"]}
+        self.stats = {"launches": 0, "launch_t": [], "wait_ms": 0.0}
+        return out
 
     @property
     def grad_scale(self) -> float:

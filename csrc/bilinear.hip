@@ -115,10 +115,15 @@ __global__ void bilinear_bwd_vec_kernel(const T *__restrict__ dy,
         long t2 = pix / Wi;
         int ih = (int)(t2 % Hi);
         int n = (int)(t2 / Hi);
-        int oh_lo = max(0, (int)floorf(((float)ih - 0.5f) / sh - 0.5f) - 1);
-        int oh_hi = min(Ho - 1, (int)ceilf(((float)ih + 1.5f) / sh - 0.5f) + 1);
-        int ow_lo = max(0, (int)floorf(((float)iw - 0.5f) / sw - 0.5f) - 1);
-        int ow_hi = min(Wo - 1, (int)ceilf(((float)iw + 1.5f) / sw - 0.5f) + 1);
+        // exact contribution range (the per-tap weight==0 guard below keeps
+        // one element of float-rounding slack); the clamped source index
+        // means the LAST input row/col receives from every larger oh/ow
+        int oh_lo = max(0, (int)floorf(((float)ih - 0.5f) / sh - 0.5f));
+        int oh_hi = (ih == Hi - 1) ? (Ho - 1)
+                  : min(Ho - 1, (int)ceilf(((float)ih + 1.5f) / sh - 0.5f));
+        int ow_lo = max(0, (int)floorf(((float)iw - 0.5f) / sw - 0.5f));
+        int ow_hi = (iw == Wi - 1) ? (Wo - 1)
+                  : min(Wo - 1, (int)ceilf(((float)iw + 1.5f) / sw - 0.5f));
         float acc[8];
 #pragma unroll
         for (int e = 0; e < 8; ++e) acc[e] = 0.f;
@@ -170,11 +175,12 @@ __global__ void bilinear_bwd_kernel(const T *__restrict__ dy,
         long t2 = pix / Wi;
         int ih = (int)(t2 % Hi);
         int n = (int)(t2 / Hi);
-        // conservative output ranges whose source index can hit ih/iw
-        int oh_lo = max(0, (int)floorf(((float)ih - 1.f + 0.5f) / sh - 0.5f) - 1);
-        int oh_hi = min(Ho - 1, (int)ceilf(((float)ih + 1.f + 0.5f) / sh - 0.5f) + 1);
-        int ow_lo = max(0, (int)floorf(((float)iw - 1.f + 0.5f) / sw - 0.5f) - 1);
-        int ow_hi = min(Wo - 1, (int)ceilf(((float)iw + 1.f + 0.5f) / sw - 0.5f) + 1);
+        int oh_lo = max(0, (int)floorf(((float)ih - 0.5f) / sh - 0.5f));
+        int oh_hi = (ih == Hi - 1) ? (Ho - 1)
+                  : min(Ho - 1, (int)ceilf(((float)ih + 1.5f) / sh - 0.5f));
+        int ow_lo = max(0, (int)floorf(((float)iw - 0.5f) / sw - 0.5f));
+        int ow_hi = (iw == Wi - 1) ? (Wo - 1)
+                  : min(Wo - 1, (int)ceilf(((float)iw + 1.5f) / sw - 0.5f));
         float acc = 0.f;
         const T *dybase = dy + (long)n * Ho * Wo * C + c;
         for (int oh = oh_lo; oh <= oh_hi; ++oh) {

@@ -23,7 +23,7 @@ from ..losses import build_loss
 from ..metrics import MAEMeter, PRFMeter
 from ..models import build_model, convert_precision, set_syncbn
 from ..parallel import GradReducer, setup, get_rank, get_world_size
-from ..utils import AvgMeter, Timer, get_logger, seed_everything
+from ..utils import AvgMeter, get_logger, seed_everything
 from ..utils.png import write_gray_png
 from ..utils.recorder import ResultsRecorder, TBRecorder
 from .checkpoint import exp_paths, load_checkpoint, make_exp_tree, save_checkpoint

@@ -15,7 +15,6 @@ from __future__ import annotations
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from ..ops import bilinear2d, dynamic_filter, upsample_add
 from .blocks import ConvBNReLU, SodConv2d

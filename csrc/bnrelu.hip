@@ -77,24 +77,44 @@ __global__ void bn_stats_vec_kernel(const T *__restrict__ x,
     }
 }
 
-// partial[nb][R] -> out[R] (deterministic fixed-order sum; 8 independent
-// accumulators keep 8 loads in flight — a serial chain is latency-bound)
+// partial[nb][R] -> out[R] (deterministic fixed-order sum).  The slab
+// dimension is split across threadIdx.y and combined with a fixed-order
+// LDS tree: a flat R-threads grid is only R/256 blocks (1-2 for BN-sized
+// R), which leaves one CU chewing through nb*R floats at latency — the
+// y-split is ~TY x more parallelism at identical determinism.  8
+// independent accumulators keep 8 loads in flight per thread.
+template <int TX, int TY>
 __global__ void slab_reduce_kernel(const float *__restrict__ partial,
                                    float *__restrict__ out, int nb, long R) {
-    long r = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    if (r >= R) return;
+    __shared__ float red[TY][TX];
+    long r = (long)blockIdx.x * TX + threadIdx.x;
     float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-    int b = 0;
-    for (; b + 8 <= nb; b += 8)
+    if (r < R) {
+        int b = threadIdx.y;
+        for (; b + 7 * TY < nb; b += 8 * TY)
 #pragma unroll
-        for (int u = 0; u < 8; ++u) a[u] += partial[(long)(b + u) * R + r];
-    for (; b < nb; ++b) a[0] += partial[(long)b * R + r];
-    out[r] = ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
+            for (int u = 0; u < 8; ++u)
+                a[u] += partial[(long)(b + u * TY) * R + r];
+        for (; b < nb; b += TY) a[0] += partial[(long)b * R + r];
+    }
+    red[threadIdx.y][threadIdx.x] =
+        ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
+    __syncthreads();
+#pragma unroll
+    for (int s = TY / 2; s > 0; s >>= 1) {
+        if (threadIdx.y < s)
+            red[threadIdx.y][threadIdx.x] += red[threadIdx.y + s][threadIdx.x];
+        __syncthreads();
+    }
+    if (threadIdx.y == 0 && r < R) out[r] = red[0][threadIdx.x];
 }
 
 // fused: reduce the per-block stat slabs AND produce mean/invstd/running
 // stats in one kernel (halves the per-BN-layer launch count on the
-// non-SyncBN forward path)
+// non-SyncBN forward path).  Same y-split as slab_reduce, but each thread
+// carries the (sum, sumsq) PAIR for its channel so the ty==0 row can
+// finalize without a cross-column exchange.
+template <int TX, int TY>
 __global__ void bn_reduce_finalize_kernel(const float *__restrict__ partial,
                                           int nb, int C, float n,
                                           float momentum, float eps,
@@ -102,24 +122,40 @@ __global__ void bn_reduce_finalize_kernel(const float *__restrict__ partial,
                                           float *__restrict__ running_var,
                                           float *__restrict__ mean_out,
                                           float *__restrict__ invstd_out) {
-    int c = blockIdx.x * blockDim.x + threadIdx.x;
-    if (c >= C) return;
+    __shared__ float red[TY][TX][2];
+    int c = blockIdx.x * TX + threadIdx.x;
     float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
     float q[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
     long R = 2L * C;
-    int b = 0;
-    for (; b + 8 <= nb; b += 8)
+    if (c < C) {
+        int b = threadIdx.y;
+        for (; b + 7 * TY < nb; b += 8 * TY)
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-            a[u] += partial[(long)(b + u) * R + c];
-            q[u] += partial[(long)(b + u) * R + C + c];
+            for (int u = 0; u < 8; ++u) {
+                a[u] += partial[(long)(b + u * TY) * R + c];
+                q[u] += partial[(long)(b + u * TY) * R + C + c];
+            }
+        for (; b < nb; b += TY) {
+            a[0] += partial[(long)b * R + c];
+            q[0] += partial[(long)b * R + C + c];
         }
-    for (; b < nb; ++b) {
-        a[0] += partial[(long)b * R + c];
-        q[0] += partial[(long)b * R + C + c];
     }
-    float sum = ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
-    float ssq = ((q[0] + q[1]) + (q[2] + q[3])) + ((q[4] + q[5]) + (q[6] + q[7]));
+    red[threadIdx.y][threadIdx.x][0] =
+        ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
+    red[threadIdx.y][threadIdx.x][1] =
+        ((q[0] + q[1]) + (q[2] + q[3])) + ((q[4] + q[5]) + (q[6] + q[7]));
+    __syncthreads();
+#pragma unroll
+    for (int s = TY / 2; s > 0; s >>= 1) {
+        if (threadIdx.y < s) {
+            red[threadIdx.y][threadIdx.x][0] += red[threadIdx.y + s][threadIdx.x][0];
+            red[threadIdx.y][threadIdx.x][1] += red[threadIdx.y + s][threadIdx.x][1];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.y != 0 || c >= C) return;
+    float sum = red[0][threadIdx.x][0];
+    float ssq = red[0][threadIdx.x][1];
     float mean = sum / n;
     // two-pass variance can go (slightly) negative by cancellation at
     // large-mean fp16 scales; rsqrt(negative) would poison the tensor
@@ -454,8 +490,8 @@ torch::Tensor bn_stats(torch::Tensor x) {
                                0, stream.stream(),
                                reinterpret_cast<const T *>(x.data_ptr()),
                                partial.data_ptr<float>(), M, C);
-            hipLaunchKernelGGL(sodkit::slab_reduce_kernel,
-                               dim3((2 * C + 255) / 256), dim3(256), 0,
+            hipLaunchKernelGGL((sodkit::slab_reduce_kernel<16, 16>),
+                               dim3((2 * C + 15) / 16), dim3(16, 16), 0,
                                stream.stream(), partial.data_ptr<float>(),
                                out.data_ptr<float>(), nb, 2L * C);
         } else
@@ -513,8 +549,8 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x, double n,
                            reinterpret_cast<const T *>(x.data_ptr()),
                            partial.data_ptr<float>(), M, C);
     });
-    hipLaunchKernelGGL(sodkit::bn_reduce_finalize_kernel,
-                       dim3((C + 255) / 256), dim3(256), 0, stream.stream(),
+    hipLaunchKernelGGL((sodkit::bn_reduce_finalize_kernel<16, 16>),
+                       dim3((C + 15) / 16), dim3(16, 16), 0, stream.stream(),
                        partial.data_ptr<float>(), nb, C, (float)n,
                        (float)momentum, (float)eps,
                        running_mean.data_ptr<float>(),
@@ -588,8 +624,8 @@ torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
                     reinterpret_cast<const T *>(x.data_ptr()),
                     meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
                     partial.data_ptr<float>(), M, C);
-                hipLaunchKernelGGL(sodkit::slab_reduce_kernel,
-                                   dim3((2 * C + 255) / 256), dim3(256), 0,
+                hipLaunchKernelGGL((sodkit::slab_reduce_kernel<16, 16>),
+                                   dim3((2 * C + 15) / 16), dim3(16, 16), 0,
                                    stream.stream(), partial.data_ptr<float>(),
                                    out.data_ptr<float>(), nb, 2L * C);
             } else

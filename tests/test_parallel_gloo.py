@@ -200,3 +200,159 @@ def test_minet_ranks_stay_identical():
     assert torch.equal(outs[0][1], outs[1][1])         # bitwise-identical ranks
     assert outs[0][0] == outs[1][0]                    # identical shared losses
     assert all(l == l and l < 100 for l in outs[0][0])  # finite
+
+
+def _syncbn_worker(rank, world, port, q):
+    """Direct SyncBN numerics: half-batch per rank through one BN layer must
+    match a single-process full-batch BN — outputs, input grads, and
+    gamma/beta grads (C4 fwd + bwd collectives)."""
+    _init(rank, world, port)
+    from sodkit.models.blocks import SodBatchNorm2d
+    torch.manual_seed(5)
+    bn = SodBatchNorm2d(6)
+    set_syncbn(bn)
+    gw = torch.Generator().manual_seed(44)
+    x_all = torch.randn(4, 6, 9, 9, generator=gw) * 3 + 1
+    res_all = torch.randn(4, 6, 9, 9, generator=gw)
+    x = x_all[rank * 2:(rank + 1) * 2].clone().requires_grad_(True)
+    res = res_all[rank * 2:(rank + 1) * 2].clone().requires_grad_(True)
+    y = bn(x, residual=res, relu=True)
+    # weight the loss so per-rank shards sum to the full-batch loss
+    (y.square().sum() / 4).backward()
+    q.put((rank, y.detach(), x.grad.detach(), res.grad.detach(),
+           bn.weight.grad.detach().clone(), bn.bias.grad.detach().clone(),
+           bn.running_mean.clone(), bn.running_var.clone()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_syncbn_matches_full_batch():
+    from sodkit.models.blocks import SodBatchNorm2d
+    port = _store_file()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_syncbn_worker, args=(r, 2, port, q)) for r in range(2)]
+    [p.start() for p in ps]
+    outs = {}
+    for _ in range(2):
+        item = q.get()
+        outs[item[0]] = item[1:]
+    [p.join() for p in ps]
+    assert all(p.exitcode == 0 for p in ps)
+
+    torch.manual_seed(5)
+    bn = SodBatchNorm2d(6)
+    gw = torch.Generator().manual_seed(44)
+    x_all = (torch.randn(4, 6, 9, 9, generator=gw) * 3 + 1).requires_grad_(True)
+    res_all = torch.randn(4, 6, 9, 9, generator=gw).requires_grad_(True)
+    y = bn(x_all, residual=res_all, relu=True)
+    (y.square().sum() / 4).backward()
+
+    for rank in (0, 1):
+        yo, xg, rg, wg, bg, rm, rv = outs[rank]
+        sl = slice(rank * 2, (rank + 1) * 2)
+        assert torch.allclose(yo, y.detach()[sl], atol=1e-5), "fwd mismatch"
+        assert torch.allclose(xg, x_all.grad[sl], atol=1e-5), "dx mismatch"
+        assert torch.allclose(rg, res_all.grad[sl], atol=1e-5), "dres mismatch"
+        # running stats use the SYNCED batch stats: identical on every rank
+        assert torch.allclose(rm, bn.running_mean, atol=1e-5)
+        assert torch.allclose(rv, bn.running_var, atol=1e-4)
+    # param grads are LOCAL shard grads by design — the GradReducer's C3
+    # all-reduce supplies the cross-rank sum, so the rank-sum must equal the
+    # full-batch grad (the DDP oracle test covers the full composition)
+    assert torch.allclose(outs[0][3] + outs[1][3], bn.weight.grad,
+                          atol=1e-5), "dgamma sum mismatch"
+    assert torch.allclose(outs[0][4] + outs[1][4], bn.bias.grad,
+                          atol=1e-5), "dbeta sum mismatch"
+
+
+class _BranchyNet(torch.nn.Module):
+    """One branch is unused on some steps: its params get NO grad, so the
+    reducer's finalize() must launch the straggler bucket defensively."""
+
+    def __init__(self):
+        super().__init__()
+        self.a = torch.nn.Linear(4, 4)
+        self.b = torch.nn.Linear(4, 4)   # unused when use_b is False
+
+    def forward(self, x, use_b):
+        h = self.a(x)
+        if use_b:
+            h = h + self.b(x)
+        return h
+
+
+def _straggler_worker(rank, world, port, q):
+    _init(rank, world, port)
+    torch.manual_seed(9)
+    m = _BranchyNet()
+    red = GradReducer(m, bucket_mb=1)
+    red.broadcast_params(m)
+    x = torch.full((2, 4), float(rank + 1))
+    # neither rank uses branch b: its bucket never fills and finalize()
+    # must all-reduce it as zeros (ranks stay in lockstep on collectives)
+    m(x, use_b=False).sum().backward()
+    red.finalize()
+    q.put((rank, m.a.weight.grad.clone(), m.b.weight.grad.clone()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_unused_param_straggler_bucket():
+    port = _store_file()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_straggler_worker, args=(r, 2, port, q)) for r in range(2)]
+    [p.start() for p in ps]
+    outs = {}
+    for _ in range(2):
+        rank, ga, gb = q.get()
+        outs[rank] = (ga, gb)
+    [p.join() for p in ps]
+    assert all(p.exitcode == 0 for p in ps)
+    assert torch.allclose(outs[0][0], outs[1][0], atol=1e-6)  # used: rank-sum
+    assert torch.count_nonzero(outs[0][1]) == 0               # unused: zeros
+    assert torch.count_nonzero(outs[1][1]) == 0
+
+
+def _metric_worker(rank, world, port, q):
+    _init(rank, world, port)
+    from sodkit.metrics import MAEMeter, PRFMeter
+    gw = torch.Generator().manual_seed(77)
+    preds = torch.rand(6, 1, 16, 16, generator=gw)
+    masks = (torch.rand(6, 1, 16, 16, generator=gw) > 0.5).float()
+    mae, prf = MAEMeter(), PRFMeter()
+    # each rank sees a disjoint shard
+    mae.update(preds[rank::2], masks[rank::2])
+    prf.update(preds[rank::2], masks[rank::2])
+    pr = prf.compute(distributed=True)
+    q.put((rank, mae.compute(distributed=True), (pr["maxf"], pr["meanf"])))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_metric_meters_reduce_across_ranks():
+    from sodkit.metrics import MAEMeter, PRFMeter
+    port = _store_file()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_metric_worker, args=(r, 2, port, q)) for r in range(2)]
+    [p.start() for p in ps]
+    outs = {}
+    for _ in range(2):
+        rank, mae, prf = q.get()
+        outs[rank] = (mae, prf)
+    [p.join() for p in ps]
+    assert all(p.exitcode == 0 for p in ps)
+
+    gw = torch.Generator().manual_seed(77)
+    preds = torch.rand(6, 1, 16, 16, generator=gw)
+    masks = (torch.rand(6, 1, 16, 16, generator=gw) > 0.5).float()
+    mae, prf = MAEMeter(), PRFMeter()
+    mae.update(preds, masks)
+    prf.update(preds, masks)
+    ref_mae, ref_prf = mae.compute(), prf.compute()
+    for rank in (0, 1):
+        assert abs(outs[rank][0] - ref_mae) < 1e-6
+        assert abs(outs[rank][1][0] - ref_prf["maxf"]) < 1e-6
+        assert abs(outs[rank][1][1] - ref_prf["meanf"]) < 1e-6

@@ -49,11 +49,11 @@ def save_checkpoint(paths: dict, epoch: int, model: torch.nn.Module,
 
 
 def load_checkpoint(paths: dict, model: torch.nn.Module, optimizer=None,
-                    map_location="cpu") -> int:
-    """Returns the next epoch to run (saved epoch + 1)."""
+                    map_location="cpu") -> "tuple[int, Optional[dict]]":
+    """Returns (next epoch to run, saved scaler state or None)."""
     ckpt = torch.load(paths["checkpoint_final"], map_location=map_location,
                       weights_only=False)
     model.load_state_dict(ckpt["net_state"])
     if optimizer is not None and ckpt.get("opti_state") is not None:
         optimizer.load_state_dict(ckpt["opti_state"])
-    return int(ckpt["epoch"]) + 1
+    return int(ckpt["epoch"]) + 1, ckpt.get("scaler")

@@ -162,11 +162,14 @@ class Solver:
                                         self.rank, self.world)
         self.total_iters = max(1, cfg.epoch_num * len(loader))
         if cfg.resume and os.path.exists(self.paths["checkpoint_final"]):
-            self.start_epoch = load_checkpoint(self.paths, self.model,
-                                               self.optimizer,
-                                               map_location=self.device)
+            self.start_epoch, scaler = load_checkpoint(self.paths, self.model,
+                                                       self.optimizer,
+                                                       map_location=self.device)
+            if scaler and self.dynamic_scale:
+                self.loss_scale = float(scaler.get("loss_scale", self.loss_scale))
             self.iter_count = self.start_epoch * len(loader)
-            self.log.info(f"resumed at epoch {self.start_epoch}")
+            self.log.info(f"resumed at epoch {self.start_epoch} "
+                          f"(loss_scale {self.loss_scale})")
         if self.rank == 0 and cfg.tb:
             self.tb = TBRecorder(self.paths["tb"], enabled=True)
 

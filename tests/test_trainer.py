@@ -53,8 +53,9 @@ def test_checkpoint_resume_bitwise(cfg):
     p = exp_paths(cfg.output_root, cfg.exp_name)
 
     s2 = Solver(cfg.replace(resume=True))
-    next_epoch = load_checkpoint(p, s2.model, s2.optimizer)
+    next_epoch, scaler_state = load_checkpoint(p, s2.model, s2.optimizer)
     assert next_epoch == 1
+    assert scaler_state is not None and "loss_scale" in scaler_state
     # params and optimizer masters bitwise equal after resume
     for a, b in zip(s.model.parameters(), s2.model.parameters()):
         assert torch.equal(a, b)
@@ -63,3 +64,39 @@ def test_checkpoint_resume_bitwise(cfg):
                            s2.optimizer.state[pb]["momentum_buffer"])
         assert torch.equal(s.optimizer.state[pa]["master"],
                            s2.optimizer.state[pb]["master"])
+
+
+@pytest.mark.timeout(600)
+def test_poly_lr_schedule_with_warmup(cfg):
+    s = Solver(cfg.replace(lr=0.1, lr_type="poly", lr_power=0.9,
+                           warmup_iters=4))
+    s.total_iters = 20
+    # linear warmup: (it+1)/warmup_iters
+    assert s._lr_at(0) == pytest.approx(0.1 * 1 / 4)
+    assert s._lr_at(3) == pytest.approx(0.1 * 4 / 4)
+    # poly decay after warmup
+    for it in (4, 10, 19):
+        assert s._lr_at(it) == pytest.approx(0.1 * (1 - it / 20) ** 0.9)
+    # last iter keeps a positive LR (frac capped at (total-1)/total)
+    assert s._lr_at(25) == pytest.approx(0.1 * (1 - 19 / 20) ** 0.9)
+    # const fallback
+    s2 = Solver(cfg.replace(lr=0.05, lr_type="const", warmup_iters=0))
+    assert s2._lr_at(0) == s2._lr_at(999) == 0.05
+
+
+@pytest.mark.timeout(600)
+def test_resume_restores_dynamic_loss_scale(cfg, tmp_path):
+    # fp16 dynamic-scaler state must survive a resume (CPU: dtype stays
+    # fp32 compute but the scaler bookkeeping is dtype-driven)
+    c = cfg.replace(dtype="fp16", loss_scale=256.0, exp_name="scaler_resume")
+    s = Solver(c)
+    s.train()
+    s.loss_scale = 64.0  # pretend overflows halved it mid-run
+    from sodkit.engine.checkpoint import save_checkpoint
+    p = exp_paths(c.output_root, c.exp_name)
+    save_checkpoint(p, 0, s.model, s.optimizer, {"loss_scale": s.loss_scale})
+
+    s2 = Solver(c.replace(resume=True, epoch_num=2))
+    s2.train()  # resumes: must pick up loss_scale=64, then run epoch 1
+    assert s2.loss_scale <= 64.0  # restored (may halve further on overflow)
+    assert s2.start_epoch == 1

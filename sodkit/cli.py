@@ -16,6 +16,10 @@ from .config import Config, PRESETS, make_config
 from .engine import Solver
 
 
+def _parse_bool(s: str) -> bool:
+    return s.lower() in ("1", "true", "yes")
+
+
 def build_argparser() -> argparse.ArgumentParser:
     ap = argparse.ArgumentParser("sodkit", description=__doc__)
     ap.add_argument("--preset", choices=sorted(PRESETS), default=None,
@@ -24,7 +28,8 @@ def build_argparser() -> argparse.ArgumentParser:
     for f in dataclasses.fields(Config):
         name = f"--{f.name}"
         if f.type == "bool" or isinstance(f.default, bool):
-            ap.add_argument(name, type=lambda s: s.lower() in ("1", "true", "yes"),
+            # both `--resume` (flag style) and `--resume true/false` work
+            ap.add_argument(name, type=_parse_bool, nargs="?", const=True,
                             default=None, metavar="BOOL")
         elif isinstance(f.default, int):
             ap.add_argument(name, type=int, default=None)
@@ -47,7 +52,18 @@ def main(argv=None) -> int:
     cfg = parse_config(argv)
     solver = Solver(cfg)
     if not cfg.test_only:
-        solver.train()
+        solver.train()   # handles --resume itself
+    elif cfg.resume:
+        # eval-only: restore the trained weights before scoring
+        import os
+        from .engine.checkpoint import load_checkpoint
+        if os.path.exists(solver.paths["checkpoint_final"]):
+            load_checkpoint(solver.paths, solver.model, solver.optimizer,
+                            map_location=solver.device)
+        else:
+            raise FileNotFoundError(
+                f"--test_only --resume: no checkpoint at "
+                f"{solver.paths['checkpoint_final']}")
     solver.test()
     return 0
 

@@ -22,3 +22,39 @@ def test_cli_overrides():
                         "--lr", "0.5", "--test_only", "true"])
     assert cfg.batch_size == 3 and cfg.lr == 0.5 and cfg.test_only
     assert cfg.model == "minet_vgg16"
+
+
+def test_cli_bool_flag_styles():
+    from sodkit.cli import parse_config
+    # value style
+    c = parse_config(["--preset", "cpu_plumbing", "--resume", "true"])
+    assert c.resume is True
+    c = parse_config(["--preset", "cpu_plumbing", "--resume", "false"])
+    assert c.resume is False
+    # bare-flag style (the docstring's `--test_only --resume`)
+    c = parse_config(["--preset", "cpu_plumbing", "--test_only", "--resume"])
+    assert c.test_only is True and c.resume is True
+
+
+def test_cli_test_only_resume_requires_checkpoint(tmp_path):
+    from sodkit.cli import main
+    import pytest as _pytest
+    with _pytest.raises(FileNotFoundError):
+        main(["--preset", "cpu_plumbing", "--output_root", str(tmp_path),
+              "--test_only", "--resume", "--size", "48",
+              "--val_samples", "2", "--num_workers", "0"])
+
+
+def test_cli_test_only_resume_restores_weights(tmp_path):
+    from sodkit.cli import main
+    import torch
+    from sodkit.engine.checkpoint import exp_paths
+    # train one tiny epoch, then eval-only with resume must succeed
+    args = ["--preset", "cpu_plumbing", "--output_root", str(tmp_path),
+            "--size", "48", "--train_samples", "4", "--val_samples", "2",
+            "--epoch_num", "1", "--num_workers", "0"]
+    assert main(args) == 0
+    p = exp_paths(str(tmp_path), "cpu_plumbing")
+    import os
+    assert os.path.exists(p["checkpoint_final"])
+    assert main(args + ["--test_only", "--resume"]) == 0

@@ -218,6 +218,16 @@ class Solver:
                         self.tb.scalar("train/loss", loss_meter.val, self.iter_count)
                         self.tb.scalar("train/lr", self._lr_at(self.iter_count - 1),
                                        self.iter_count)
+            if self.tb is not None:   # reference TB parity: image panel per epoch
+                self.model.eval()     # don't touch BN running stats (no SyncBN
+                with torch.no_grad():  # collective in eval: rank-0-only is safe)
+                    pred = torch.sigmoid(
+                        (self.model(img, depth) if depth is not None
+                         else self.model(img))[:1].float())
+                self.model.train()
+                self.tb.image("train/img", img[0].float().cpu(), epoch)
+                self.tb.image("train/mask", mask[0].float().cpu(), epoch)
+                self.tb.image("train/pred", pred[0].cpu(), epoch)
             if self.rank == 0:
                 save_checkpoint(self.paths, epoch, self.model, self.optimizer,
                                 {"loss_scale": self.loss_scale})

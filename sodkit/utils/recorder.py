@@ -26,6 +26,14 @@ class TBRecorder:
         if self.writer is not None:
             self.writer.add_scalar(tag, value, step)
 
+    def image(self, tag: str, img, step: int) -> None:
+        """img: [C,H,W] or [H,W] tensor in [0,1] (reference TB parity:
+        periodic input/pred/mask image panels)."""
+        if self.writer is not None:
+            if img.ndim == 2:
+                img = img.unsqueeze(0)
+            self.writer.add_image(tag, img.float().clamp(0, 1), step)
+
     def close(self) -> None:
         if self.writer is not None:
             self.writer.close()

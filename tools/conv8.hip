@@ -1,0 +1,339 @@
+// 8-phase 256x256 implicit-GEMM conv PROTOTYPE on the verified gemm8
+// base (tools/gemm8.hip: 1008 TF, 25/25 race screen — docs/IGEMM8_PLAN.md).
+// STANDALONE, NOT wired into sodkit dispatch; round-2 must rerun the full
+// verification discipline before any integration.
+//   hipcc -O3 --offload-arch=gfx950 tools/conv8.hip -o conv8 && ./conv8
+//
+// Mapping (same phase schedule / staging / swizzle as gemm8):
+//   "A" operand = weights KRSC [Cout][K], K = R*S*Cin  (C rows = cout,
+//       so each lane's 4 accum regs are 4 consecutive couts -> packed
+//       8 B NHWC stores, identical to the shipped 128-square kernel)
+//   "B" operand = im2col gather over NHWC x: pixel row -> (n,oh,ow)
+//       precomputed per thread; k -> (tap r,s, cin) per stage call
+//       (divisions per stage; shipped kernel's incremental state is the
+//       round-2 optimization).  Out-of-image taps and K/M tails read a
+//       zero page — contributions multiply to 0.
+//   Tails: M%256 via zero-page gather + guarded stores; K%128 via the
+//       absolute-phase loop (odd K-tile count runs a 4-phase half
+//       iteration); Cout%256==0 required (eligible layers only).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdio>
+#include <cstdlib>
+#include <cmath>
+#include <vector>
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef short s16x8 __attribute__((ext_vector_type(8)));
+typedef short s16x4 __attribute__((ext_vector_type(4)));
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef __attribute__((address_space(3))) void lds_void_t;
+typedef __attribute__((address_space(1))) const void gbl_void_t;
+
+#define DEVINL __device__ __forceinline__
+
+DEVINL void glds16(const void *gsrc, char *lds_dst) {
+    __builtin_amdgcn_global_load_lds((const gbl_void_t *)gsrc,
+                                     (lds_void_t *)lds_dst, 16, 0, 0);
+}
+
+DEVINL int swz(int x) { return x ^ (((x >> 9) & 1) << 5); }
+
+constexpr int IMG_BYTES = 16384;   // [256 rows][32 k] bf16
+DEVINL int img_off(int buf, int op, int kh) {
+    return ((buf * 2 + op) * 2 + kh) * IMG_BYTES;
+}
+
+struct ConvShape {
+    int N, H, W, Cin, Cout, R, S, stride, pad, dil, OH, OW;
+    long M() const { return (long)N * OH * OW; }
+    int K() const { return R * S * Cin; }
+};
+
+__global__ __launch_bounds__(512, 1)
+void conv8_kernel(const __hip_bfloat16 *__restrict__ x,     // NHWC
+                  const __hip_bfloat16 *__restrict__ w,     // KRSC [Cout][K]
+                  __hip_bfloat16 *__restrict__ out,         // [M][Cout]
+                  const __hip_bfloat16 *__restrict__ zero_page,
+                  int N_, int H, int W, int Cin, int Cout,
+                  int R, int S, int stride, int pad, int dil,
+                  int OH, int OW) {
+    extern __shared__ char lds[];
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6, lane = tid & 63;
+    const int wm = wave >> 2, wn = wave & 3;   // wm: cout half, wn: pixel quarter
+    const int K = R * S * Cin;
+    const long M = (long)N_ * OH * OW;
+
+    // bijective XCD remap
+    int nwg = gridDim.x;
+    int q = nwg / 8, r8 = nwg % 8;
+    int xcd = blockIdx.x % 8, seq = blockIdx.x / 8;
+    int wgid = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q) + seq;
+    const int ntile = (int)((M + 255) / 256);
+    const int tc = wgid / ntile;               // cout tile
+    const long tp = (long)(wgid % ntile);      // pixel tile
+    const size_t Kb = (size_t)K * 2;
+    const __hip_bfloat16 *Abase = w + (size_t)tc * 256 * K;
+
+    const int NKT = (K + 63) / 64;
+
+    // --- staging state ---
+    int srow[2], scol[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        int s = swz((j * 8192 + tid * 16) & (IMG_BYTES - 1));
+        srow[j] = s >> 6;
+        scol[j] = s & 63;
+    }
+    // B gather: pixel row -> (n, oh, ow), fixed per thread
+    int pn[2], poh[2], pow_[2];
+    bool pvalid[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        long m = tp * 256 + srow[j];
+        pvalid[j] = m < M;
+        long mm = pvalid[j] ? m : 0;
+        pn[j] = (int)(mm / ((long)OH * OW));
+        int rem = (int)(mm % ((long)OH * OW));
+        poh[j] = rem / OW;
+        pow_[j] = rem % OW;
+    }
+
+    auto stage = [&](int kt, int op, int kh) {
+        int ktc = kt < NKT ? kt : 0;
+        char *dst = lds + img_off(kt & 1, op, kh);
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+            int k = ktc * 64 + kh * 32 + (scol[j] >> 1);
+            const char *g;
+            if (op == 0) {                       // weights: linear k-major
+                g = (const char *)(k < K ? (const void *)((const char *)Abase +
+                        (size_t)srow[j] * Kb + (size_t)k * 2)
+                        : (const void *)zero_page);
+            } else {                             // im2col gather
+                int tap = k / Cin, c = k - tap * Cin;
+                int rr = tap / S, ss = tap - rr * S;
+                int ih = poh[j] * stride - pad + rr * dil;
+                int iw = pow_[j] * stride - pad + ss * dil;
+                bool ok = pvalid[j] && k < K &&
+                          ih >= 0 && ih < H && iw >= 0 && iw < W;
+                g = ok ? (const char *)(x + (((size_t)pn[j] * H + ih) * W + iw)
+                                                * Cin + c)
+                       : (const char *)zero_page;
+            }
+            glds16(g, dst + j * 8192 + tid * 16);
+        }
+    };
+    // absolute phase a: stage half hp=(a+2)&3 of K-tile (a+6)>>2
+    auto stage_for = [&](int a) {
+        int hp = (a + 2) & 3;
+        stage((a + 6) >> 2, hp & 1, hp >> 1);
+    };
+
+    // --- fragment offsets ---
+    int a_off[8], b_off[2][2];
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+        a_off[mf] = swz((wm * 128 + mf * 16 + (lane & 15)) * 64 +
+                        ((lane >> 4) << 4));
+#pragma unroll
+    for (int nq = 0; nq < 2; ++nq)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+            b_off[nq][nf] = swz((wn * 64 + nq * 32 + nf * 16 + (lane & 15)) * 64 +
+                                ((lane >> 4) << 4));
+
+    f32x4 acc[8][4];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    // --- prologue (6 halves, steady-state order) ---
+    stage(0, 0, 0); stage(0, 1, 0); stage(0, 0, 1);
+    stage(0, 1, 1); stage(1, 0, 0); stage(1, 1, 0);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __syncthreads();
+
+    s16x8 afrag[8], bfrag[2];
+    // --- main loop over absolute phases (handles odd NKT) ---
+    for (int kt = 0; kt < NKT; ++kt) {
+        const int buf = kt & 1;
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            const int a = kt * 4 + p;
+            const int kh = p >> 1, nq = p & 1;
+            const char *ai = lds + img_off(buf, 0, kh);
+            const char *bi = lds + img_off(buf, 1, kh);
+            if (p == 0) {                 // checkpoint phase
+                stage_for(a);
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+#pragma unroll
+                for (int mf = 0; mf < 8; ++mf)
+                    afrag[mf] = *(const s16x8 *)(ai + a_off[mf]);
+#pragma unroll
+                for (int nf = 0; nf < 2; ++nf)
+                    bfrag[nf] = *(const s16x8 *)(bi + b_off[nq][nf]);
+            } else {
+                if (nq == 0)
+#pragma unroll
+                    for (int mf = 0; mf < 8; ++mf)
+                        afrag[mf] = *(const s16x8 *)(ai + a_off[mf]);
+#pragma unroll
+                for (int nf = 0; nf < 2; ++nf)
+                    bfrag[nf] = *(const s16x8 *)(bi + b_off[nq][nf]);
+                stage_for(a);
+            }
+            __builtin_amdgcn_s_barrier();
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+                for (int nf = 0; nf < 2; ++nf)
+                    acc[mf][nq * 2 + nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        (bf16x8)afrag[mf], (bf16x8)bfrag[nf],
+                        acc[mf][nq * 2 + nf], 0, 0, 0);
+            __builtin_amdgcn_s_setprio(0);
+            __builtin_amdgcn_s_barrier();
+        }
+    }
+
+    // --- epilogue: packed 8 B stores of 4 consecutive couts ---
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    const int c0 = tc * 256 + wm * 128;
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf) {
+            long pix = tp * 256 + wn * 64 + nf * 16 + (lane & 15);
+            if (pix >= M) continue;
+            int cc = c0 + mf * 16 + ((lane >> 4) << 2);
+            s16x4 v;
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+                __hip_bfloat16 t = __float2bfloat16(acc[mf][nf][e]);
+                v[e] = *reinterpret_cast<const short *>(&t);
+            }
+            *reinterpret_cast<s16x4 *>(
+                (__hip_bfloat16 *)out + pix * Cout + cc) = v;
+        }
+}
+
+// ---------------------------------------------------------------------------
+#define HIP_CHECK(c) do { hipError_t e = (c); if (e != hipSuccess) { \
+    fprintf(stderr, "HIP %s @%d\n", hipGetErrorString(e), __LINE__); \
+    exit(1); } } while (0)
+
+static float lcg_float(unsigned &s) {
+    s = s * 1664525u + 1013904223u;
+    return (float)(s >> 8) / (float)(1u << 24) * 2.f - 1.f;
+}
+
+static void run_case(ConvShape cs, int iters, bool check) {
+    const long M = cs.M();
+    const int K = cs.K();
+    std::vector<__hip_bfloat16> hx((size_t)cs.N * cs.H * cs.W * cs.Cin);
+    std::vector<__hip_bfloat16> hw((size_t)cs.Cout * K);
+    unsigned seed = 999;
+    for (auto &v : hx) v = __float2bfloat16(lcg_float(seed));
+    for (auto &v : hw) v = __float2bfloat16(lcg_float(seed));
+
+    __hip_bfloat16 *dx, *dw, *dy, *dz;
+    HIP_CHECK(hipMalloc(&dx, hx.size() * 2));
+    HIP_CHECK(hipMalloc(&dw, hw.size() * 2));
+    HIP_CHECK(hipMalloc(&dy, (size_t)M * cs.Cout * 2));
+    HIP_CHECK(hipMalloc(&dz, 16384));
+    HIP_CHECK(hipMemset(dz, 0, 16384));
+    HIP_CHECK(hipMemcpy(dx, hx.data(), hx.size() * 2, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(dw, hw.data(), hw.size() * 2, hipMemcpyHostToDevice));
+
+    dim3 grid((unsigned)((M + 255) / 256) * (cs.Cout / 256)), block(512);
+    size_t shmem = 128 * 1024;
+    HIP_CHECK(hipFuncSetAttribute((const void *)conv8_kernel,
+              hipFuncAttributeMaxDynamicSharedMemorySize, shmem));
+    auto launch = [&] {
+        hipLaunchKernelGGL(conv8_kernel, grid, block, shmem, 0, dx, dw, dy, dz,
+                           cs.N, cs.H, cs.W, cs.Cin, cs.Cout, cs.R, cs.S,
+                           cs.stride, cs.pad, cs.dil, cs.OH, cs.OW);
+    };
+    launch();
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipGetLastError());
+
+    if (check) {
+        std::vector<__hip_bfloat16> hy((size_t)M * cs.Cout);
+        HIP_CHECK(hipMemcpy(hy.data(), dy, hy.size() * 2, hipMemcpyDeviceToHost));
+        // sampled fp64 reference
+        unsigned s2 = 31337;
+        double max_rel = 0; int bad = 0; const int samples = 2048;
+        for (int t = 0; t < samples; ++t) {
+            s2 = s2 * 1664525u + 1013904223u; long pix = (s2 >> 8) % M;
+            s2 = s2 * 1664525u + 1013904223u; int co = (s2 >> 8) % cs.Cout;
+            int n = (int)(pix / ((long)cs.OH * cs.OW));
+            int rem = (int)(pix % ((long)cs.OH * cs.OW));
+            int oh = rem / cs.OW, ow = rem % cs.OW;
+            double ref = 0;
+            for (int rr = 0; rr < cs.R; ++rr)
+                for (int ss = 0; ss < cs.S; ++ss) {
+                    int ih = oh * cs.stride - cs.pad + rr * cs.dil;
+                    int iw = ow * cs.stride - cs.pad + ss * cs.dil;
+                    if (ih < 0 || ih >= cs.H || iw < 0 || iw >= cs.W) continue;
+                    for (int c = 0; c < cs.Cin; ++c)
+                        ref += (double)__bfloat162float(
+                                   hx[(((size_t)n * cs.H + ih) * cs.W + iw) *
+                                      cs.Cin + c]) *
+                               (double)__bfloat162float(
+                                   hw[(size_t)co * K + (rr * cs.S + ss) *
+                                      cs.Cin + c]);
+                }
+            double got = __bfloat162float(hy[(size_t)pix * cs.Cout + co]);
+            double rel = fabs(got - ref) / (fabs(ref) + 1e-3);
+            if (rel > max_rel) max_rel = rel;
+            if (rel > 0.05) ++bad;
+        }
+        printf("conv8 refcheck N%d %dx%d Ci%d Co%d %dx%d s%d d%d: "
+               "max_rel %.4g bad %d/%d -> %s\n",
+               cs.N, cs.H, cs.W, cs.Cin, cs.Cout, cs.R, cs.S, cs.stride,
+               cs.dil, max_rel, bad, samples, bad == 0 ? "PASS" : "FAIL");
+    }
+    if (iters > 0) {
+        hipEvent_t t0, t1;
+        HIP_CHECK(hipEventCreate(&t0)); HIP_CHECK(hipEventCreate(&t1));
+        for (int i = 0; i < 3; ++i) launch();
+        HIP_CHECK(hipEventRecord(t0));
+        for (int i = 0; i < iters; ++i) launch();
+        HIP_CHECK(hipEventRecord(t1));
+        HIP_CHECK(hipEventSynchronize(t1));
+        float ms = 0;
+        HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+        double tf = 2.0 * M * cs.Cout * K * iters / (ms * 1e-3) / 1e12;
+        printf("conv8 time  N%d %dx%d Ci%d Co%d %dx%d: %.3f ms, %.1f TF\n",
+               cs.N, cs.H, cs.W, cs.Cin, cs.Cout, cs.R, cs.S, ms / iters, tf);
+    }
+    hipFree(dx); hipFree(dw); hipFree(dy); hipFree(dz);
+}
+
+static ConvShape mk(int N, int HW, int Ci, int Co, int k, int st, int dl) {
+    ConvShape c{N, HW, HW, Ci, Co, k, k, st, (k / 2) * dl, dl, 0, 0};
+    c.OH = (c.H + 2 * c.pad - dl * (k - 1) - 1) / st + 1;
+    c.OW = c.OH;
+    return c;
+}
+
+int main(int argc, char **argv) {
+    int iters = argc > 1 ? atoi(argv[1]) : 20;
+    // refcheck: small shapes covering 1x1 / 3x3 / odd-NKT K / M-tail /
+    // stride / dilation
+    run_case(mk(2, 22, 64, 256, 3, 1, 1), 0, true);    // K=576 (odd NKT), M-tail
+    run_case(mk(2, 16, 256, 256, 1, 1, 1), 0, true);   // pure GEMM 1x1
+    run_case(mk(2, 23, 72, 256, 3, 2, 1), 0, true);    // stride 2, ragged
+    run_case(mk(2, 20, 64, 512, 3, 1, 2), 0, true);    // dilation 2
+    // timing: MINet-R50 hot shapes (batch 32)
+    run_case(mk(32, 44, 256, 256, 3, 1, 1), iters, false);
+    run_case(mk(32, 44, 1024, 256, 1, 1, 1), iters, false);
+    run_case(mk(32, 22, 512, 512, 3, 1, 1), iters, false);
+    return 0;
+}

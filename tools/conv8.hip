@@ -51,6 +51,13 @@ struct ConvShape {
     int K() const { return R * S * Cin; }
 };
 
+// PIPE=0: the verified phase body (reads in the barrier shadow).
+// PIPE=1: the docs/IGEMM8_PLAN.md interleave lever — phase p+1's fragment
+// reads issue inside phase p's setprio(1) MFMA region (legal for p<3; the
+// p3->p0 hop crosses the vmcnt checkpoint and must not prefetch).  Only
+// a second B fragment pair is added (+8 VGPRs); the A set is re-read via
+// compiler-ordered WAR after its last MFMA use.  UNMEASURED — round-2 A/B.
+template <int PIPE>
 __global__ __launch_bounds__(512, 1)
 void conv8_kernel(const __hip_bfloat16 *__restrict__ x,     // NHWC
                   const __hip_bfloat16 *__restrict__ w,     // KRSC [Cout][K]
@@ -157,10 +164,12 @@ void conv8_kernel(const __hip_bfloat16 *__restrict__ x,     // NHWC
     asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     __syncthreads();
 
-    s16x8 afrag[8], bfrag[2];
+    s16x8 afrag[8], bfrag[2], bfrag2[2];
     // --- main loop over absolute phases (handles odd NKT) ---
     for (int kt = 0; kt < NKT; ++kt) {
         const int buf = kt & 1;
+        const char *ai1 = lds + img_off(buf, 0, 1);   // kh=1 images (PIPE
+        const char *bi1 = lds + img_off(buf, 1, 1);   // prefetch targets)
 #pragma unroll
         for (int p = 0; p < 4; ++p) {
             const int a = kt * 4 + p;
@@ -177,25 +186,52 @@ void conv8_kernel(const __hip_bfloat16 *__restrict__ x,     // NHWC
                 for (int nf = 0; nf < 2; ++nf)
                     bfrag[nf] = *(const s16x8 *)(bi + b_off[nq][nf]);
             } else {
-                if (nq == 0)
+                if (PIPE == 0) {
+                    if (nq == 0)
 #pragma unroll
-                    for (int mf = 0; mf < 8; ++mf)
-                        afrag[mf] = *(const s16x8 *)(ai + a_off[mf]);
+                        for (int mf = 0; mf < 8; ++mf)
+                            afrag[mf] = *(const s16x8 *)(ai + a_off[mf]);
 #pragma unroll
-                for (int nf = 0; nf < 2; ++nf)
-                    bfrag[nf] = *(const s16x8 *)(bi + b_off[nq][nf]);
+                    for (int nf = 0; nf < 2; ++nf)
+                        bfrag[nf] = *(const s16x8 *)(bi + b_off[nq][nf]);
+                }
                 stage_for(a);
             }
             __builtin_amdgcn_s_barrier();
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_setprio(1);
+            {
+                // PIPE=1: odd phases consume the second B set
+                const s16x8 *bsel = (PIPE == 1 && nq == 1) ? bfrag2 : bfrag;
 #pragma unroll
-            for (int mf = 0; mf < 8; ++mf)
+                for (int mf = 0; mf < 8; ++mf)
 #pragma unroll
-                for (int nf = 0; nf < 2; ++nf)
-                    acc[mf][nq * 2 + nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)afrag[mf], (bf16x8)bfrag[nf],
-                        acc[mf][nq * 2 + nf], 0, 0, 0);
+                    for (int nf = 0; nf < 2; ++nf)
+                        acc[mf][nq * 2 + nf] =
+                            __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                (bf16x8)afrag[mf], (bf16x8)bsel[nf],
+                                acc[mf][nq * 2 + nf], 0, 0, 0);
+            }
+            if (PIPE == 1) {
+                // prefetch phase p+1's fragments inside the MFMA region
+                // (compiler orders the afrag re-reads after their last use)
+                if (p == 0) {
+#pragma unroll
+                    for (int nf = 0; nf < 2; ++nf)
+                        bfrag2[nf] = *(const s16x8 *)(bi + b_off[1][nf]);
+                } else if (p == 1) {
+#pragma unroll
+                    for (int mf = 0; mf < 8; ++mf)
+                        afrag[mf] = *(const s16x8 *)(ai1 + a_off[mf]);
+#pragma unroll
+                    for (int nf = 0; nf < 2; ++nf)
+                        bfrag[nf] = *(const s16x8 *)(bi1 + b_off[0][nf]);
+                } else if (p == 2) {
+#pragma unroll
+                    for (int nf = 0; nf < 2; ++nf)
+                        bfrag2[nf] = *(const s16x8 *)(bi1 + b_off[1][nf]);
+                }                             // p==3: crosses the checkpoint
+            }
             __builtin_amdgcn_s_setprio(0);
             __builtin_amdgcn_s_barrier();
         }
@@ -232,7 +268,7 @@ static float lcg_float(unsigned &s) {
     return (float)(s >> 8) / (float)(1u << 24) * 2.f - 1.f;
 }
 
-static void run_case(ConvShape cs, int iters, bool check) {
+static void run_case(ConvShape cs, int iters, bool check, int pipe = 0) {
     const long M = cs.M();
     const int K = cs.K();
     std::vector<__hip_bfloat16> hx((size_t)cs.N * cs.H * cs.W * cs.Cin);
@@ -252,10 +288,11 @@ static void run_case(ConvShape cs, int iters, bool check) {
 
     dim3 grid((unsigned)((M + 255) / 256) * (cs.Cout / 256)), block(512);
     size_t shmem = 128 * 1024;
-    HIP_CHECK(hipFuncSetAttribute((const void *)conv8_kernel,
+    auto kern = pipe ? conv8_kernel<1> : conv8_kernel<0>;
+    HIP_CHECK(hipFuncSetAttribute((const void *)kern,
               hipFuncAttributeMaxDynamicSharedMemorySize, shmem));
     auto launch = [&] {
-        hipLaunchKernelGGL(conv8_kernel, grid, block, shmem, 0, dx, dw, dy, dz,
+        hipLaunchKernelGGL(kern, grid, block, shmem, 0, dx, dw, dy, dz,
                            cs.N, cs.H, cs.W, cs.Cin, cs.Cout, cs.R, cs.S,
                            cs.stride, cs.pad, cs.dil, cs.OH, cs.OW);
     };
@@ -325,15 +362,16 @@ static ConvShape mk(int N, int HW, int Ci, int Co, int k, int st, int dl) {
 
 int main(int argc, char **argv) {
     int iters = argc > 1 ? atoi(argv[1]) : 20;
+    int pipe = argc > 2 ? atoi(argv[2]) : 0;   // 1 = interleave A/B variant
     // refcheck: small shapes covering 1x1 / 3x3 / odd-NKT K / M-tail /
     // stride / dilation
-    run_case(mk(2, 22, 64, 256, 3, 1, 1), 0, true);    // K=576 (odd NKT), M-tail
-    run_case(mk(2, 16, 256, 256, 1, 1, 1), 0, true);   // pure GEMM 1x1
-    run_case(mk(2, 23, 72, 256, 3, 2, 1), 0, true);    // stride 2, ragged
-    run_case(mk(2, 20, 64, 512, 3, 1, 2), 0, true);    // dilation 2
+    run_case(mk(2, 22, 64, 256, 3, 1, 1), 0, true, pipe);  // K=576 odd NKT
+    run_case(mk(2, 16, 256, 256, 1, 1, 1), 0, true, pipe); // pure GEMM 1x1
+    run_case(mk(2, 23, 72, 256, 3, 2, 1), 0, true, pipe);  // stride 2
+    run_case(mk(2, 20, 64, 512, 3, 1, 2), 0, true, pipe);  // dilation 2
     // timing: MINet-R50 hot shapes (batch 32)
-    run_case(mk(32, 44, 256, 256, 3, 1, 1), iters, false);
-    run_case(mk(32, 44, 1024, 256, 1, 1, 1), iters, false);
-    run_case(mk(32, 22, 512, 512, 3, 1, 1), iters, false);
+    run_case(mk(32, 44, 256, 256, 3, 1, 1), iters, false, pipe);
+    run_case(mk(32, 44, 1024, 256, 1, 1, 1), iters, false, pipe);
+    run_case(mk(32, 22, 512, 512, 3, 1, 1), iters, false, pipe);
     return 0;
 }

@@ -10,11 +10,12 @@
 
 namespace sodkit {
 
-// ---- per-channel sum + sumsq -> out[2][C] (fp32, atomics across blocks) ----
+// ---- per-channel sum + sumsq -> per-block slabs out[nb][2C] (fp32) ----
 // Vector path (C % 8 == 0): each thread owns an 8-channel chunk; the 256
 // threads cover min(C/8, 256) chunks x rows_pb rows per iteration, so a
 // whole iteration's loads are one contiguous coalesced span.  Cross-row
-// partials combine in LDS, one atomicAdd set per block.
+// partials combine in LDS; each block writes a plain [2C] slab (the
+// scalar fallback below is the only atomicAdd path).
 template <typename T>
 __global__ void bn_stats_vec_kernel(const T *__restrict__ x,
                                     float *__restrict__ out, long M, int C) {

@@ -100,3 +100,30 @@ def test_resume_restores_dynamic_loss_scale(cfg, tmp_path):
     s2.train()  # resumes: must pick up loss_scale=64, then run epoch 1
     assert s2.loss_scale <= 64.0  # restored (may halve further on overflow)
     assert s2.start_epoch == 1
+
+
+@pytest.mark.timeout(600)
+def test_profiler_writes_trace(cfg):
+    s = Solver(cfg.replace(profile_steps=3, exp_name="prof"))
+    s.train()
+    tb_dir = exp_paths(cfg.output_root, "prof")["tb"]
+    prof_dir = os.path.join(tb_dir, "profiler")
+    assert os.path.isdir(prof_dir)
+    files = os.listdir(prof_dir)
+    assert any("pt.trace" in f or f.endswith(".json") for f in files), files
+
+
+def test_reducer_detach_releases_hooks():
+    import torch as t
+    from sodkit.parallel import GradReducer
+    m = t.nn.Linear(4, 4)
+    red = GradReducer(m, bucket_mb=1)
+    x = t.ones(2, 4)
+    m(x).sum().backward()
+    red.finalize()
+    g1 = m.weight.grad.clone()
+    red.detach()
+    red.zero_grad()
+    m(x).sum().backward()     # hooks gone: plain accumulation into the views
+    assert t.allclose(m.weight.grad, g1)  # same grads, no reducer involvement
+    assert not red._hooks

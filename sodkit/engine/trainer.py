@@ -181,6 +181,16 @@ class Solver:
                               os.path.join(self.paths["tb"], "profiler"),
                               self.rank)
         prof_step = prof_cm.__enter__()
+        try:
+            self._train_epochs(cfg, loader, sampler, loss_meter, ips_meter,
+                               prof_step)
+        finally:
+            prof_cm.__exit__(None, None, None)
+            if self.tb is not None:
+                self.tb.close()
+
+    def _train_epochs(self, cfg, loader, sampler, loss_meter, ips_meter,
+                      prof_step) -> None:
         for epoch in range(self.start_epoch, cfg.epoch_num):
             if sampler is not None:
                 sampler.set_epoch(epoch)
@@ -232,9 +242,6 @@ class Solver:
                 save_checkpoint(self.paths, epoch, self.model, self.optimizer,
                                 {"loss_scale": self.loss_scale})
             self._barrier()
-        prof_cm.__exit__(None, None, None)
-        if self.tb is not None:
-            self.tb.close()
 
     # ------------------------------------------------------------------
     @torch.no_grad()

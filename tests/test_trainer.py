@@ -31,6 +31,15 @@ def test_train_then_test_writes_reference_layout(cfg):
     pre = os.path.join(p["pre"], "synthetic_val")
     assert os.path.isdir(pre) and len(os.listdir(pre)) > 0
     assert 0 <= res["mae"] <= 1
+    # the deploy artifact round-trips strict into a fresh model
+    from sodkit.models import build_model
+    state = torch.load(p["state_final"], map_location="cpu",
+                       weights_only=True)
+    fresh = build_model(cfg.model)
+    fresh.load_state_dict(state, strict=True)
+    for (n1, p1), (n2, p2) in zip(s.model.state_dict().items(),
+                                  fresh.state_dict().items()):
+        assert n1 == n2 and torch.equal(p1, p2)
 
 
 @pytest.mark.timeout(600)

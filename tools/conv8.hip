@@ -57,7 +57,12 @@ struct ConvShape {
 // p3->p0 hop crosses the vmcnt checkpoint and must not prefetch).  Only
 // a second B fragment pair is added (+8 VGPRs); the A set is re-read via
 // compiler-ordered WAR after its last MFMA use.  UNMEASURED — round-2 A/B.
-template <int PIPE>
+// MODE=0: forward conv.  MODE=1: dgrad — the caller passes x=dy (NHWC over
+// the conv's OUTPUT grid, dims H/W here), w permuted to [Cin][R][S][Cout]
+// (so "Cout" here = the conv's Cin = dx channels, "Cin" here = the conv's
+// Cout = dy channels), OH/OW = the dx grid; the gather inverts the conv
+// geometry per tap: oh = (ih + pad - r*dil)/stride, valid iff exact.
+template <int PIPE, int MODE>
 __global__ __launch_bounds__(512, 1)
 void conv8_kernel(const __hip_bfloat16 *__restrict__ x,     // NHWC
                   const __hip_bfloat16 *__restrict__ w,     // KRSC [Cout][K]
@@ -122,10 +127,20 @@ void conv8_kernel(const __hip_bfloat16 *__restrict__ x,     // NHWC
             } else {                             // im2col gather
                 int tap = k / Cin, c = k - tap * Cin;
                 int rr = tap / S, ss = tap - rr * S;
-                int ih = poh[j] * stride - pad + rr * dil;
-                int iw = pow_[j] * stride - pad + ss * dil;
-                bool ok = pvalid[j] && k < K &&
-                          ih >= 0 && ih < H && iw >= 0 && iw < W;
+                int ih, iw;
+                bool ok = pvalid[j] && k < K;
+                if constexpr (MODE == 0) {
+                    ih = poh[j] * stride - pad + rr * dil;
+                    iw = pow_[j] * stride - pad + ss * dil;
+                    ok = ok && ih >= 0 && ih < H && iw >= 0 && iw < W;
+                } else {                         // dgrad: invert the geometry
+                    int nh = poh[j] + pad - rr * dil;
+                    int nw = pow_[j] + pad - ss * dil;
+                    ih = nh / stride; iw = nw / stride;
+                    ok = ok && nh >= 0 && nw >= 0 &&
+                         nh == ih * stride && nw == iw * stride &&
+                         ih < H && iw < W;
+                }
                 g = ok ? (const char *)(x + (((size_t)pn[j] * H + ih) * W + iw)
                                                 * Cin + c)
                        : (const char *)zero_page;
@@ -288,7 +303,7 @@ static void run_case(ConvShape cs, int iters, bool check, int pipe = 0) {
 
     dim3 grid((unsigned)((M + 255) / 256) * (cs.Cout / 256)), block(512);
     size_t shmem = 128 * 1024;
-    auto kern = pipe ? conv8_kernel<1> : conv8_kernel<0>;
+    auto kern = pipe ? conv8_kernel<1, 0> : conv8_kernel<0, 0>;
     HIP_CHECK(hipFuncSetAttribute((const void *)kern,
               hipFuncAttributeMaxDynamicSharedMemorySize, shmem));
     auto launch = [&] {
@@ -353,6 +368,106 @@ static void run_case(ConvShape cs, int iters, bool check, int pipe = 0) {
     hipFree(dx); hipFree(dw); hipFree(dy); hipFree(dz);
 }
 
+// dgrad refcheck: dx = conv8_kernel<.,1> over dy with permuted weights,
+// checked against a sampled fp64 transposed-conv reference.  Requires the
+// conv's Cin % 256 == 0 (dx channels are the kernel's cout tile).
+static void run_dgrad_case(ConvShape cs, int iters, bool check) {
+    const long Mo = cs.M();                       // dy pixels
+    const long Mi = (long)cs.N * cs.H * cs.W;     // dx pixels
+    const int K = cs.R * cs.S * cs.Cout;          // dgrad GEMM K
+    std::vector<__hip_bfloat16> hdy((size_t)Mo * cs.Cout);
+    std::vector<__hip_bfloat16> hw((size_t)cs.Cout * cs.R * cs.S * cs.Cin);
+    unsigned seed = 4242;
+    for (auto &v : hdy) v = __float2bfloat16(lcg_float(seed));
+    for (auto &v : hw) v = __float2bfloat16(lcg_float(seed));
+    // permute KRSC [Co][R][S][Ci] -> [Ci][R][S][Co]
+    std::vector<__hip_bfloat16> hwt((size_t)cs.Cin * cs.R * cs.S * cs.Cout);
+    for (int co = 0; co < cs.Cout; ++co)
+        for (int t = 0; t < cs.R * cs.S; ++t)
+            for (int ci = 0; ci < cs.Cin; ++ci)
+                hwt[((size_t)ci * cs.R * cs.S + t) * cs.Cout + co] =
+                    hw[((size_t)co * cs.R * cs.S + t) * cs.Cin + ci];
+
+    __hip_bfloat16 *ddy, *dwt, *ddx, *dz;
+    HIP_CHECK(hipMalloc(&ddy, hdy.size() * 2));
+    HIP_CHECK(hipMalloc(&dwt, hwt.size() * 2));
+    HIP_CHECK(hipMalloc(&ddx, (size_t)Mi * cs.Cin * 2));
+    HIP_CHECK(hipMalloc(&dz, 16384));
+    HIP_CHECK(hipMemset(dz, 0, 16384));
+    HIP_CHECK(hipMemcpy(ddy, hdy.data(), hdy.size() * 2, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(dwt, hwt.data(), hwt.size() * 2, hipMemcpyHostToDevice));
+
+    dim3 grid((unsigned)((Mi + 255) / 256) * (cs.Cin / 256)), block(512);
+    size_t shmem = 128 * 1024;
+    auto kern = conv8_kernel<0, 1>;
+    HIP_CHECK(hipFuncSetAttribute((const void *)kern,
+              hipFuncAttributeMaxDynamicSharedMemorySize, shmem));
+    auto launch = [&] {
+        // swapped geometry: x=dy (its grid is OHxOW), out grid = HxW
+        hipLaunchKernelGGL(kern, grid, block, shmem, 0, ddy, dwt, ddx, dz,
+                           cs.N, cs.OH, cs.OW, cs.Cout, cs.Cin, cs.R, cs.S,
+                           cs.stride, cs.pad, cs.dil, cs.H, cs.W);
+    };
+    launch();
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipGetLastError());
+
+    if (check) {
+        std::vector<__hip_bfloat16> hdx((size_t)Mi * cs.Cin);
+        HIP_CHECK(hipMemcpy(hdx.data(), ddx, hdx.size() * 2,
+                            hipMemcpyDeviceToHost));
+        unsigned s2 = 515151;
+        double max_rel = 0; int bad = 0; const int samples = 2048;
+        for (int t = 0; t < samples; ++t) {
+            s2 = s2 * 1664525u + 1013904223u; long pix = (s2 >> 8) % Mi;
+            s2 = s2 * 1664525u + 1013904223u; int ci = (s2 >> 8) % cs.Cin;
+            int n = (int)(pix / ((long)cs.H * cs.W));
+            int rem = (int)(pix % ((long)cs.H * cs.W));
+            int ih = rem / cs.W, iw = rem % cs.W;
+            double ref = 0;
+            for (int rr = 0; rr < cs.R; ++rr)
+                for (int ss = 0; ss < cs.S; ++ss) {
+                    int nh = ih + cs.pad - rr * cs.dil;
+                    int nw = iw + cs.pad - ss * cs.dil;
+                    if (nh < 0 || nw < 0 || nh % cs.stride || nw % cs.stride)
+                        continue;
+                    int oh = nh / cs.stride, ow = nw / cs.stride;
+                    if (oh >= cs.OH || ow >= cs.OW) continue;
+                    for (int co = 0; co < cs.Cout; ++co)
+                        ref += (double)__bfloat162float(
+                                   hdy[(((size_t)n * cs.OH + oh) * cs.OW + ow) *
+                                       cs.Cout + co]) *
+                               (double)__bfloat162float(
+                                   hw[((size_t)co * cs.R * cs.S +
+                                       rr * cs.S + ss) * cs.Cin + ci]);
+                }
+            double got = __bfloat162float(hdx[(size_t)pix * cs.Cin + ci]);
+            double rel = fabs(got - ref) / (fabs(ref) + 1e-3);
+            if (rel > max_rel) max_rel = rel;
+            if (rel > 0.05) ++bad;
+        }
+        printf("conv8 DGRAD refcheck N%d %dx%d Ci%d Co%d %dx%d s%d d%d: "
+               "max_rel %.4g bad %d/%d -> %s\n",
+               cs.N, cs.H, cs.W, cs.Cin, cs.Cout, cs.R, cs.S, cs.stride,
+               cs.dil, max_rel, bad, samples, bad == 0 ? "PASS" : "FAIL");
+    }
+    if (iters > 0) {
+        hipEvent_t t0, t1;
+        HIP_CHECK(hipEventCreate(&t0)); HIP_CHECK(hipEventCreate(&t1));
+        for (int i = 0; i < 3; ++i) launch();
+        HIP_CHECK(hipEventRecord(t0));
+        for (int i = 0; i < iters; ++i) launch();
+        HIP_CHECK(hipEventRecord(t1));
+        HIP_CHECK(hipEventSynchronize(t1));
+        float ms = 0;
+        HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+        double tf = 2.0 * Mi * cs.Cin * K * iters / (ms * 1e-3) / 1e12;
+        printf("conv8 DGRAD time N%d %dx%d Ci%d Co%d %dx%d: %.3f ms, %.1f TF\n",
+               cs.N, cs.H, cs.W, cs.Cin, cs.Cout, cs.R, cs.S, ms / iters, tf);
+    }
+    hipFree(ddy); hipFree(dwt); hipFree(ddx); hipFree(dz);
+}
+
 static ConvShape mk(int N, int HW, int Ci, int Co, int k, int st, int dl) {
     ConvShape c{N, HW, HW, Ci, Co, k, k, st, (k / 2) * dl, dl, 0, 0};
     c.OH = (c.H + 2 * c.pad - dl * (k - 1) - 1) / st + 1;
@@ -369,9 +484,14 @@ int main(int argc, char **argv) {
     run_case(mk(2, 16, 256, 256, 1, 1, 1), 0, true, pipe); // pure GEMM 1x1
     run_case(mk(2, 23, 72, 256, 3, 2, 1), 0, true, pipe);  // stride 2
     run_case(mk(2, 20, 64, 512, 3, 1, 2), 0, true, pipe);  // dilation 2
+    // dgrad (MODE=1): cs.Cin is the kernel's cout tile -> Cin % 256
+    run_dgrad_case(mk(2, 22, 256, 64, 3, 1, 1), 0, true);   // K=576 odd NKT
+    run_dgrad_case(mk(2, 18, 256, 128, 3, 2, 1), 0, true);  // stride 2
+    run_dgrad_case(mk(2, 20, 512, 64, 3, 1, 2), 0, true);   // dilation 2
     // timing: MINet-R50 hot shapes (batch 32)
     run_case(mk(32, 44, 256, 256, 3, 1, 1), iters, false, pipe);
     run_case(mk(32, 44, 1024, 256, 1, 1, 1), iters, false, pipe);
     run_case(mk(32, 22, 512, 512, 3, 1, 1), iters, false, pipe);
+    run_dgrad_case(mk(32, 44, 256, 256, 3, 1, 1), iters, false);
     return 0;
 }

@@ -60,6 +60,8 @@ class Config:
     # distributed / engine
     backend: str = "auto"                # auto -> nccl(RCCL) on GPU, gloo on CPU
     bucket_mb: int = 25                  # gradient bucket size for the RCCL reducer
+    accum_steps: int = 1                 # gradient-accumulation micro-batches
+                                         # per optimizer step (no_sync comm)
     sync_bn: bool = False
     overlap_comm: bool = True            # all-reduce on side stream during backward
     seed: int = 3407

@@ -88,6 +88,7 @@ class Solver:
         self.start_epoch = 0
         self.iter_count = 0
         self.total_iters = 1
+        self._accum_count = 0
         self.tb = None
 
     # ------------------------------------------------------------------
@@ -123,7 +124,8 @@ class Solver:
             sq = n if sq is None else sq + n
         # norm of the SUMMED (not yet averaged/unscaled) grads
         raw_norm = torch.sqrt(sq)
-        eff = self.reducer.grad_scale / self.loss_scale  # avg + unscale
+        accum = max(1, cfg.accum_steps)
+        eff = self.reducer.grad_scale / (self.loss_scale * accum)  # avg+unscale
         norm = (raw_norm * eff).item()
         if self.dynamic_scale and not (norm == norm and norm < float("inf")):
             self.loss_scale = max(self.loss_scale / 2.0, 1.0)
@@ -140,18 +142,35 @@ class Solver:
 
     def step(self, img: torch.Tensor, mask: torch.Tensor,
              depth: Optional[torch.Tensor] = None) -> torch.Tensor:
-        """One optimization step on already-on-device tensors.  Returns the
-        (detached, device-side) unscaled loss."""
-        self.reducer.zero_grad()
+        """One loader batch.  With ``accum_steps == 1`` (default) this is a
+        full optimization step.  With accumulation, the first N-1 calls are
+        no_sync micro-batches (grads accumulate into the bucket views, no
+        collectives); the N-th runs the overlapped all-reduce of the
+        accumulated sums + the fused SGD with 1/(world*accum) folded in.
+        Returns the (detached, device-side) unscaled micro-batch loss."""
+        accum = max(1, self.cfg.accum_steps)
+        first = self._accum_count == 0
+        final = self._accum_count == accum - 1
+        if first:
+            self.reducer.zero_grad()
         logits = self.model(img, depth) if depth is not None else self.model(img)
         loss = self.criterion(logits, mask)
-        (loss * self.loss_scale).backward() if self.loss_scale != 1.0 else loss.backward()
+        scaled = loss * self.loss_scale if self.loss_scale != 1.0 else loss
+        if final:
+            scaled.backward()
+        else:
+            with self.reducer.no_sync():
+                scaled.backward()
+            self._accum_count += 1
+            return loss.detach()
+        self._accum_count = 0
         self.reducer.finalize()
         clip_scale = self._grad_norm_and_clip_scale()
         if clip_scale != 0.0:   # 0.0 == overflow: skip step, scale halved
             lr = self._lr_at(self.iter_count)
             self.optimizer.step(
-                lr, self.reducer.grad_scale / self.loss_scale * clip_scale)
+                lr, self.reducer.grad_scale / (self.loss_scale * accum)
+                    * clip_scale)
         self.iter_count += 1
         return loss.detach()
 
@@ -160,14 +179,15 @@ class Solver:
         cfg = self.cfg
         loader, sampler = create_loader(cfg, "train", self.distributed,
                                         self.rank, self.world)
-        self.total_iters = max(1, cfg.epoch_num * len(loader))
+        accum = max(1, cfg.accum_steps)
+        self.total_iters = max(1, cfg.epoch_num * len(loader) // accum)
         if cfg.resume and os.path.exists(self.paths["checkpoint_final"]):
             self.start_epoch, scaler = load_checkpoint(self.paths, self.model,
                                                        self.optimizer,
                                                        map_location=self.device)
             if scaler and self.dynamic_scale:
                 self.loss_scale = float(scaler.get("loss_scale", self.loss_scale))
-            self.iter_count = self.start_epoch * len(loader)
+            self.iter_count = self.start_epoch * len(loader) // accum
             self.log.info(f"resumed at epoch {self.start_epoch} "
                           f"(loss_scale {self.loss_scale})")
         if self.rank == 0 and cfg.tb:

@@ -57,6 +57,7 @@ class GradReducer:
                  bucket_mb: int = 25, overlap: bool = True):
         self.pg = process_group
         self.overlap = overlap and torch.cuda.is_available()
+        self._sync = True            # False inside no_sync() micro-batches
         self.assert_mode = os.environ.get("SODKIT_COMM_ASSERT", "0") == "1"
         params = [p for p in model.parameters() if p.requires_grad]
         params.reverse()  # reverse registration order ≈ backward arrival order
@@ -117,11 +118,30 @@ class GradReducer:
             dist.broadcast(t.data, src=0, group=self.pg)
 
     # ------------------------------------------------------------------
+    def no_sync(self):
+        """Context manager: backwards inside skip the bucket all-reduce
+        (gradient-accumulation micro-batches; grads keep accumulating into
+        the bucket views).  Same semantics as DDP.no_sync()."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def cm():
+            prev = self._sync
+            self._sync = False
+            try:
+                yield
+            finally:
+                self._sync = prev
+        return cm()
+
+    # ------------------------------------------------------------------
     def _on_grad_ready(self, p: torch.nn.Parameter) -> None:
         b = self._p2bucket[p]
         b.arrived += 1
         if b.arrived == len(b.params):
-            self._launch(b)
+            b.arrived = 0          # cycle per backward (accumulation-safe)
+            if self._sync:
+                self._launch(b)
 
     def _launch(self, b: _Bucket) -> None:
         if _world(self.pg) <= 1:
@@ -153,12 +173,16 @@ class GradReducer:
 
     # ------------------------------------------------------------------
     def finalize(self) -> None:
-        """Call after loss.backward(): wait for every in-flight collective
-        (and launch any straggler buckets defensively)."""
+        """Call after the last (sync) backward of the step: wait for every
+        in-flight collective (and launch any straggler buckets — a bucket
+        whose params got no grad this backward has arrived != 0 ...== 0 via
+        the cycling counter only when it completed)."""
         for b in self.buckets:
-            if b.arrived != len(b.params) and _world(self.pg) > 1:
-                # a param got no grad this step (unused path) — its slot is
-                # zero, all-reduce anyway so ranks agree
+            if b.work is None and _world(self.pg) > 1 and self._sync:
+                # straggler: either some param got no grad this step
+                # (arrived != 0) or the bucket completed during a no_sync
+                # micro-batch — its slot holds the accumulated sum,
+                # all-reduce now so ranks agree
                 self._launch(b)
             if b.work is not None:
                 if self.stats_mode:

@@ -136,3 +136,23 @@ def test_reducer_detach_releases_hooks():
     m(x).sum().backward()     # hooks gone: plain accumulation into the views
     assert t.allclose(m.weight.grad, g1)  # same grads, no reducer involvement
     assert not red._hooks
+
+
+@pytest.mark.timeout(600)
+def test_solver_grad_accumulation(cfg):
+    c = cfg.replace(accum_steps=2, train_samples=8, exp_name="accum")
+    s = Solver(c)
+    s.train()
+    # 4 loader batches / accum 2 -> 2 optimizer steps per epoch
+    assert s.iter_count == 2
+    assert s.total_iters == 2
+    # and the accumulated path still trains: compare one manual pair
+    g = torch.Generator().manual_seed(6)
+    x = torch.rand(2, 3, 48, 48, generator=g)
+    t = (torch.rand(2, 1, 48, 48, generator=g) > 0.5).float()
+    before = s.iter_count
+    l1 = s.step(x, t)          # micro-batch: no optimizer step
+    assert s.iter_count == before
+    l2 = s.step(x, t)          # final: optimizer steps
+    assert s.iter_count == before + 1
+    assert torch.isfinite(l1) and torch.isfinite(l2)

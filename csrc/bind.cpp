@@ -10,6 +10,14 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
                            int64_t S, int64_t stride, int64_t pad, int64_t dil,
                            int64_t ldx, int64_t lddy);
 torch::Tensor channel_sum(torch::Tensor x, int64_t ld);
+// 8-phase opt-in conv backend (SODKIT_IGEMM_8P=1; docs/IGEMM8_PLAN.md)
+bool conv8_eligible(int64_t cout, int64_t cin);
+torch::Tensor conv2d_fwd_8p(torch::Tensor x, torch::Tensor w,
+                            torch::Tensor bias, int64_t stride, int64_t pad,
+                            int64_t dil, int64_t ld);
+torch::Tensor conv2d_dgrad_8p(torch::Tensor dy, torch::Tensor w_t, int64_t H,
+                              int64_t W, int64_t stride, int64_t pad,
+                              int64_t dil, int64_t lddy);
 
 torch::Tensor bn_stats(torch::Tensor x);
 std::vector<torch::Tensor> bn_finalize(torch::Tensor stats, double n,
@@ -55,6 +63,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv2d_dgrad", &conv2d_dgrad);
     m.def("conv2d_wgrad", &conv2d_wgrad);
     m.def("channel_sum", &channel_sum);
+    m.def("conv8_eligible", &conv8_eligible);
+    m.def("conv2d_fwd_8p", &conv2d_fwd_8p);
+    m.def("conv2d_dgrad_8p", &conv2d_dgrad_8p);
     m.def("bn_stats", &bn_stats);
     m.def("bn_finalize", &bn_finalize);
     m.def("bn_stats_finalize", &bn_stats_finalize);

@@ -36,6 +36,7 @@ _invalidate_stale_objects()
 SRC = [
     "csrc/bind.cpp",
     "csrc/conv_igemm.hip",
+    "csrc/conv_igemm8.hip",
     "csrc/bnrelu.hip",
     "csrc/bilinear.hip",
     "csrc/loss.hip",

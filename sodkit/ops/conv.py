@@ -13,6 +13,7 @@ NHW-reduction GEMM.
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -21,6 +22,13 @@ from . import reference
 from ._backend import ext
 
 _CL = torch.channels_last
+
+# opt-in 8-phase 256-square conv backend (csrc/conv_igemm8.hip,
+# docs/IGEMM8_PLAN.md).  Default OFF this round: refcheck-verified on
+# MI355X but not yet A/B'd across the full shape table.  Eligible shapes
+# (Cout % 256 == 0 fwd / Cin % 256 == 0 dgrad) route to it when set;
+# everything else stays on the 128-square kernel either way.
+_USE_8P = os.environ.get("SODKIT_IGEMM_8P", "0") == "1"
 
 
 def _cl_ld(x: torch.Tensor):
@@ -46,8 +54,11 @@ class _ConvFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, bias, stride, padding, dilation):
         x, ldx = _cl_ld(x)
-        y = ext().conv2d_fwd(x, w, bias if bias is not None else torch.Tensor(),
-                             stride, padding, dilation, ldx)
+        b = bias if bias is not None else torch.Tensor()
+        if _USE_8P and w.shape[0] % 256 == 0 and w.shape[3] % 8 == 0:
+            y = ext().conv2d_fwd_8p(x, w, b, stride, padding, dilation, ldx)
+        else:
+            y = ext().conv2d_fwd(x, w, b, stride, padding, dilation, ldx)
         ctx.save_for_backward(x, w)
         ctx.meta = (stride, padding, dilation, bias is not None, ldx)
         return y
@@ -61,8 +72,12 @@ class _ConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             # dgrad A-operand layout [Cin, R, S, Cout] (a small permuted copy)
             w_t = w.permute(3, 1, 2, 0).contiguous()
-            dx = ext().conv2d_dgrad(dy, w_t, x.shape[2], x.shape[3],
-                                    stride, padding, dilation, lddy)
+            if _USE_8P and w_t.shape[0] % 256 == 0 and w_t.shape[3] % 8 == 0:
+                dx = ext().conv2d_dgrad_8p(dy, w_t, x.shape[2], x.shape[3],
+                                           stride, padding, dilation, lddy)
+            else:
+                dx = ext().conv2d_dgrad(dy, w_t, x.shape[2], x.shape[3],
+                                        stride, padding, dilation, lddy)
         if ctx.needs_input_grad[1]:
             dw = ext().conv2d_wgrad(x, dy, w.shape[1], w.shape[2],
                                     stride, padding, dilation, ldx,

@@ -450,3 +450,74 @@ def test_conv_k_tail_not_multiple_of_64():
     yr = F.conv2d(x.float(), w.float().permute(0, 3, 1, 2), None,
                   stride=2, padding=1)
     close(y, yr, what="K=504 tail")
+
+
+# ---- 8-phase opt-in conv backend (csrc/conv_igemm8.hip) -------------------
+IG8_SHAPES = [
+    # (N, Cin, H, Cout, k, stride, pad, dil) — all Cout%256==0 eligible
+    (2, 64, 22, 256, 3, 1, 1, 1),      # odd K-tile count + M%256 tail
+    (2, 256, 16, 256, 1, 1, 0, 1),     # pure GEMM 1x1
+    (2, 72, 23, 256, 3, 2, 1, 1),      # stride 2, ragged spatial
+    (2, 64, 20, 512, 3, 1, 2, 2),      # dilation 2
+    (2, 1024, 14, 256, 1, 1, 0, 1),    # deep-K 1x1
+]
+
+
+@pytest.mark.parametrize("shape", IG8_SHAPES)
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_conv8p_fwd_matches_reference(shape, dtype):
+    from sodkit.ops._backend import ext
+    N, Cin, H, Cout, k, st, pad, dil = shape
+    torch.manual_seed(0)
+    x = _cl(torch.randn(N, Cin, H, H, device="cuda").to(dtype))
+    w = (torch.randn(Cout, k, k, Cin, device="cuda") * 0.1).to(dtype)
+    b = torch.randn(Cout, device="cuda").to(dtype)
+    y = ext().conv2d_fwd_8p(x, w, b, st, pad, dil, Cin)
+    y_ref = F.conv2d(x.float(), w.float().permute(0, 3, 1, 2), b.float(),
+                     stride=st, padding=pad, dilation=dil)
+    assert y.shape == y_ref.shape
+    close(y, y_ref, what=f"8p fwd {shape} {dtype}")
+
+
+@pytest.mark.parametrize("shape", [s for s in IG8_SHAPES if s[1] % 256 == 0
+                                   or s[3] % 256 == 0])
+def test_conv8p_dgrad_matches_reference(shape):
+    """dgrad eligibility is Cin%256 (dx channels) — test it by swapping
+    roles: dgrad of a conv with Cin=256-eligible shapes."""
+    from sodkit.ops._backend import ext
+    N, Cin, H, Cout, k, st, pad, dil = shape
+    if Cin % 256 != 0:
+        Cin, Cout = Cout, max(64, Cin)   # make dx channels eligible
+    torch.manual_seed(1)
+    dtype = torch.bfloat16
+    x = torch.randn(N, Cin, H, H, device="cuda", requires_grad=True)
+    wf = torch.randn(Cout, Cin, k, k, device="cuda") * 0.1
+    y_ref = F.conv2d(x, wf, None, stride=st, padding=pad, dilation=dil)
+    gy = torch.randn_like(y_ref)
+    (dx_ref,) = torch.autograd.grad(y_ref, x, gy)
+
+    w_krsc = wf.permute(0, 2, 3, 1).contiguous().to(dtype)       # [Co][R][S][Ci]
+    w_t = w_krsc.permute(3, 1, 2, 0).contiguous()                # [Ci][R][S][Co]
+    dy = _cl(gy.to(dtype))
+    dx = ext().conv2d_dgrad_8p(dy, w_t, H, H, st, pad, dil, Cout)
+    close(dx, dx_ref, what=f"8p dgrad {shape}")
+
+
+def test_conv8p_matches_default_kernel_e2e():
+    """The opt-in dispatch produces the same numbers as the default kernel
+    on an eligible layer (both vs their common fp32 reference)."""
+    from sodkit.ops import conv as conv_mod
+    from sodkit import ops
+    torch.manual_seed(2)
+    x = _cl(torch.randn(2, 64, 22, 22, device="cuda").to(torch.bfloat16))
+    w = (torch.randn(256, 3, 3, 64, device="cuda") * 0.1).to(torch.bfloat16)
+    ref = F.conv2d(x.float(), w.float().permute(0, 3, 1, 2), None, padding=1)
+    y0 = ops.conv2d(x, w, None, 1, 1, 1)
+    prev = conv_mod._USE_8P
+    conv_mod._USE_8P = True
+    try:
+        y1 = ops.conv2d(x, w, None, 1, 1, 1)
+    finally:
+        conv_mod._USE_8P = prev
+    close(y0, ref, what="default vs ref")
+    close(y1, ref, what="8p vs ref")

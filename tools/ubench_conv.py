@@ -38,7 +38,7 @@ def bench(fn, iters=10):
 
 def main():
     dev = "cuda"
-    print(f"{'shape':>12} {'M':>8} {'K':>6} {'N':>6} | {'fwd_tf':>7} {'dgrad_tf':>8} {'wgrad_tf':>8}")
+    print(f"{'shape':>12} {'M':>8} {'K':>6} {'N':>6} | {'fwd_tf':>7} {'dgrad_tf':>8} {'wgrad_tf':>8} | {'fwd8p':>7} {'dgrad8p':>7}")
     for name, N, Cin, H, Cout, k, st, pad, dil in SHAPES:
         x = torch.randn(N, Cin, H, H, device=dev).to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)
@@ -54,9 +54,18 @@ def main():
         w_t = w.permute(3, 1, 2, 0).contiguous()
         t_d = bench(lambda: _C.conv2d_dgrad(y, w_t, H, H, st, pad, dil, Cout))
         t_w = bench(lambda: _C.conv2d_wgrad(x, y, k, k, st, pad, dil, Cin, Cout))
+        # 8-phase opt-in backend A/B where the shape is eligible
+        f8 = d8 = "      -"
+        if Cout % 256 == 0 and Cin % 8 == 0:
+            t = bench(lambda: _C.conv2d_fwd_8p(x, w, nob, st, pad, dil, Cin))
+            f8 = f"{flop / t / 1e12:7.0f}"
+        if Cin % 256 == 0 and Cout % 8 == 0:
+            t = bench(lambda: _C.conv2d_dgrad_8p(y, w_t, H, H, st, pad, dil,
+                                                 Cout))
+            d8 = f"{flop / t / 1e12:7.0f}"
         print(f"{name:>12} {M:>8} {K:>6} {Cout:>6} | "
               f"{flop / t_f / 1e12:7.0f} {flop / t_d / 1e12:8.0f} "
-              f"{flop / t_w / 1e12:8.0f}")
+              f"{flop / t_w / 1e12:8.0f} | {f8} {d8}")
 
 
 if __name__ == "__main__":

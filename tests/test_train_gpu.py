@@ -113,3 +113,25 @@ def test_grad_accumulation_on_gpu(tmp_path):
     assert s.iter_count == 1
     assert torch.isfinite(l1) and torch.isfinite(l2)
     assert not torch.equal(p0, s.model.classifier.weight.detach())
+
+
+def test_minet_step_with_8p_backend(tmp_path):
+    """Full MINet-R50 step with the opt-in 8-phase conv backend enabled:
+    eligible layers (Cout%256 fwd / Cin%256 dgrad) route to
+    conv_igemm8.hip, the rest stay on the 128-square kernel — the mixed
+    dispatch must train."""
+    from sodkit.ops import conv as conv_mod
+    s = _mk_solver("minet_r50_352", batch_size=2, size=96,
+                   output_root=str(tmp_path), exp_name="ig8")
+    s.total_iters = 100
+    x, t, _ = _batch(s, n=2, size=96)
+    l0 = s.step(x, t).item()
+    prev = conv_mod._USE_8P
+    conv_mod._USE_8P = True
+    try:
+        l1 = s.step(x, t).item()
+        l2 = s.step(x, t).item()
+    finally:
+        conv_mod._USE_8P = prev
+    assert l0 == l0 and l1 == l1 and l2 == l2, "NaN loss under 8p"
+    assert l2 < l0 * 1.5, (l0, l1, l2)   # still optimizing, not diverging

@@ -429,3 +429,42 @@ def test_grad_accumulation_matches_full_batch():
         opt.step(lr=0.01, grad_scale=1.0 / 4)
     assert torch.allclose(outs[0][0], m[3].weight, atol=2e-5)
     assert torch.allclose(outs[0][1], m[3].bias, atol=2e-5)
+
+
+def _accum_straggler_worker(rank, world, port, q):
+    """A param used during no_sync micro-batches but UNUSED in the final
+    sync backward: its bucket never completes in the final pass, so
+    finalize()'s straggler launch must all-reduce the accumulated sum."""
+    _init(rank, world, port)
+    torch.manual_seed(21)
+    m = _BranchyNet()
+    red = GradReducer(m, bucket_mb=1)
+    red.broadcast_params(m)
+    x = torch.full((2, 4), float(rank + 1))
+    red.zero_grad()
+    with red.no_sync():
+        m(x, use_b=True).sum().backward()    # micro-batch: b used
+    m(x, use_b=False).sum().backward()       # final: b unused
+    red.finalize()
+    q.put((rank, m.a.weight.grad.clone(), m.b.weight.grad.clone()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_accum_straggler_bucket_still_reduced():
+    port = _store_file()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_accum_straggler_worker, args=(r, 2, port, q))
+          for r in range(2)]
+    [p.start() for p in ps]
+    outs = {}
+    for _ in range(2):
+        rank, ga, gb = q.get()
+        outs[rank] = (ga, gb)
+    [p.join() for p in ps]
+    assert all(p.exitcode == 0 for p in ps)
+    # both grads are rank-summed (incl. b's micro-batch-only contribution)
+    assert torch.allclose(outs[0][0], outs[1][0], atol=1e-6)
+    assert torch.allclose(outs[0][1], outs[1][1], atol=1e-6)
+    assert torch.count_nonzero(outs[0][1]) > 0   # b DID accumulate

@@ -51,6 +51,15 @@ class Solver:
         # identical seed on every rank -> identical random init
         seed_everything(cfg.seed)
         model = build_model(cfg.model)
+        if self.is_gpu and self.dtype == torch.float32:
+            # the MFMA conv kernels are bf16/fp16 (fp32 accumulate); a full
+            # f32-MFMA conv path (mfma_f32_16x16x4f32) is not built — fail
+            # here with guidance instead of deep in the first conv dispatch
+            raise ValueError(
+                "dtype=fp32 on GPU is not supported: the CDNA4 conv kernels "
+                "compute in bf16/fp16 with fp32 accumulation (BN/loss/SGD "
+                "already run fp32 internally). Use --dtype bf16 (default) "
+                "or fp16; fp32 end-to-end runs on CPU (device=cpu).")
         if self.is_gpu:
             # NOTE: no model.to(memory_format=channels_last) — sodkit weights
             # are KRSC-contiguous by construction and activations are made

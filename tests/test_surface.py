@@ -69,3 +69,14 @@ def test_kernel_inventory_sources_present():
         assert re.search(sym, src), f"kernel symbol missing: {sym}"
     assert re.search(r"dynfilter|dynamic_filter|df_fwd",
                      (root / "dynfilter.hip").read_text())
+
+
+def test_8p_eligibility_predicate_parity():
+    """The python dispatch predicate (sodkit/ops/conv.py) and the C++
+    TORCH_CHECK predicate (_C.conv8_eligible) must agree, or an eligible-
+    looking layer would raise at runtime on GPU."""
+    from sodkit import _C
+    for cout in (1, 8, 64, 128, 255, 256, 512, 1024, 2048):
+        for cin in (1, 3, 7, 8, 64, 72, 256):
+            py = cout % 256 == 0 and cin % 8 == 0     # ops/conv.py fwd gate
+            assert _C.conv8_eligible(cout, cin) == py, (cout, cin)

@@ -40,3 +40,30 @@ def test_meters():
     assert m.avg == 2.0 and m.val == 3.0
     t = Timer()
     assert t.elapsed() >= 0.0 and t.lap() >= 0.0
+
+
+def test_timer_laps():
+    import time
+    from sodkit.utils import Timer
+    t = Timer()
+    time.sleep(0.01)
+    l1 = t.lap()
+    assert 0.005 < l1 < 1.0
+    l2 = t.lap()
+    assert l2 < l1            # immediate second lap is shorter
+    assert t.elapsed() >= l1
+
+
+def test_loader_with_workers():
+    """num_workers > 0: the seeded generator must stay deterministic across
+    worker processes (each index derives its own generator)."""
+    from sodkit.config import make_config
+    from sodkit.data import create_loader
+    cfg = make_config("cpu_plumbing", size=32, train_samples=8, batch_size=4,
+                      num_workers=2)
+    l1, _ = create_loader(cfg, "val", distributed=False)
+    l2, _ = create_loader(cfg, "val", distributed=False)
+    import torch
+    b1 = next(iter(l1))
+    b2 = next(iter(l2))
+    assert torch.equal(b1[0], b2[0]) and torch.equal(b1[1], b2[1])

@@ -33,7 +33,13 @@
 // LDS swizzle: st_16x32 per the guide, byte ^= ((byte>>9)&1)<<5 within
 // each 1024 B (16 row x 64 B) subtile; SWZ=2 additionally XORs bit8->bit4
 // (16 B granule, keeps b128 contiguity) which the bank model says removes
-// the residual 2-way conflict — A/B template param to A/B on hardware.
+// the residual 2-way conflict — MEASURED: swz1 1008 TF beats swz2 940 TF
+// at 4k^3, keep the guide's swizzle (profiles/gemm8.log).
+//
+// The pipelined-fragment-read interleave experiment (issue phase p+1's
+// ds_reads inside phase p's MFMA region) lives in tools/conv8.hip as the
+// PIPE=1 template arm — A/B it there (a 1x1 conv IS this GEMM), keeping
+// this file frozen as the verified reference implementation.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>

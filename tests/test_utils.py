@@ -67,3 +67,13 @@ def test_loader_with_workers():
     b1 = next(iter(l1))
     b2 = next(iter(l2))
     assert torch.equal(b1[0], b2[0]) and torch.equal(b1[1], b2[1])
+
+
+def test_tb_recorder_image_noop_without_tb(tmp_path):
+    import torch
+    from sodkit.utils.recorder import TBRecorder
+    r = TBRecorder(str(tmp_path), enabled=True)  # tb not installed -> no-op
+    r.image("x", torch.rand(3, 8, 8), 0)         # must not raise
+    r.image("y", torch.rand(8, 8), 1)            # 2D promoted to [1,H,W]
+    r.scalar("z", 1.0, 0)
+    r.close()

@@ -118,3 +118,25 @@ def test_dummy_stages_target_dead_or_safe_regions():
                 assert p > last_read or replaced_earlier, (
                     f"nkt={nkt}: dummy staging ({kt},{op},{kh}) at phase {p}"
                     f" could race tile {t} read until {last_read}")
+
+
+def test_pipe_prefetch_legality_rule():
+    """The PIPE=1 interleave (tools/conv8.hip) issues phase p+1's fragment
+    reads inside phase p's MFMA region.  Legal iff every image phase p+1
+    consumes is retired by a checkpoint that executed at or before phase p.
+    The derived rule — prefetch on every phase except the last of each
+    4-phase group (p % 4 == 3) — must be exactly what the automaton
+    permits."""
+    for nkt in range(1, 9):
+        stages, phases, _ = simulate(nkt)
+        for p in range(nkt * 4 - 1):
+            horizon = retire_horizon(stages, p)
+            legal = True
+            for (kt, op, kh) in phases[p + 1]["consumes"]:
+                cands = [i for i, (ph, k, o, h) in enumerate(stages)
+                         if k == kt and o == op and h == kh]
+                if not cands or cands[0] >= horizon:
+                    legal = False
+            assert legal == (p % 4 != 3), (
+                f"nkt={nkt} phase {p}: prefetch legality {legal} != rule "
+                f"{(p % 4 != 3)}")

@@ -80,3 +80,17 @@ def test_8p_eligibility_predicate_parity():
         for cin in (1, 3, 7, 8, 64, 72, 256):
             py = cout % 256 == 0 and cin % 8 == 0     # ops/conv.py fwd gate
             assert _C.conv8_eligible(cout, cin) == py, (cout, cin)
+
+
+def test_8p_host_guards_without_gpu():
+    """The 8p entry points exist in the built extension and their guards
+    fire cleanly on CPU tensors (binding arity + TORCH_CHECK path)."""
+    import pytest
+    import torch
+    from sodkit import _C
+    x = torch.randn(1, 8, 8, 8)
+    w = torch.randn(256, 1, 1, 8)
+    with pytest.raises(RuntimeError, match="is_cuda"):
+        _C.conv2d_fwd_8p(x, w, torch.Tensor(), 1, 0, 1, 8)
+    with pytest.raises(RuntimeError, match="is_cuda"):
+        _C.conv2d_dgrad_8p(x, w, 8, 8, 1, 0, 1, 256)

@@ -79,4 +79,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dynfilter_fwd", &dynfilter_fwd);
     m.def("dynfilter_bwd", &dynfilter_bwd);
     m.def("sgd_fused", &sgd_fused);
+    // build metadata (diagnostics: confirms the in-tree gfx950 build is
+    // the one loaded, e.g. in the driver's native-code check)
+    m.attr("BUILD_ARCH") = "gfx950";
+    m.attr("BUILD_DATE") = __DATE__ " " __TIME__;
 }

@@ -29,6 +29,10 @@ def _init(rank: int, world: int, store_file: str):
     # FileStore rendezvous: no TCP port races between tests
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
+    # each spawned worker would otherwise open an all-CPU intraop pool:
+    # 2 workers x ncpu threads oversubscribes the host and makes the
+    # spawn-heavy suite slow/flaky under parallel test runners
+    torch.set_num_threads(2)
     dist.init_process_group("gloo", rank=rank, world_size=world,
                             init_method=f"file://{store_file}")
 

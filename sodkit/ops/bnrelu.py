@@ -8,8 +8,9 @@ GPU path: hand-written HIP kernels (csrc/bnrelu.hip):
 
 SyncBN is the same op with a process group: the per-channel
 stats (and backward reduction terms) are all-reduced once per call as a single
-coalesced [2C+1] buffer — no per-layer small-message storm beyond one
-all-reduce per BN layer, and the kernel path is identical.
+coalesced [2, C] buffer (per-rank counts are equal by construction —
+DistributedSampler drop_last — so no count exchange and no host sync), and
+the kernel path is identical to the non-sync one.
 """
 
 from __future__ import annotations

@@ -18,8 +18,9 @@ def test_minet_res2net_forward_shape():
     assert m(x).shape == (1, 1, 64, 64)
 
 
-def test_hdfnet_forward_shape():
-    m = build_model("hdfnet_res50")
+@pytest.mark.parametrize("name", ["hdfnet_res50", "hdfnet_vgg16"])
+def test_hdfnet_forward_shape(name):
+    m = build_model(name)
     rgb = torch.randn(1, 3, 64, 64)
     d = torch.randn(1, 1, 64, 64)
     assert m(rgb, d).shape == (1, 1, 64, 64)

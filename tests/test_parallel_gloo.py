@@ -472,3 +472,49 @@ def test_accum_straggler_bucket_still_reduced():
     assert torch.allclose(outs[0][0], outs[1][0], atol=1e-6)
     assert torch.allclose(outs[0][1], outs[1][1], atol=1e-6)
     assert torch.count_nonzero(outs[0][1]) > 0   # b DID accumulate
+
+
+def _world3_worker(rank, port, q):
+    _init(rank, 3, port)
+    torch.manual_seed(7)
+    m = torch.nn.Linear(4, 2)
+    red = GradReducer(m, bucket_mb=1)
+    red.broadcast_params(m)
+    m(torch.full((1, 4), float(rank + 1))).sum().backward()
+    red.finalize()
+    from sodkit.metrics import MAEMeter
+    gw = torch.Generator().manual_seed(3)
+    preds = torch.rand(5, 1, 8, 8, generator=gw)
+    masks = (torch.rand(5, 1, 8, 8, generator=gw) > 0.5).float()
+    meter = MAEMeter()
+    sl = slice(rank, 5, 3)             # shards of size 2/2/1
+    meter.update(preds[sl], masks[sl])
+    q.put((rank, m.weight.grad.clone(), meter.compute(distributed=True)))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_reducer_world3_odd_world_size():
+    """Non-power-of-two world: 3 ranks, uneven metric shards (5 images)."""
+    port = _store_file()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_world3_worker, args=(r, port, q))
+          for r in range(3)]
+    [p.start() for p in ps]
+    outs = {}
+    for _ in range(3):
+        rank, g, mae = q.get()
+        outs[rank] = (g, mae)
+    [p.join() for p in ps]
+    assert all(p.exitcode == 0 for p in ps)
+    assert torch.allclose(outs[0][0], outs[1][0], atol=1e-6)
+    assert torch.allclose(outs[0][0], outs[2][0], atol=1e-6)
+    from sodkit.metrics import MAEMeter
+    gw = torch.Generator().manual_seed(3)
+    preds = torch.rand(5, 1, 8, 8, generator=gw)
+    masks = (torch.rand(5, 1, 8, 8, generator=gw) > 0.5).float()
+    ref = MAEMeter()
+    ref.update(preds, masks)
+    for r in range(3):
+        assert abs(outs[r][1] - ref.compute()) < 1e-6

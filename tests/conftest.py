@@ -1,3 +1,10 @@
+"""Shared test config: registers the `gpu` marker (auto-skipped off-GPU).
+
+The suite is designed for the driver's SERIAL runs (`-x -q -m "not gpu"`);
+it also passes under pytest-xdist, but the gloo spawn tests want a couple
+of cores each (worker intraop threads are pinned to 2 in test_parallel_gloo
+for exactly that reason)."""
+
 import pytest
 import torch
 

@@ -94,3 +94,17 @@ def test_8p_host_guards_without_gpu():
         _C.conv2d_fwd_8p(x, w, torch.Tensor(), 1, 0, 1, 8)
     with pytest.raises(RuntimeError, match="is_cuda"):
         _C.conv2d_dgrad_8p(x, w, 8, 8, 1, 0, 1, 256)
+
+
+def test_backend_loader_contract():
+    """has_ext() finds the in-tree build; ext() returns the module whose
+    file lives INSIDE the repo (not site-packages) — the driver's
+    native-code check records exactly this."""
+    import os
+    import sodkit
+    from sodkit.ops._backend import ext, has_ext
+    assert has_ext()
+    mod = ext()
+    pkg_dir = os.path.dirname(os.path.abspath(sodkit.__file__))
+    assert os.path.dirname(os.path.abspath(mod.__file__)) == pkg_dir
+    assert mod.BUILD_ARCH == "gfx950"

@@ -41,7 +41,8 @@ def _world(pg) -> int:
 
 
 class _Bucket:
-    __slots__ = ("params", "flat", "arrived", "work", "ready_event", "index")
+    __slots__ = ("params", "flat", "arrived", "work", "ready_event", "index",
+                 "launched")
 
     def __init__(self, index: int):
         self.index = index
@@ -50,6 +51,9 @@ class _Bucket:
         self.arrived = 0
         self.work = None
         self.ready_event = None
+        self.launched = False  # all-reduce issued this step (survives the
+        #                        assert-mode synchronous wait, which nulls
+        #                        `work` — finalize() must not re-launch)
 
 
 class GradReducer:
@@ -146,6 +150,7 @@ class GradReducer:
     def _launch(self, b: _Bucket) -> None:
         if _world(self.pg) <= 1:
             return
+        b.launched = True
         if self.stats_mode:
             import time
             self.stats["launches"] += 1
@@ -178,11 +183,13 @@ class GradReducer:
         whose params got no grad this backward has arrived != 0 ...== 0 via
         the cycling counter only when it completed)."""
         for b in self.buckets:
-            if b.work is None and _world(self.pg) > 1 and self._sync:
+            if not b.launched and _world(self.pg) > 1 and self._sync:
                 # straggler: either some param got no grad this step
                 # (arrived != 0) or the bucket completed during a no_sync
                 # micro-batch — its slot holds the accumulated sum,
-                # all-reduce now so ranks agree
+                # all-reduce now so ranks agree.  (`launched`, not
+                # `work is None`: assert mode waits synchronously and nulls
+                # `work`, which must not look like a straggler.)
                 self._launch(b)
             if b.work is not None:
                 if self.stats_mode:
@@ -194,11 +201,13 @@ class GradReducer:
                     b.work.wait()  # current stream waits for the collective
                 b.work = None
             b.arrived = 0
+            b.launched = False
 
     def zero_grad(self) -> None:
         for b in self.buckets:
             b.flat.zero_()
             b.arrived = 0
+            b.launched = False
 
     def pop_stats(self) -> dict:
         """Comm-overlap summary since the last call (stats mode only):

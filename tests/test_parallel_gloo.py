@@ -474,6 +474,47 @@ def test_accum_straggler_bucket_still_reduced():
     assert torch.count_nonzero(outs[0][1]) > 0   # b DID accumulate
 
 
+def _assert_mode_worker(rank, world, port, q, assert_mode):
+    """SODKIT_COMM_ASSERT=1 must be numerics-neutral: the synchronous
+    checksum wait in _launch() nulls `work`, and finalize() must NOT treat
+    that as a straggler and all-reduce the bucket a second time."""
+    os.environ["SODKIT_COMM_ASSERT"] = "1" if assert_mode else "0"
+    _init(rank, world, port)
+    torch.manual_seed(7)
+    m = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.ReLU(),
+                            torch.nn.Linear(16, 4))
+    red = GradReducer(m, bucket_mb=1)
+    red.broadcast_params(m)
+    x = torch.full((2, 8), float(rank + 1))
+    m(x).sum().backward()
+    red.finalize()
+    q.put((rank, [p.grad.clone() for p in m.parameters()]))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_comm_assert_mode_is_numerics_neutral():
+    outs = {}
+    for assert_mode in (False, True):
+        port = _store_file()
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        ps = [ctx.Process(target=_assert_mode_worker,
+                          args=(r, 2, port, q, assert_mode)) for r in range(2)]
+        [p.start() for p in ps]
+        res = {}
+        for _ in range(2):
+            rank, grads = q.get()
+            res[rank] = grads
+        [p.join() for p in ps]
+        assert all(p.exitcode == 0 for p in ps)
+        outs[assert_mode] = res
+    for rank in (0, 1):
+        for g_off, g_on in zip(outs[False][rank], outs[True][rank]):
+            assert torch.equal(g_off, g_on), \
+                "assert mode changed gradient numerics (double all-reduce?)"
+
+
 def _world3_worker(rank, port, q):
     _init(rank, 3, port)
     torch.manual_seed(7)

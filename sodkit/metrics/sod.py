@@ -64,39 +64,42 @@ class MAEMeter:
 
 
 class PRFMeter:
-    """Distributed PR-curve / F-measure accumulator over NBINS thresholds."""
+    """Distributed PR-curve / F-measure accumulator over NBINS thresholds.
+
+    Reference protocol (utils/metric.py per SURVEY §2.1): precision/recall
+    curves are computed PER IMAGE and macro-averaged across the dataset
+    before the F-measure — NOT micro-averaged from pooled TP/FP histograms.
+    The meter therefore sums per-image P/R arrays (linear, so the
+    distributed all-reduce of the sums + image count is exact)."""
 
     def __init__(self) -> None:
-        self.tp = np.zeros(NBINS)
-        self.fp = np.zeros(NBINS)
-        self.npos = 0.0
+        self.prec_sum = np.zeros(NBINS)
+        self.rec_sum = np.zeros(NBINS)
         self.count = 0
 
     @torch.no_grad()
     def update(self, pred: torch.Tensor, mask: torch.Tensor) -> None:
         p = pred.float().clamp(0, 1)
-        gt = mask > 0.5
         for i in range(p.shape[0]):
             pi = p[i].reshape(-1).cpu().numpy()
-            gi = gt[i].reshape(-1).cpu().numpy()
-            hist_fg, _ = np.histogram(pi[gi], bins=NBINS, range=(0, 1))
-            hist_bg, _ = np.histogram(pi[~gi], bins=NBINS, range=(0, 1))
-            self.tp += np.cumsum(hist_fg[::-1])[::-1]
-            self.fp += np.cumsum(hist_bg[::-1])[::-1]
-            self.npos += gi.sum()
+            mi = mask[i].float().reshape(-1).cpu().numpy()
+            prec, rec, _, _ = cal_pr_maxf(pi, mi)
+            self.prec_sum += prec
+            self.rec_sum += rec
             self.count += 1
 
     def compute(self, distributed: bool = False):
-        tp, fp, npos = self.tp, self.fp, self.npos
+        prec_sum, rec_sum, count = self.prec_sum, self.rec_sum, self.count
         if distributed and dist.is_initialized() and dist.get_world_size() > 1:
-            t = torch.from_numpy(np.concatenate([tp, fp, [npos]]))
+            t = torch.from_numpy(
+                np.concatenate([prec_sum, rec_sum, [float(count)]]))
             if torch.cuda.is_available():
                 t = t.cuda()
             dist.all_reduce(t)
             arr = t.cpu().numpy()
-            tp, fp, npos = arr[:NBINS], arr[NBINS:2 * NBINS], arr[-1]
-        prec = tp / np.maximum(tp + fp, 1)
-        rec = tp / max(npos, 1)
+            prec_sum, rec_sum, count = arr[:NBINS], arr[NBINS:2 * NBINS], arr[-1]
+        prec = prec_sum / max(count, 1)
+        rec = rec_sum / max(count, 1)
         f = (1 + BETA2) * prec * rec / np.maximum(BETA2 * prec + rec, 1e-8)
         return {"precision": prec, "recall": rec,
                 "maxf": float(f.max()), "meanf": float(f.mean())}

@@ -59,6 +59,15 @@ def test_meters_match_per_image_fns():
     pf.update(pred, mask)
     out = pf.compute()
     assert 0 <= out["maxf"] <= 1
+    # macro-average protocol: the meter's curve is the MEAN of per-image
+    # P/R curves (reference utils/metric.py protocol), not pooled TP/FP
+    ps, rs = [], []
+    for i in range(3):
+        p_i, r_i, _, _ = cal_pr_maxf(pred[i, 0].numpy(), mask[i, 0].numpy())
+        ps.append(p_i)
+        rs.append(r_i)
+    assert np.allclose(out["precision"], np.mean(ps, axis=0), atol=1e-9)
+    assert np.allclose(out["recall"], np.mean(rs, axis=0), atol=1e-9)
 
 
 def test_cel_and_hel_losses():

@@ -20,7 +20,7 @@ in as deterministic per-index augmentation choices of the generator itself.
 from __future__ import annotations
 
 import torch
-from torch.utils.data import DataLoader, Dataset, DistributedSampler
+from torch.utils.data import DataLoader, Dataset, DistributedSampler, Sampler
 
 
 def _blob_image(g: torch.Generator, size: int, with_depth: bool):
@@ -100,6 +100,25 @@ class SyntheticSOD(Dataset):
         return (*parts, name)
 
 
+class EvalShardSampler(Sampler):
+    """Uneven-shard eval sampler: rank takes ``indices[rank::world]`` with
+    NO padding.  A padded DistributedSampler (drop_last=False) duplicates up
+    to world-1 samples, which double-counts images in the eval meters; here
+    each val image is visited by exactly one rank exactly once."""
+
+    def __init__(self, length: int, rank: int, world_size: int):
+        self._idx = range(rank, length, world_size)
+
+    def __iter__(self):
+        return iter(self._idx)
+
+    def __len__(self) -> int:
+        return len(self._idx)
+
+    def set_epoch(self, epoch: int) -> None:  # API parity; val order is fixed
+        pass
+
+
 def create_loader(cfg, split: str, distributed: bool, rank: int = 0,
                   world_size: int = 1) -> tuple[DataLoader, "DistributedSampler | None"]:
     """Reference-parity loader factory (SURVEY §2.1 create_loader)."""
@@ -108,8 +127,12 @@ def create_loader(cfg, split: str, distributed: bool, rank: int = 0,
     ds = SyntheticSOD(split, cfg.size, length, seed=cfg.data_seed, with_depth=with_depth)
     sampler = None
     if distributed:
-        sampler = DistributedSampler(ds, num_replicas=world_size, rank=rank,
-                                     shuffle=(split == "train"), drop_last=(split == "train"))
+        if split == "train":
+            sampler = DistributedSampler(ds, num_replicas=world_size,
+                                         rank=rank, shuffle=True,
+                                         drop_last=True)
+        else:
+            sampler = EvalShardSampler(length, rank, world_size)
     loader = DataLoader(
         ds,
         batch_size=cfg.batch_size,

@@ -39,3 +39,14 @@ def test_loader_distributed_sharding():
     i0 = list(iter(s0))
     i1 = list(iter(s1))
     assert set(i0).isdisjoint(i1)
+
+
+def test_val_loader_uneven_shards_cover_exactly_once():
+    # 5 val images over 2 ranks: shards of 3 + 2, no padding/duplication
+    cfg = make_config("cpu_plumbing", val_samples=5, batch_size=2, num_workers=0)
+    seen = []
+    for rank in range(2):
+        loader, _ = create_loader(cfg, "val", True, rank=rank, world_size=2)
+        for batch in loader:
+            seen.extend(batch[-1])  # names
+    assert sorted(seen) == [f"val_{i:06d}" for i in range(5)]

@@ -19,7 +19,16 @@ __global__ void sgd_fused_kernel(const long *__restrict__ ptrs,
                                  const long *__restrict__ numels,
                                  const float *__restrict__ wds,
                                  const long *__restrict__ chunks,
-                                 float lr, float mu, float gscale) {
+                                 float lr, float mu, float gscale,
+                                 const float *__restrict__ scale) {
+    if (scale != nullptr) {
+        // device-side dynamic-loss-scale multiplier (fp16 path, sync-free):
+        // 0.0 == overflow detected this step -> skip the whole update
+        // (weight decay / momentum must not advance on a skipped step)
+        float s = scale[0];
+        if (s == 0.0f) return;
+        gscale *= s;
+    }
     long ti = chunks[blockIdx.x * 2 + 0];
     long off = chunks[blockIdx.x * 2 + 1];
     T *param = reinterpret_cast<T *>(ptrs[ti * 4 + 0]);
@@ -45,8 +54,14 @@ __global__ void sgd_fused_kernel(const long *__restrict__ ptrs,
 
 void sgd_fused(torch::Tensor ptrs, torch::Tensor numels, torch::Tensor wds,
                torch::Tensor chunks, int64_t dtype_code, double lr, double mu,
-               double gscale, bool nesterov) {
+               double gscale, bool nesterov, torch::Tensor scale) {
     int K = chunks.size(0);
+    const float *scale_ptr = nullptr;
+    if (scale.defined() && scale.numel() > 0) {
+        TORCH_CHECK(scale.is_cuda() && scale.scalar_type() == torch::kFloat32,
+                    "sgd scale multiplier must be a device fp32 scalar");
+        scale_ptr = scale.data_ptr<float>();
+    }
     auto stream = at::hip::getCurrentHIPStream();
     auto launch = [&](auto t_tag, auto nv) {
         using T = typename decltype(t_tag)::type;
@@ -54,7 +69,7 @@ void sgd_fused(torch::Tensor ptrs, torch::Tensor numels, torch::Tensor wds,
                            dim3(K), dim3(256), 0, stream.stream(),
                            ptrs.data_ptr<long>(), numels.data_ptr<long>(),
                            wds.data_ptr<float>(), chunks.data_ptr<long>(),
-                           (float)lr, (float)mu, (float)gscale);
+                           (float)lr, (float)mu, (float)gscale, scale_ptr);
     };
     struct F32 { using type = float; };
     struct BF { using type = __hip_bfloat16; };

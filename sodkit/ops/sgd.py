@@ -82,16 +82,23 @@ class SodSGD:
             self._tables.append((codes[dtype], ptr_t, numel_t, wd_t, chunk_t))
 
     @torch.no_grad()
-    def step(self, lr: float, grad_scale: float = 1.0):
+    def step(self, lr: float, grad_scale: float = 1.0,
+             scale_t: "torch.Tensor | None" = None):
+        """``scale_t`` (GPU only): a device fp32 scalar multiplied into
+        grad_scale INSIDE the kernel; 0.0 skips the update entirely.  This is
+        the sync-free fp16 dynamic-scaler contract — the overflow decision
+        never touches the host."""
         first = next((p for g in self.groups for p in g["params"]), None)
         if first is None:
             return
         if first.is_cuda:
             if self._tables is None:
                 self._build_tables()
+            st = scale_t if scale_t is not None else torch.Tensor()
             for dt, ptr_t, numel_t, wd_t, chunk_t in self._tables:
                 ext().sgd_fused(ptr_t, numel_t, wd_t, chunk_t, dt, float(lr),
-                                self.momentum, float(grad_scale), self.nesterov)
+                                self.momentum, float(grad_scale),
+                                self.nesterov, st)
         else:
             for g in self.groups:
                 for p in g["params"]:

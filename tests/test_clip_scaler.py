@@ -62,3 +62,44 @@ def test_dynamic_scaler_skips_on_overflow():
     s.reducer.finalize = orig_finalize
     s.step(x, t)  # recovers
     assert s._good_steps == 1
+
+
+def test_device_scaler_state_machine():
+    """The GPU sync-free scaler's tensor state machine, exercised on CPU
+    tensors (identical torch ops; only the fused-SGD consumption is GPU)."""
+    s = _solver(output_root="/tmp/sodscal3", grad_clip=0.0)
+    s.dynamic_scale = True
+    s._use_device_scaler = True
+    s._growth_interval = 2
+    s._scale_t = torch.tensor(64.0)
+    s._good_t = torch.zeros(())
+    s._skips_t = torch.zeros(())
+    g = torch.Generator().manual_seed(2)
+    x = torch.rand(1, 3, 32, 32, generator=g)
+    t = (torch.rand(1, 1, 32, 32, generator=g) > 0.5).float()
+    s.reducer.zero_grad()
+    (s.criterion(s.model(x), t) * s._scale_t).backward()
+    s.reducer.finalize()
+
+    mult = s._device_clip_and_scale()
+    assert abs(mult.item() - 1 / 64.0) < 1e-9   # unscale folded in
+    assert s._scale_t.item() == 64.0 and s._good_t.item() == 1
+
+    # second good step hits the growth interval: scale doubles, counter resets
+    s._device_clip_and_scale()
+    assert s._scale_t.item() == 128.0 and s._good_t.item() == 0
+
+    # overflow: multiplier 0.0 (skip), scale halves, skip counted
+    s.reducer.buckets[0].flat[0] = float("inf")
+    mult = s._device_clip_and_scale()
+    assert mult.item() == 0.0
+    assert s._scale_t.item() == 64.0
+    assert s._skips_t.item() == 1 and s._good_t.item() == 0
+    ls, sk = s._scaler_state()
+    assert ls == 64.0 and sk == 1
+
+    # grad_clip composes: tiny clip forces mult < 1/scale
+    s.reducer.buckets[0].flat[0] = 0.0
+    s.cfg.grad_clip = 1e-8
+    mult = s._device_clip_and_scale()
+    assert 0.0 < mult.item() < 1 / 64.0

@@ -224,6 +224,40 @@ def test_sgd_fused_matches_reference():
                   what=f"sgd step {step}")
 
 
+def test_sgd_fused_device_scale_multiplier():
+    """The sync-free fp16 scaler contract: a device fp32 scalar multiplies
+    grad_scale inside the kernel, and 0.0 skips the update entirely
+    (momentum/weight-decay must not advance on a skipped step)."""
+    from sodkit import ops
+    torch.manual_seed(8)
+    w = torch.randn(33, 17, device="cuda").to(torch.bfloat16)
+    g = torch.randn(33, 17, device="cuda").to(torch.bfloat16)
+
+    def fresh():
+        p = torch.nn.Parameter(w.clone())
+        p.grad = g.clone()
+        return p
+
+    # scale_t = 0.0 -> bitwise no-op (params, master, momentum all untouched)
+    p0 = fresh()
+    opt0 = ops.SodSGD([{"params": [p0], "weight_decay": 1e-2}], momentum=0.9)
+    mom_before = opt0.state[p0]["momentum_buffer"].clone()
+    opt0.step(lr=0.1, grad_scale=1.0,
+              scale_t=torch.zeros(1, device="cuda"))
+    assert torch.equal(p0.detach(), w)
+    assert torch.equal(opt0.state[p0]["momentum_buffer"], mom_before)
+
+    # scale_t = 0.5 on gscale 1.0 == host gscale 0.5
+    pa, pb = fresh(), fresh()
+    opta = ops.SodSGD([{"params": [pa], "weight_decay": 1e-2}], momentum=0.9)
+    optb = ops.SodSGD([{"params": [pb], "weight_decay": 1e-2}], momentum=0.9)
+    opta.step(lr=0.1, grad_scale=1.0,
+              scale_t=torch.full((1,), 0.5, device="cuda"))
+    optb.step(lr=0.1, grad_scale=0.5)
+    assert torch.equal(pa.detach(), pb.detach())
+    assert torch.equal(opta.state[pa]["master"], optb.state[pb]["master"])
+
+
 def test_native_ext_is_loaded_and_mandatory():
     import sodkit._C as C
     assert C.__file__ and "sodkit" in C.__file__

@@ -37,6 +37,36 @@ def test_res2net101_fp16_step():
     assert last == last and last < first * 1.5, (first, last)
 
 
+def test_fp16_device_scaler_skips_on_overflow():
+    """Sync-free fp16 path: an inf gradient must skip the fused-SGD update
+    via the device multiplier (no host round-trip) and halve the scale."""
+    s = _mk_solver("minet_res2net101_512", batch_size=2, size=96,
+                   output_root="/tmp/sodkit_devscal")
+    assert s._use_device_scaler and s.dynamic_scale
+    x, t, _ = _batch(s)
+    s.total_iters = 10**9
+    s.step(x, t)
+    scale0, _ = s._scaler_state()
+    before = [p.detach().clone() for p in s.model.parameters()]
+    orig_finalize = s.reducer.finalize
+
+    def poisoned():
+        orig_finalize()
+        s.reducer.buckets[0].flat.view(-1)[0] = float("inf")
+
+    s.reducer.finalize = poisoned
+    s.step(x, t)
+    s.reducer.finalize = orig_finalize
+    scale1, skips = s._scaler_state()
+    assert scale1 == scale0 / 2 and skips == 1
+    for p, b in zip(s.model.parameters(), before):
+        assert torch.equal(p.detach(), b)
+    s.step(x, t)  # recovers: params move again
+    moved = any(not torch.equal(p.detach(), b)
+                for p, b in zip(s.model.parameters(), before))
+    assert moved
+
+
 def test_minet_vgg16_bf16_step():
     s = _mk_solver("minet_r50_352", model="minet_vgg16", batch_size=2, size=96,
                    output_root="/tmp/sodkit_vgg")

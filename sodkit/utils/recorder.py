@@ -1,13 +1,62 @@
 """Observability recorders — parity with the reference's utils/recorder.py
 (TBRecorder, Timer, XLSXRecorder).  TensorBoard is an
 optional dependency (absent in this image: the recorder degrades to a no-op
-with a one-line notice); results tables always write CSV and additionally
-XLSX when an engine is importable."""
+with a one-line notice); results tables write CSV and XLSX unconditionally —
+the XLSX comes from the dependency-free writer below (an .xlsx file is a zip
+of a few fixed XML parts plus one worksheet), so the reference's results.xlsx
+artifact exists even with no pandas/openpyxl engine installed."""
 
 from __future__ import annotations
 
 import csv
 import os
+import zipfile
+from xml.sax.saxutils import escape
+
+_XLSX_CONTENT_TYPES = (
+    '<?xml version="1.0" encoding="UTF-8" standalone="yes"?>'
+    '<Types xmlns="http://schemas.openxmlformats.org/package/2006/content-types">'
+    '<Default Extension="rels" ContentType="application/vnd.openxmlformats-package.relationships+xml"/>'
+    '<Default Extension="xml" ContentType="application/xml"/>'
+    '<Override PartName="/xl/workbook.xml" ContentType="application/vnd.openxmlformats-officedocument.spreadsheetml.sheet.main+xml"/>'
+    '<Override PartName="/xl/worksheets/sheet1.xml" ContentType="application/vnd.openxmlformats-officedocument.spreadsheetml.worksheet+xml"/>'
+    '</Types>')
+_XLSX_RELS = (
+    '<?xml version="1.0" encoding="UTF-8" standalone="yes"?>'
+    '<Relationships xmlns="http://schemas.openxmlformats.org/package/2006/relationships">'
+    '<Relationship Id="rId1" Type="http://schemas.openxmlformats.org/officeDocument/2006/relationships/officeDocument" Target="xl/workbook.xml"/>'
+    '</Relationships>')
+_XLSX_WORKBOOK = (
+    '<?xml version="1.0" encoding="UTF-8" standalone="yes"?>'
+    '<workbook xmlns="http://schemas.openxmlformats.org/spreadsheetml/2006/main" '
+    'xmlns:r="http://schemas.openxmlformats.org/officeDocument/2006/relationships">'
+    '<sheets><sheet name="results" sheetId="1" r:id="rId1"/></sheets></workbook>')
+_XLSX_WORKBOOK_RELS = (
+    '<?xml version="1.0" encoding="UTF-8" standalone="yes"?>'
+    '<Relationships xmlns="http://schemas.openxmlformats.org/package/2006/relationships">'
+    '<Relationship Id="rId1" Type="http://schemas.openxmlformats.org/officeDocument/2006/relationships/worksheet" Target="worksheets/sheet1.xml"/>'
+    '</Relationships>')
+
+
+def _xlsx_cell(value) -> str:
+    if isinstance(value, (int, float)) and not isinstance(value, bool):
+        return f'<c t="n"><v>{value!r}</v></c>'
+    return f'<c t="inlineStr"><is><t>{escape(str(value))}</t></is></c>'
+
+
+def write_xlsx(path: str, rows: list[list]) -> None:
+    """Write ``rows`` (first row = header) as a single-sheet .xlsx."""
+    body = "".join(
+        "<row>" + "".join(_xlsx_cell(v) for v in row) + "</row>" for row in rows)
+    sheet = ('<?xml version="1.0" encoding="UTF-8" standalone="yes"?>'
+             '<worksheet xmlns="http://schemas.openxmlformats.org/spreadsheetml/2006/main">'
+             f'<sheetData>{body}</sheetData></worksheet>')
+    with zipfile.ZipFile(path, "w", zipfile.ZIP_DEFLATED) as z:
+        z.writestr("[Content_Types].xml", _XLSX_CONTENT_TYPES)
+        z.writestr("_rels/.rels", _XLSX_RELS)
+        z.writestr("xl/workbook.xml", _XLSX_WORKBOOK)
+        z.writestr("xl/_rels/workbook.xml.rels", _XLSX_WORKBOOK_RELS)
+        z.writestr("xl/worksheets/sheet1.xml", sheet)
 
 
 class TBRecorder:
@@ -59,9 +108,5 @@ class ResultsRecorder:
             w.writeheader()
             for r in self.rows:
                 w.writerow(r)
-        try:
-            import pandas as pd  # xlsx parity when an engine exists
-            pd.DataFrame(self.rows).to_excel(
-                os.path.join(self.out_dir, "results.xlsx"), index=False)
-        except Exception:
-            pass
+        write_xlsx(os.path.join(self.out_dir, "results.xlsx"),
+                   [keys] + [[r.get(k, "") for k in keys] for r in self.rows])

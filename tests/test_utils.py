@@ -25,6 +25,14 @@ def test_results_recorder_csv(tmp_path):
     rec.save()
     lines = open(tmp_path / "results.csv").read().strip().split("\n")
     assert lines[0] == "dataset,mae,maxf" and len(lines) == 3
+    # results.xlsx is written unconditionally by the dependency-free writer
+    import zipfile
+    xp = os.path.join(str(tmp_path), "results.xlsx")
+    assert os.path.exists(xp)
+    with zipfile.ZipFile(xp) as z:
+        assert z.testzip() is None
+        sheet = z.read("xl/worksheets/sheet1.xml").decode()
+    assert "dsA" in sheet and "0.9" in sheet
 
 
 def test_tb_recorder_degrades_gracefully(tmp_path):

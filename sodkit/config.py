@@ -33,6 +33,10 @@ class Config:
     batch_size: int = 32                 # per-GPU
     num_workers: int = 8
     data_seed: int = 20260913
+    shuffle: bool = True                 # off => sequential sharding, so an
+    #                                      N-rank global batch is the SAME
+    #                                      sample set as the 1-rank batch
+    #                                      (deterministic DDP parity runs)
 
     # optimization
     epoch_num: int = 3

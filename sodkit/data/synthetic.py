@@ -125,18 +125,21 @@ def create_loader(cfg, split: str, distributed: bool, rank: int = 0,
     with_depth = cfg.model.startswith("hdfnet")
     length = cfg.train_samples if split == "train" else cfg.val_samples
     ds = SyntheticSOD(split, cfg.size, length, seed=cfg.data_seed, with_depth=with_depth)
+    shuffle = getattr(cfg, "shuffle", True)
     sampler = None
     if distributed:
         if split == "train":
+            # shuffle=False shards indices[rank::world]: the union of rank
+            # batches at step k is exactly the 1-rank batch k (parity runs)
             sampler = DistributedSampler(ds, num_replicas=world_size,
-                                         rank=rank, shuffle=True,
+                                         rank=rank, shuffle=shuffle,
                                          drop_last=True)
         else:
             sampler = EvalShardSampler(length, rank, world_size)
     loader = DataLoader(
         ds,
         batch_size=cfg.batch_size,
-        shuffle=(sampler is None and split == "train"),
+        shuffle=(sampler is None and split == "train" and shuffle),
         sampler=sampler,
         num_workers=cfg.num_workers,
         pin_memory=torch.cuda.is_available(),

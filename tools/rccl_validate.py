@@ -1,0 +1,294 @@
+"""RCCL multi-rank validation on a single MI355X (VERDICT r1 top item).
+
+The headline metric is multi-GPU DDP, but the builder's lease is 1 GPU —
+so this script executes the REAL `nccl` (=RCCL on ROCm) code path by
+putting 2 ranks on cuda:0 (sodkit.parallel.dist modulo mapping), which
+exercises exactly the code with zero hardware coverage in round 1:
+`GradReducer._launch`'s comm-stream branch, the coalesced SyncBN
+all-reduce, and the distributed eval reductions — over a real RCCL
+communicator instead of gloo.
+
+Stages (each survives the previous failing; RCCL may refuse duplicate
+GPUs in some versions, in which case stage 1 records the refusal and the
+rest fall back to documenting it):
+
+1. probe   — ws=2 nccl init + all_reduce of a known tensor.
+2. oracle  — lockstep 3-step training of a small conv+SyncBN net on the
+             sodkit kernels: 2-rank half-batches vs 1-proc full batch
+             (loss trajectory match within bf16 tolerance; ranks bitwise
+             identical after each step).
+3. solver  — full Solver train+eval, 2 ranks x batch 4 vs 1 rank x
+             batch 8 on the fixed-seed synthetic val split: the
+             MAE-parity gate (BASELINE.md protocol, within 1e-3), with
+             SODKIT_COMM_STATS overlap stats in the rank logs.
+
+Usage (on the GPU box):  python tools/rccl_validate.py
+Writes gpurun_out/rccl/{summary.json,rank*.log} and prints the summary.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+
+OUT = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                   "gpurun_out", "rccl")
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+# ----------------------------------------------------------------- workers
+def _init_dist(rank: int, world: int, port: int):
+    import datetime
+    import torch
+    import torch.distributed as dist
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.cuda.set_device(rank % torch.cuda.device_count())
+    dist.init_process_group("nccl", rank=rank, world_size=world,
+                            timeout=datetime.timedelta(seconds=120))
+
+
+def worker_probe(rank: int, world: int, port: int):
+    import torch
+    import torch.distributed as dist
+    _init_dist(rank, world, port)
+    t = torch.full((1024,), float(rank + 1), device="cuda")
+    dist.all_reduce(t)
+    torch.cuda.synchronize()
+    want = float(sum(range(1, world + 1)))
+    assert torch.all(t == want), (t[0].item(), want)
+    if rank == 0:
+        print(f"PROBE_OK world={world} sum={t[0].item()} "
+              f"nccl_version={torch.cuda.nccl.version()}")
+    dist.destroy_process_group()
+
+
+def _small_net():
+    import torch
+    from sodkit.models.blocks import ConvBNReLU, SodConv2d
+    torch.manual_seed(3)
+    net = torch.nn.Sequential(
+        ConvBNReLU(3, 32, 3, padding=1),
+        ConvBNReLU(32, 32, 3, padding=1),
+        ConvBNReLU(32, 32, 3, padding=1),
+        SodConv2d(32, 1, 1, bias=True),
+    )
+    return net
+
+
+def _prep_gpu_net(net):
+    import torch
+    from sodkit.models import convert_precision
+    net = net.to("cuda")
+    convert_precision(net, torch.bfloat16)
+    return net
+
+
+def _run_oracle_steps(net, world_batch, ranks_see, steps=3):
+    """Run `steps` of train on `net`; ranks_see = slice of the global batch
+    this process consumes. Returns (losses, checksum)."""
+    import torch
+    import torch.distributed as dist
+    from sodkit.ops import SodSGD, split_decay_groups, hybrid_loss
+    from sodkit.parallel import GradReducer
+    red = GradReducer(net, bucket_mb=1)
+    red.broadcast_params(net)
+    decay, no_decay = split_decay_groups(net)
+    opt = SodSGD([{"params": decay, "weight_decay": 1e-4},
+                  {"params": no_decay, "weight_decay": 0.0}])
+    gw = torch.Generator().manual_seed(123)
+    losses = []
+    for step in range(steps):
+        x_all = torch.rand(world_batch, 3, 64, 64, generator=gw)
+        t_all = (torch.rand(world_batch, 1, 64, 64, generator=gw) > 0.5).float()
+        x = x_all[ranks_see].to("cuda").to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        t = t_all[ranks_see].to("cuda")
+        red.zero_grad()
+        loss = hybrid_loss(net(x), t)
+        loss.backward()
+        red.finalize()
+        opt.step(lr=0.01, grad_scale=red.grad_scale)
+        lt = loss.detach().clone()
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dist.all_reduce(lt)
+            lt /= dist.get_world_size()
+        losses.append(float(lt.item()))
+    cksum = float(sum(p.detach().float().sum().item()
+                      for p in net.parameters()))
+    return losses, cksum
+
+
+def worker_oracle(rank: int, world: int, port: int):
+    import torch
+    import torch.distributed as dist
+    from sodkit.models import set_syncbn
+    os.environ["SODKIT_COMM_STATS"] = "1"
+    _init_dist(rank, world, port)
+    net = _prep_gpu_net(_small_net())
+    set_syncbn(net)
+    half = 2  # per-rank batch; global = 4
+    losses, cksum = _run_oracle_steps(
+        net, world_batch=4, ranks_see=slice(rank * half, (rank + 1) * half))
+    # cross-rank bitwise check: all-gather per-rank checksums
+    cks = [torch.zeros(1, device="cuda") for _ in range(world)]
+    dist.all_gather(cks, torch.tensor([cksum], device="cuda"))
+    if rank == 0:
+        print("ORACLE_2RANK", json.dumps({
+            "losses": losses,
+            "rank_checksums": [float(c.item()) for c in cks]}))
+    dist.destroy_process_group()
+
+
+def worker_oracle_ref():
+    """1-process full-batch reference for the oracle (no dist)."""
+    losses, cksum = _run_oracle_steps(
+        _prep_gpu_net(_small_net()), world_batch=4, ranks_see=slice(0, 4))
+    print("ORACLE_1RANK", json.dumps({"losses": losses, "checksum": cksum}))
+
+
+def worker_solver(rank: int, world: int, port: int, batch: int):
+    """Full Solver train+eval path under nccl (or single process)."""
+    from sodkit.config import make_config
+    from sodkit.engine import Solver
+    os.environ["SODKIT_COMM_STATS"] = "1"
+    if world > 1:
+        os.environ["RANK"] = str(rank)
+        os.environ["LOCAL_RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+    cfg = make_config(
+        "minet_r50_352", size=128, batch_size=batch, epoch_num=2,
+        train_samples=128, val_samples=16, num_workers=2, log_interval=4,
+        sync_bn=True, shuffle=False,  # lockstep global batches: the 2-rank
+        # union at step k IS the 1-rank batch k, so MAE parity is a real
+        # numerics gate, not a training-noise comparison
+        backend="nccl" if world > 1 else "auto",
+        output_root=os.path.join(OUT, f"exp_w{world}"),
+        exp_name=f"rccl_w{world}")
+    s = Solver(cfg)
+    s.train()
+    res = s.test(save_png=False)
+    if rank == 0:
+        print(f"SOLVER_W{world}", json.dumps(
+            {"mae": res["mae"], "maxf": res["maxf"], "world": world}))
+    from sodkit.parallel import cleanup
+    cleanup()
+
+
+# ----------------------------------------------------------------- driver
+def _spawn(stage: str, world: int, port: int, extra=(), timeout=240):
+    """Launch `world` copies of this script as rank workers; returns
+    (ok, rank0_stdout). Kills exact child PIDs on timeout."""
+    procs = []
+    logs = []
+    os.makedirs(OUT, exist_ok=True)
+    for r in range(world):
+        lf = open(os.path.join(OUT, f"{stage}_rank{r}.log"), "w")
+        env = dict(os.environ)
+        env["NCCL_DEBUG"] = "WARN"
+        p = subprocess.Popen(
+            [sys.executable, os.path.abspath(__file__), "--worker", stage,
+             str(r), str(world), str(port), *map(str, extra)],
+            stdout=lf, stderr=subprocess.STDOUT, env=env, cwd=REPO)
+        procs.append(p)
+        logs.append(lf)
+    deadline = time.time() + timeout
+    ok = True
+    for p in procs:
+        left = max(1, deadline - time.time())
+        try:
+            rc = p.wait(timeout=left)
+            ok = ok and rc == 0
+        except subprocess.TimeoutExpired:
+            ok = False
+            for q in procs:
+                if q.poll() is None:
+                    q.send_signal(signal.SIGKILL)
+    for lf in logs:
+        lf.close()
+    out = open(os.path.join(OUT, f"{stage}_rank0.log")).read()
+    return ok, out
+
+
+def main():
+    if len(sys.argv) > 1 and sys.argv[1] == "--worker":
+        stage, rank, world, port = (sys.argv[2], int(sys.argv[3]),
+                                    int(sys.argv[4]), int(sys.argv[5]))
+        if stage == "probe":
+            worker_probe(rank, world, port)
+        elif stage == "oracle":
+            worker_oracle(rank, world, port)
+        elif stage == "oracle_ref":
+            worker_oracle_ref()
+        elif stage == "solver":
+            worker_solver(rank, world, port, batch=int(sys.argv[6]))
+        return
+
+    summary = {"backend": "nccl(RCCL)", "stages": {}}
+    base_port = 29811
+
+    ok, out = _spawn("probe", 2, base_port, timeout=180)
+    probe_ok = ok and "PROBE_OK" in out
+    summary["stages"]["probe"] = {"ok": probe_ok,
+                                  "detail": out.strip().splitlines()[-3:]}
+    print(f"[probe] {'OK' if probe_ok else 'FAILED'}")
+
+    if probe_ok:
+        ok, out = _spawn("oracle", 2, base_port + 1, timeout=300)
+        o2 = o1 = None
+        for line in out.splitlines():
+            if line.startswith("ORACLE_2RANK"):
+                o2 = json.loads(line.split(" ", 1)[1])
+        ok1, out1 = _spawn("oracle_ref", 1, base_port + 2, timeout=300)
+        for line in out1.splitlines():
+            if line.startswith("ORACLE_1RANK"):
+                o1 = json.loads(line.split(" ", 1)[1])
+        st = {"ok": bool(ok and ok1 and o2 and o1)}
+        if o2 and o1:
+            deltas = [abs(a - b) for a, b in zip(o2["losses"], o1["losses"])]
+            st.update({
+                "losses_2rank": o2["losses"], "losses_1rank": o1["losses"],
+                "max_loss_delta": max(deltas),
+                "ranks_bitwise_identical":
+                    o2["rank_checksums"][0] == o2["rank_checksums"][1],
+                "loss_match": max(deltas) < 2e-2,
+            })
+            st["ok"] = st["ok"] and st["loss_match"] and \
+                st["ranks_bitwise_identical"]
+        summary["stages"]["oracle"] = st
+        print(f"[oracle] {st}")
+
+        ok2, out2 = _spawn("solver", 2, base_port + 3, extra=(4,), timeout=600)
+        ok1, out1 = _spawn("solver", 1, base_port + 4, extra=(8,), timeout=600)
+        m2 = m1 = None
+        for line in out2.splitlines():
+            if line.startswith("SOLVER_W2"):
+                m2 = json.loads(line.split(" ", 1)[1])
+        for line in out1.splitlines():
+            if line.startswith("SOLVER_W1"):
+                m1 = json.loads(line.split(" ", 1)[1])
+        st = {"ok": bool(ok2 and ok1 and m2 and m1)}
+        if m2 and m1:
+            st.update({"mae_2rank": m2["mae"], "mae_1rank": m1["mae"],
+                       "mae_delta": abs(m2["mae"] - m1["mae"]),
+                       "mae_parity_1e-3": abs(m2["mae"] - m1["mae"]) < 1e-3})
+        summary["stages"]["solver_parity"] = st
+        print(f"[solver] {st}")
+
+    summary["ok"] = all(s.get("ok") for s in summary["stages"].values())
+    with open(os.path.join(OUT, "summary.json"), "w") as f:
+        json.dump(summary, f, indent=1)
+    print("RCCL_SUMMARY " + json.dumps(summary))
+
+
+if __name__ == "__main__":
+    main()

@@ -23,12 +23,24 @@ from ._backend import ext
 
 _CL = torch.channels_last
 
-# opt-in 8-phase 256-square conv backend (csrc/conv_igemm8.hip,
-# docs/IGEMM8_PLAN.md).  Default OFF this round: refcheck-verified on
-# MI355X but not yet A/B'd across the full shape table.  Eligible shapes
-# (Cout % 256 == 0 fwd / Cin % 256 == 0 dgrad) route to it when set;
-# everything else stays on the 128-square kernel either way.
-_USE_8P = os.environ.get("SODKIT_IGEMM_8P", "0") == "1"
+# 8-phase 256-square conv backend (csrc/conv_igemm8.hip, docs/
+# IGEMM8_PLAN.md).  Measured A/B across the model shape table
+# (gpurun_out/ubench_conv r2): the 256x256 tile only beats the 128-square
+# kernel when the launch fills the chip — its win shapes have
+# (ceil(M/256) * Cout/256) >= ~256 workgroups (e.g. 1x1 expand 64->256
+# @88x88: 143 vs 114 TF); below that it underfills 256 CUs and LOSES
+# (3x3 512ch @11x11: 110 vs 323 TF).  Dispatch is therefore gated on the
+# workgroup count; SODKIT_IGEMM_8P=1/0 forces it on/off for A/B runs.
+_8P_ENV = os.environ.get("SODKIT_IGEMM_8P", "")
+_8P_MIN_WG = 256
+
+
+def _use_8p(m_tiles: int, cout: int) -> bool:
+    if _8P_ENV == "0":
+        return False
+    if _8P_ENV == "1":
+        return True
+    return m_tiles * (cout // 256) >= _8P_MIN_WG
 
 
 def _cl_ld(x: torch.Tensor):
@@ -55,7 +67,16 @@ class _ConvFn(torch.autograd.Function):
     def forward(ctx, x, w, bias, stride, padding, dilation):
         x, ldx = _cl_ld(x)
         b = bias if bias is not None else torch.Tensor()
-        if _USE_8P and w.shape[0] % 256 == 0 and w.shape[3] % 8 == 0:
+        if w.shape[0] % 256 == 0 and w.shape[3] % 8 == 0:
+            p = (x.shape[2] + 2 * padding - dilation * (w.shape[1] - 1) - 1) \
+                // stride + 1
+            q = (x.shape[3] + 2 * padding - dilation * (w.shape[2] - 1) - 1) \
+                // stride + 1
+            m_tiles = -(-x.shape[0] * p * q // 256)
+            use8 = _use_8p(m_tiles, w.shape[0])
+        else:
+            use8 = False
+        if use8:
             y = ext().conv2d_fwd_8p(x, w, b, stride, padding, dilation, ldx)
         else:
             y = ext().conv2d_fwd(x, w, b, stride, padding, dilation, ldx)
@@ -72,7 +93,9 @@ class _ConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             # dgrad A-operand layout [Cin, R, S, Cout] (a small permuted copy)
             w_t = w.permute(3, 1, 2, 0).contiguous()
-            if _USE_8P and w_t.shape[0] % 256 == 0 and w_t.shape[3] % 8 == 0:
+            if w_t.shape[0] % 256 == 0 and w_t.shape[3] % 8 == 0 and _use_8p(
+                    -(-x.shape[0] * x.shape[2] * x.shape[3] // 256),
+                    w_t.shape[0]):
                 dx = ext().conv2d_dgrad_8p(dy, w_t, x.shape[2], x.shape[3],
                                            stride, padding, dilation, lddy)
             else:

@@ -547,11 +547,11 @@ def test_conv8p_matches_default_kernel_e2e():
     w = (torch.randn(256, 3, 3, 64, device="cuda") * 0.1).to(torch.bfloat16)
     ref = F.conv2d(x.float(), w.float().permute(0, 3, 1, 2), None, padding=1)
     y0 = ops.conv2d(x, w, None, 1, 1, 1)
-    prev = conv_mod._USE_8P
-    conv_mod._USE_8P = True
+    prev = conv_mod._8P_ENV
+    conv_mod._8P_ENV = "1"
     try:
         y1 = ops.conv2d(x, w, None, 1, 1, 1)
     finally:
-        conv_mod._USE_8P = prev
+        conv_mod._8P_ENV = prev
     close(y0, ref, what="default vs ref")
     close(y1, ref, what="8p vs ref")

@@ -156,12 +156,12 @@ def test_minet_step_with_8p_backend(tmp_path):
     s.total_iters = 100
     x, t, _ = _batch(s, n=2, size=96)
     l0 = s.step(x, t).item()
-    prev = conv_mod._USE_8P
-    conv_mod._USE_8P = True
+    prev = conv_mod._8P_ENV
+    conv_mod._8P_ENV = "1"
     try:
         l1 = s.step(x, t).item()
         l2 = s.step(x, t).item()
     finally:
-        conv_mod._USE_8P = prev
+        conv_mod._8P_ENV = prev
     assert l0 == l0 and l1 == l1 and l2 == l2, "NaN loss under 8p"
     assert l2 < l0 * 1.5, (l0, l1, l2)   # still optimizing, not diverging

@@ -16,8 +16,13 @@ SHAPES = [
     ("l1_3x3",     32, 64, 88, 64, 3, 1, 1, 1),
     ("l1_1x1e",    32, 64, 88, 256, 1, 1, 0, 1),
     ("l2_3x3",     32, 128, 44, 128, 3, 1, 1, 1),
+    ("l2_1x1r",    32, 512, 44, 128, 1, 1, 0, 1),
+    ("l2_1x1e",    32, 128, 44, 512, 1, 1, 0, 1),
     ("l3_3x3",     32, 256, 22, 256, 3, 1, 1, 1),
+    ("l3_1x1r",    32, 1024, 22, 256, 1, 1, 0, 1),
+    ("l3_1x1e",    32, 256, 22, 1024, 1, 1, 0, 1),
     ("l4_3x3",     32, 512, 11, 512, 3, 1, 1, 1),
+    ("l4_1x1r",    32, 2048, 11, 512, 1, 1, 0, 1),
     ("l4_1x1e",    32, 512, 11, 2048, 1, 1, 0, 1),
     ("dec_3x3_176", 32, 64, 176, 64, 3, 1, 1, 1),
     ("dec_3x3_88", 32, 64, 88, 64, 3, 1, 1, 1),
@@ -38,7 +43,7 @@ def bench(fn, iters=10):
 
 def main():
     dev = "cuda"
-    print(f"{'shape':>12} {'M':>8} {'K':>6} {'N':>6} | {'fwd_tf':>7} {'dgrad_tf':>8} {'wgrad_tf':>8} | {'fwd8p':>7} {'dgrad8p':>7}")
+    print(f"{'shape':>12} {'M':>8} {'K':>6} {'N':>6} | {'fwd_tf':>7} {'dgrad_tf':>8} {'wgrad_tf':>8} | {'fwd8p':>7} {'dgrad8p':>7} wg8")
     for name, N, Cin, H, Cout, k, st, pad, dil in SHAPES:
         x = torch.randn(N, Cin, H, H, device=dev).to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)
@@ -54,8 +59,10 @@ def main():
         w_t = w.permute(3, 1, 2, 0).contiguous()
         t_d = bench(lambda: _C.conv2d_dgrad(y, w_t, H, H, st, pad, dil, Cout))
         t_w = bench(lambda: _C.conv2d_wgrad(x, y, k, k, st, pad, dil, Cin, Cout))
-        # 8-phase opt-in backend A/B where the shape is eligible
+        # 8-phase backend A/B where the shape is eligible (wg8 = its launch
+        # size; the dispatch gate needs >= ~256 to fill the chip)
         f8 = d8 = "      -"
+        wg8 = ((M + 255) // 256) * (Cout // 256)
         if Cout % 256 == 0 and Cin % 8 == 0:
             t = bench(lambda: _C.conv2d_fwd_8p(x, w, nob, st, pad, dil, Cin))
             f8 = f"{flop / t / 1e12:7.0f}"
@@ -65,7 +72,7 @@ def main():
             d8 = f"{flop / t / 1e12:7.0f}"
         print(f"{name:>12} {M:>8} {K:>6} {Cout:>6} | "
               f"{flop / t_f / 1e12:7.0f} {flop / t_d / 1e12:8.0f} "
-              f"{flop / t_w / 1e12:8.0f} | {f8} {d8}")
+              f"{flop / t_w / 1e12:8.0f} | {f8} {d8} wg8={wg8}")
 
 
 if __name__ == "__main__":

@@ -31,8 +31,11 @@ _CL = torch.channels_last
 # @88x88: 143 vs 114 TF); below that it underfills 256 CUs and LOSES
 # (3x3 512ch @11x11: 110 vs 323 TF).  Dispatch is therefore gated on the
 # workgroup count; SODKIT_IGEMM_8P=1/0 forces it on/off for A/B runs.
+# Threshold from the full r2 table (gpurun_out/ubench_conv.log): every
+# measured shape at >= 244 wgs wins (l3_1x1e fwd +30%, l3_1x1r dgrad
+# +23%, l2_1x1e, l1_1x1e), every shape at <= 128 loses; 232 splits them.
 _8P_ENV = os.environ.get("SODKIT_IGEMM_8P", "")
-_8P_MIN_WG = 256
+_8P_MIN_WG = 232
 
 
 def _use_8p(m_tiles: int, cout: int) -> bool:

@@ -38,9 +38,10 @@ import subprocess
 import sys
 import time
 
-OUT = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
-                   "gpurun_out", "rccl")
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+OUT = os.path.join(REPO, "gpurun_out", "rccl")
+sys.path.insert(0, REPO)   # workers run as `python tools/<me> --worker`:
+#                            python puts tools/ on sys.path, not the repo
 
 
 # ----------------------------------------------------------------- workers

@@ -117,7 +117,6 @@ void wgrad_dir_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     // fragment read geometry (lane&15 = row-in-16, lane>>4 = k octet)
     const int arow = wm * 32 + (lane & 15);          // dy rows for mf=0
     const int brow = wj * 16 + (lane & 15);          // x^T c row
-    const char *brow_ptr0 = nullptr;                 // set per window row
 
     __syncthreads();
 
@@ -129,10 +128,11 @@ void wgrad_dir_kernel(const T *__restrict__ x, const T *__restrict__ dy,
         s16x8 xv[3][4], dyv[4];
         int my_pq = -1, my_c8 = -1;
         {
-            int ch = ((t >> 5) & 7) * n_pq + ((t & 31) + ((t >> 8) << 5));
-            // ch -> (c8, pq); threads whose pq exceeds the row are idle
-            int c8 = ch / n_pq, pq = ch % n_pq;
-            bool on = ((t & 31) + ((t >> 8) << 5)) < n_pq && c8 < 8;
+            // thread -> (c8 group, pixel quad); this split keeps the b64
+            // transpose-write bank conflicts <= 2-way (header)
+            int c8 = (t >> 5) & 7;
+            int pq = (t & 31) + ((t >> 8) << 5);
+            bool on = pq < n_pq;
             if (on) { my_pq = pq; my_c8 = c8; }
 #pragma unroll
             for (int d = 0; d < 3; ++d) {
@@ -193,7 +193,6 @@ void wgrad_dir_kernel(const T *__restrict__ x, const T *__restrict__ dy,
         __syncthreads();
 
         // ---- MFMA sweep over the q k-tiles ----
-        const char *br = nullptr;
         for (int kt = 0; kt < ktiles; ++kt) {
             const int kb = kt * 64 + ((lane >> 4) << 4);   // dy k byte
             s16x8 a0 = read16(dybuf + arow * dypitch, kb);
@@ -201,7 +200,7 @@ void wgrad_dir_kernel(const T *__restrict__ x, const T *__restrict__ dy,
             const int xb = kb + 16;                        // + q0 element pad
 #pragma unroll
             for (int r = 0; r < 3; ++r) {
-                br = xbuf[r] + brow * xpitch;
+                const char *br = xbuf[r] + brow * xpitch;
                 s16x8 b0 = shifted_read<0>(br, xb);
                 s16x8 b1 = shifted_read<1>(br, xb);
                 s16x8 b2 = shifted_read<2>(br, xb);
@@ -273,7 +272,9 @@ inline int bankfree_pitch(int need) {
 bool wgrad_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
                         int64_t dil, int64_t C, int64_t W) {
     if (R != 3 || S != 3 || stride != 1 || pad != 1 || dil != 1) return false;
-    if (C % 16 != 0) return false;
+    if (C % 64 != 0) return false;   // stage reads 8c vectors across a full
+    //                                  64-row chunk; partial chunks would
+    //                                  read past the channel range
     if (W > 184 || W < 8) return false;
     int ktiles = (int)((W + 31) / 32);
     int xpitch = bankfree_pitch((ktiles * 32 + 24) * 2);
@@ -291,11 +292,8 @@ torch::Tensor conv2d_wgrad_dir(torch::Tensor x, torch::Tensor dy,
     TORCH_CHECK(wgrad_dir_eligible(3, 3, 1, 1, 1, C, W), "shape not eligible");
     const int Kw = 9 * C;
     const int total_rows = N * H;
-    const int cc_chunks = C / 64 + (C % 64 ? 1 : 0);
-    TORCH_CHECK(C % 64 == 0 || C == 16 || C == 32 || C == 48,
-                "C must be 64k or a single sub-64 chunk");
     const int cq_chunks = (Cout + 63) / 64;
-    const int chunks = cq_chunks * (C + 63) / 64;
+    const int chunks = cq_chunks * (C / 64);
     int splitk = std::max(1, std::min(512, 512 / std::max(chunks, 1)));
     splitk = std::min(splitk, total_rows);
     int rows_per_z = (total_rows + splitk - 1) / splitk;
@@ -312,7 +310,7 @@ torch::Tensor conv2d_wgrad_dir(torch::Tensor x, torch::Tensor dy,
     auto dw = torch::empty({Cout, 3, 3, C}, x.options());
     long n_elems = (long)Cout * Kw;
     int rblocks = std::min<long>((n_elems + 255) / 256, 2048);
-    dim3 grid(cq_chunks * ((C + 63) / 64), splitk);
+    dim3 grid(cq_chunks * (C / 64), splitk);
 
     AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_wgrad_dir",
         AT_DISPATCH_CASE(at::kBFloat16, [&] {
@@ -331,7 +329,7 @@ torch::Tensor conv2d_wgrad_dir(torch::Tensor x, torch::Tensor dy,
                                partial.data_ptr<float>(), N, H, W, C,
                                (int)ldx, (int)lddy, Cout, ktiles, xpitch,
                                dypitch, rows_per_z, total_rows,
-                               (C + 63) / 64, Kw);
+                               C / 64, Kw);
             hipLaunchKernelGGL((sodkit::wdir::wdir_reduce_kernel<T>),
                                dim3(rblocks), dim3(256), 0, stream.stream(),
                                partial.data_ptr<float>(),
@@ -354,7 +352,7 @@ torch::Tensor conv2d_wgrad_dir(torch::Tensor x, torch::Tensor dy,
                                partial.data_ptr<float>(), N, H, W, C,
                                (int)ldx, (int)lddy, Cout, ktiles, xpitch,
                                dypitch, rows_per_z, total_rows,
-                               (C + 63) / 64, Kw);
+                               C / 64, Kw);
             hipLaunchKernelGGL((sodkit::wdir::wdir_reduce_kernel<T>),
                                dim3(rblocks), dim3(256), 0, stream.stream(),
                                partial.data_ptr<float>(),

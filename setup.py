@@ -37,6 +37,7 @@ SRC = [
     "csrc/bind.cpp",
     "csrc/conv_igemm.hip",
     "csrc/conv_igemm8.hip",
+    "csrc/wgrad_dir.hip",
     "csrc/bnrelu.hip",
     "csrc/bilinear.hip",
     "csrc/loss.hip",

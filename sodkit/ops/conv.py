@@ -46,6 +46,21 @@ def _use_8p(m_tiles: int, cout: int) -> bool:
     return m_tiles * (cout // 256) >= _8P_MIN_WG
 
 
+# direct sliding-window wgrad for 3x3 s1 d1 p1 (csrc/wgrad_dir.hip): stages
+# each x row once and computes all nine taps from shifted LDS reads instead
+# of re-gathering per tap.  SODKIT_WGRAD_DIR=0 disables (A/B).
+_WDIR_ENV = os.environ.get("SODKIT_WGRAD_DIR", "")
+
+
+def _use_wgrad_dir(w: torch.Tensor, x: torch.Tensor, stride, padding,
+                   dilation) -> bool:
+    if _WDIR_ENV == "0":
+        return False
+    return (w.shape[1] == 3 and w.shape[2] == 3 and stride == 1
+            and padding == 1 and dilation == 1 and w.shape[3] % 64 == 0
+            and 8 <= x.shape[3] <= 184)
+
+
 def _cl_ld(x: torch.Tensor):
     """Return (tensor, channel pitch).  A channel-slice of a channels_last
     tensor (torch.split / cat-backward narrow views) keeps NHWC pixel layout
@@ -105,9 +120,12 @@ class _ConvFn(torch.autograd.Function):
                 dx = ext().conv2d_dgrad(dy, w_t, x.shape[2], x.shape[3],
                                         stride, padding, dilation, lddy)
         if ctx.needs_input_grad[1]:
-            dw = ext().conv2d_wgrad(x, dy, w.shape[1], w.shape[2],
-                                    stride, padding, dilation, ldx,
-                                    lddy).to(w.dtype)
+            if _use_wgrad_dir(w, x, stride, padding, dilation):
+                dw = ext().conv2d_wgrad_dir(x, dy, ldx, lddy).to(w.dtype)
+            else:
+                dw = ext().conv2d_wgrad(x, dy, w.shape[1], w.shape[2],
+                                        stride, padding, dilation, ldx,
+                                        lddy).to(w.dtype)
         if has_bias and ctx.needs_input_grad[2]:
             db = ext().channel_sum(dy, lddy).to(w.dtype)
         return dx, dw, db, None, None, None

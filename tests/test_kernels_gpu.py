@@ -224,6 +224,57 @@ def test_sgd_fused_matches_reference():
                   what=f"sgd step {step}")
 
 
+WDIR_SHAPES = [
+    # (N, C, H, Cout): 3x3 s1 d1 p1 only — the kernel's whole domain
+    (2, 64, 33, 64),     # decoder/AIM class, odd W (k-tail + stage tail)
+    (2, 64, 176, 64),    # the big dec_3x3_176 class (max supported W)
+    (2, 128, 44, 128),   # layer2 class, multi-chunk C and Cout
+    (2, 64, 40, 32),     # Cout < 64 (masked cout rows, e.g. head conv)
+    (1, 256, 22, 64),    # layer3-C with small W (k-tile padding)
+    (3, 64, 11, 72),     # Cout % 8 != 0 tail + tiny W
+]
+
+
+@pytest.mark.parametrize("shape", WDIR_SHAPES)
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_wgrad_dir_matches_reference(shape, dtype):
+    """Direct sliding-window wgrad vs fp32 torch on its eligible domain."""
+    from sodkit import _C
+    N, C, H, Cout = shape
+    torch.manual_seed(7)
+    x = torch.randn(N, C, H, H, device="cuda").to(dtype)
+    dy = torch.randn(N, Cout, H, H, device="cuda").to(dtype)
+    assert _C.wgrad_dir_eligible(3, 3, 1, 1, 1, C, H)
+    dw = _C.conv2d_wgrad_dir(_cl(x), _cl(dy), C, Cout)
+    # reference wgrad via autograd on the weight
+    wr = torch.zeros(Cout, C, 3, 3, device="cuda", requires_grad=True)
+    yr = F.conv2d(x.float(), wr, None, padding=1)
+    yr.backward(dy.float())
+    close(dw, wr.grad.permute(0, 2, 3, 1), rtol=3e-2, atol=3e-2,
+          what=f"wgrad_dir {shape} {dtype}")
+
+
+def test_wgrad_dir_dispatch_is_used():
+    """The conv backward must actually route eligible 3x3 wgrads through
+    the direct kernel (guards against a silently-dead dispatch)."""
+    from sodkit.ops import conv as conv_mod
+    from sodkit import ops
+    torch.manual_seed(3)
+    x = _cl(torch.randn(2, 64, 40, 40, device="cuda").to(torch.bfloat16))
+    w = (torch.randn(64, 3, 3, 64, device="cuda") * 0.1).to(torch.bfloat16)
+    assert conv_mod._use_wgrad_dir(w, x, 1, 1, 1)
+    xg = x.clone().requires_grad_()
+    wg = w.clone().requires_grad_()
+    y = ops.conv2d(xg, wg, None, 1, 1, 1)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    wr = w.float().permute(0, 3, 1, 2).detach().requires_grad_()
+    yr = F.conv2d(x.float(), wr, None, padding=1)
+    yr.backward(gy.float())
+    close(wg.grad, wr.grad.permute(0, 2, 3, 1), rtol=3e-2, atol=3e-2,
+          what="dispatched wgrad_dir")
+
+
 def test_sgd_fused_device_scale_multiplier():
     """The sync-free fp16 scaler contract: a device fp32 scalar multiplies
     grad_scale inside the kernel, and 0.0 skips the update entirely

@@ -43,7 +43,7 @@ def bench(fn, iters=10):
 
 def main():
     dev = "cuda"
-    print(f"{'shape':>12} {'M':>8} {'K':>6} {'N':>6} | {'fwd_tf':>7} {'dgrad_tf':>8} {'wgrad_tf':>8} | {'fwd8p':>7} {'dgrad8p':>7} wg8")
+    print(f"{'shape':>12} {'M':>8} {'K':>6} {'N':>6} | {'fwd_tf':>7} {'dgrad_tf':>8} {'wgrad_tf':>8} | {'fwd8p':>7} {'dgrad8p':>7} {'wdir':>7} wg8")
     for name, N, Cin, H, Cout, k, st, pad, dil in SHAPES:
         x = torch.randn(N, Cin, H, H, device=dev).to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)
@@ -70,9 +70,13 @@ def main():
             t = bench(lambda: _C.conv2d_dgrad_8p(y, w_t, H, H, st, pad, dil,
                                                  Cout))
             d8 = f"{flop / t / 1e12:7.0f}"
+        wd = "      -"
+        if _C.wgrad_dir_eligible(k, k, st, pad, dil, Cin, H):
+            t = bench(lambda: _C.conv2d_wgrad_dir(x, y, Cin, Cout))
+            wd = f"{flop / t / 1e12:7.0f}"
         print(f"{name:>12} {M:>8} {K:>6} {Cout:>6} | "
               f"{flop / t_f / 1e12:7.0f} {flop / t_d / 1e12:8.0f} "
-              f"{flop / t_w / 1e12:8.0f} | {f8} {d8} wg8={wg8}")
+              f"{flop / t_w / 1e12:8.0f} | {f8} {d8} {wd} wg8={wg8}")
 
 
 if __name__ == "__main__":

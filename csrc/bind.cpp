@@ -33,16 +33,19 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x, double n,
                                              double momentum, double eps,
                                              torch::Tensor running_mean,
                                              torch::Tensor running_var);
-torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd,
-                       torch::Tensor gamma, torch::Tensor beta, bool relu,
-                       torch::Tensor residual);
+std::vector<torch::Tensor> bn_apply(torch::Tensor x, torch::Tensor mean,
+                                    torch::Tensor invstd, torch::Tensor gamma,
+                                    torch::Tensor beta, bool relu,
+                                    torch::Tensor residual, bool want_mask);
 torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
-                            torch::Tensor mean, torch::Tensor invstd, bool relu);
+                            torch::Tensor mean, torch::Tensor invstd, bool relu,
+                            torch::Tensor mask);
 std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor dy, torch::Tensor y,
                                         torch::Tensor x, torch::Tensor mean,
                                         torch::Tensor invstd, torch::Tensor gamma,
                                         torch::Tensor sums, double n_total,
-                                        bool relu, bool need_dres);
+                                        bool relu, bool need_dres,
+                                        torch::Tensor mask);
 
 torch::Tensor bilinear_fwd(torch::Tensor x, int64_t Ho, int64_t Wo,
                            torch::Tensor add);

@@ -217,7 +217,13 @@ __global__ void bn_apply_vec_kernel(const T *__restrict__ x,
                                     const float *__restrict__ invstd,
                                     const float *__restrict__ gamma,
                                     const float *__restrict__ beta,
-                                    T *__restrict__ y, long total8, int C8) {
+                                    T *__restrict__ y,
+                                    unsigned char *__restrict__ mask,
+                                    long total8, int C8) {
+    // mask != nullptr (training, RELU): emit a 1-bit-per-element ReLU-mask
+    // plane (one byte per 8-channel vector) so the backward kernels read
+    // 1/16 of y's bytes instead of y itself (bn_bwd_reduce was 3-tensor
+    // memory-bound at ~2.7 TB/s; this removes a third of its traffic).
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long stride = (long)gridDim.x * blockDim.x;
     for (; i < total8; i += stride) {
@@ -226,6 +232,7 @@ __global__ void bn_apply_vec_kernel(const T *__restrict__ x,
         s16x8 vr;
         if (RES) vr = *reinterpret_cast<const s16x8 *>(res + i * 8);
         s16x8 vo;
+        unsigned int mb = 0;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
             float sc = invstd[c0 + e] * gamma[c0 + e];
@@ -235,8 +242,12 @@ __global__ void bn_apply_vec_kernel(const T *__restrict__ x,
             if (RELU) v = fmaxf(v, 0.f);
             T tv = ElemTraits<T>::from_f32(v);
             vo[e] = *reinterpret_cast<const short *>(&tv);
+            if (RELU && mask != nullptr)
+                // bit from the ROUNDED value: bit-exact with the y>0 path
+                mb |= (ElemTraits<T>::to_f32(tv) > 0.f ? 1u : 0u) << e;
         }
         *reinterpret_cast<s16x8 *>(y + i * 8) = vo;
+        if (RELU && mask != nullptr) mask[i] = (unsigned char)mb;
     }
 }
 
@@ -263,13 +274,17 @@ __global__ void bn_apply_kernel(const T *__restrict__ x,
 
 // ---- backward reduce: g = dy * relumask(y); out[0][c]=sum g,
 //      out[1][c]=sum g*xhat ----
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool MASKED>
 __global__ void bn_bwd_reduce_vec_kernel(const T *__restrict__ dy,
                                          const T *__restrict__ y,
                                          const T *__restrict__ x,
+                                         const unsigned char *__restrict__ mask,
                                          const float *__restrict__ mean,
                                          const float *__restrict__ invstd,
                                          float *__restrict__ out, long M, int C) {
+    // MASKED: the ReLU mask comes from bn_apply's bit plane (1/16 of y's
+    // bytes) instead of re-reading y — drops a third of the kernel's
+    // memory traffic (BASELINE "bn_bwd_reduce" lever).
     __shared__ float red[256 * 16];
     int C8 = C >> 3;
     int cc0 = blockIdx.y * 256;
@@ -281,6 +296,7 @@ __global__ void bn_bwd_reduce_vec_kernel(const T *__restrict__ dy,
 #pragma unroll
     for (int e = 0; e < 16; ++e) acc[e] = 0.f;
     int c0 = (cc0 + lane_c) * 8;
+    const long moff = cc0 + lane_c;          // mask plane column
     float mu[8], is[8];
 #pragma unroll
     for (int e = 0; e < 8; ++e) { mu[e] = mean[c0 + e]; is[e] = invstd[c0 + e]; }
@@ -295,15 +311,28 @@ __global__ void bn_bwd_reduce_vec_kernel(const T *__restrict__ dy,
             s16x8 vx0 = *reinterpret_cast<const s16x8 *>(x + i0);
             s16x8 vx1 = *reinterpret_cast<const s16x8 *>(x + i1);
             s16x8 vy0, vy1;
+            unsigned int mb0 = 0xff, mb1 = 0xff;
             if (RELU) {
-                vy0 = *reinterpret_cast<const s16x8 *>(y + i0);
-                vy1 = *reinterpret_cast<const s16x8 *>(y + i1);
+                if (MASKED) {
+                    mb0 = mask[m * C8 + moff];
+                    mb1 = mask[(m + mstep) * C8 + moff];
+                } else {
+                    vy0 = *reinterpret_cast<const s16x8 *>(y + i0);
+                    vy1 = *reinterpret_cast<const s16x8 *>(y + i1);
+                }
             }
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
                 float g0 = s16_as_f32<T>(vdy0[e]), g1 = s16_as_f32<T>(vdy1[e]);
-                if (RELU && s16_as_f32<T>(vy0[e]) <= 0.f) g0 = 0.f;
-                if (RELU && s16_as_f32<T>(vy1[e]) <= 0.f) g1 = 0.f;
+                if (RELU) {
+                    if (MASKED) {
+                        if (!((mb0 >> e) & 1)) g0 = 0.f;
+                        if (!((mb1 >> e) & 1)) g1 = 0.f;
+                    } else {
+                        if (s16_as_f32<T>(vy0[e]) <= 0.f) g0 = 0.f;
+                        if (s16_as_f32<T>(vy1[e]) <= 0.f) g1 = 0.f;
+                    }
+                }
                 float xh0 = (s16_as_f32<T>(vx0[e]) - mu[e]) * is[e];
                 float xh1 = (s16_as_f32<T>(vx1[e]) - mu[e]) * is[e];
                 acc[e] += g0 + g1;
@@ -314,11 +343,18 @@ __global__ void bn_bwd_reduce_vec_kernel(const T *__restrict__ dy,
             long i = m * C + off0;
             s16x8 vdy = *reinterpret_cast<const s16x8 *>(dy + i);
             s16x8 vy, vx = *reinterpret_cast<const s16x8 *>(x + i);
-            if (RELU) vy = *reinterpret_cast<const s16x8 *>(y + i);
+            unsigned int mb = 0xff;
+            if (RELU) {
+                if (MASKED) mb = mask[m * C8 + moff];
+                else vy = *reinterpret_cast<const s16x8 *>(y + i);
+            }
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
                 float g = s16_as_f32<T>(vdy[e]);
-                if (RELU && s16_as_f32<T>(vy[e]) <= 0.f) g = 0.f;
+                if (RELU) {
+                    if (MASKED) { if (!((mb >> e) & 1)) g = 0.f; }
+                    else if (s16_as_f32<T>(vy[e]) <= 0.f) g = 0.f;
+                }
                 float xh = (s16_as_f32<T>(vx[e]) - mu[e]) * is[e];
                 acc[e] += g;
                 acc[8 + e] += g * xh;
@@ -366,10 +402,11 @@ __global__ void bn_bwd_reduce_kernel(const T *__restrict__ dy,
 }
 
 // ---- vectorized backward apply (C % 8 == 0) ----
-template <typename T, bool RELU, bool NEED_DRES>
+template <typename T, bool RELU, bool NEED_DRES, bool MASKED>
 __global__ void bn_bwd_apply_vec_kernel(const T *__restrict__ dy,
                                         const T *__restrict__ y,
                                         const T *__restrict__ x,
+                                        const unsigned char *__restrict__ mask,
                                         const float *__restrict__ mean,
                                         const float *__restrict__ invstd,
                                         const float *__restrict__ gamma,
@@ -385,13 +422,20 @@ __global__ void bn_bwd_apply_vec_kernel(const T *__restrict__ dy,
         s16x8 vdy = *reinterpret_cast<const s16x8 *>(dy + i * 8);
         s16x8 vx = *reinterpret_cast<const s16x8 *>(x + i * 8);
         s16x8 vy;
-        if (RELU) vy = *reinterpret_cast<const s16x8 *>(y + i * 8);
+        unsigned int mb = 0xff;
+        if (RELU) {
+            if (MASKED) mb = mask[i];
+            else vy = *reinterpret_cast<const s16x8 *>(y + i * 8);
+        }
         s16x8 vo, vdr;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
             int c = c0 + e;
             float g = s16_as_f32<T>(vdy[e]);
-            if (RELU && s16_as_f32<T>(vy[e]) <= 0.f) g = 0.f;
+            if (RELU) {
+                if (MASKED) { if (!((mb >> e) & 1)) g = 0.f; }
+                else if (s16_as_f32<T>(vy[e]) <= 0.f) g = 0.f;
+            }
             if (NEED_DRES) {
                 T tg = ElemTraits<T>::from_f32(g);
                 vdr[e] = *reinterpret_cast<const short *>(&tg);
@@ -560,13 +604,18 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x, double n,
     return {mean, invstd};
 }
 
-torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd,
-                       torch::Tensor gamma, torch::Tensor beta, bool relu,
-                       torch::Tensor residual) {
+std::vector<torch::Tensor> bn_apply(torch::Tensor x, torch::Tensor mean,
+                                    torch::Tensor invstd, torch::Tensor gamma,
+                                    torch::Tensor beta, bool relu,
+                                    torch::Tensor residual, bool want_mask) {
     TORCH_CHECK(x.is_cuda() && x.dim() == 4);
     int C = x.size(1);
     long total = x.numel();
     auto y = torch::empty_like(x);
+    torch::Tensor mask;   // ReLU bit plane for the backward kernels
+    bool emit_mask = want_mask && relu && C % 8 == 0 && x.element_size() == 2;
+    if (emit_mask)
+        mask = torch::empty({total / 8}, x.options().dtype(torch::kByte));
     auto stream = at::hip::getCurrentHIPStream();
     bool has_res = residual.defined() && residual.numel() > 0;
     auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
@@ -584,7 +633,9 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd
                     reinterpret_cast<const T *>(x.data_ptr()), resp,
                     meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
                     gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
-                    reinterpret_cast<T *>(y.data_ptr()), total8, C / 8);
+                    reinterpret_cast<T *>(y.data_ptr()),
+                    emit_mask ? mask.data_ptr<unsigned char>() : nullptr,
+                    total8, C / 8);
             } else
                 hipLaunchKernelGGL(
                     (sodkit::bn_apply_kernel<T, decltype(rl)::value, decltype(rs)::value>),
@@ -598,31 +649,38 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean, torch::Tensor invstd
         if (relu) { if (has_res) launch(tt{}, tt{}); else launch(tt{}, ff{}); }
         else      { if (has_res) launch(ff{}, tt{}); else launch(ff{}, ff{}); }
     });
-    return y;
+    if (!mask.defined())
+        mask = torch::empty({0}, x.options().dtype(torch::kByte));
+    return {y, mask};
 }
 
 torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
-                            torch::Tensor mean, torch::Tensor invstd, bool relu) {
+                            torch::Tensor mean, torch::Tensor invstd, bool relu,
+                            torch::Tensor mask) {
     int C = x.size(1);
     long M = x.numel() / C;
     bool vec = (C % 8 == 0) && x.element_size() == 2;
+    bool masked = relu && vec && mask.defined() && mask.numel() == x.numel() / 8;
     auto out = vec ? torch::empty({2, C}, x.options().dtype(torch::kFloat32))
                    : torch::zeros({2, C}, x.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
     auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
     DISPATCH_T(x, "bn_bwd_reduce", [&] {
-        auto launch = [&](auto rl) {
+        auto launch = [&](auto rl, auto mk) {
             if (C % 8 == 0 && sizeof(T) == 2) {
                 dim3 g = vec_stats_grid(M, C);
                 int nb = g.x;
                 auto partial = torch::empty({nb, 2L * C},
                                             x.options().dtype(torch::kFloat32));
                 hipLaunchKernelGGL(
-                    (sodkit::bn_bwd_reduce_vec_kernel<T, decltype(rl)::value>),
+                    (sodkit::bn_bwd_reduce_vec_kernel<T, decltype(rl)::value,
+                                                      decltype(mk)::value>),
                     g, dim3(256), 0, stream.stream(),
                     reinterpret_cast<const T *>(dy.data_ptr()),
                     reinterpret_cast<const T *>(y.data_ptr()),
                     reinterpret_cast<const T *>(x.data_ptr()),
+                    decltype(mk)::value ? mask.data_ptr<unsigned char>()
+                                        : nullptr,
                     meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
                     partial.data_ptr<float>(), M, C);
                 hipLaunchKernelGGL((sodkit::slab_reduce_kernel<16, 16>),
@@ -639,7 +697,9 @@ torch::Tensor bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
                     meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
                     out.data_ptr<float>(), M, C);
         };
-        if (relu) launch(std::true_type{}); else launch(std::false_type{});
+        using tt = std::true_type; using ff = std::false_type;
+        if (relu) { if (masked) launch(tt{}, tt{}); else launch(tt{}, ff{}); }
+        else launch(ff{}, ff{});
     });
     return out;
 }
@@ -648,27 +708,34 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor dy, torch::Tensor y,
                                         torch::Tensor x, torch::Tensor mean,
                                         torch::Tensor invstd, torch::Tensor gamma,
                                         torch::Tensor sums, double n_total,
-                                        bool relu, bool need_dres) {
+                                        bool relu, bool need_dres,
+                                        torch::Tensor mask) {
     int C = x.size(1);
     long total = x.numel();
     auto dx = torch::empty_like(x);
     torch::Tensor dres;
     if (need_dres) dres = torch::empty_like(x);
     float inv_n = n_total > 0 ? (float)(1.0 / n_total) : 0.f;
+    bool vec8 = (C % 8 == 0) && x.element_size() == 2;
+    bool masked = relu && vec8 && mask.defined() && mask.numel() == total / 8;
     auto stream = at::hip::getCurrentHIPStream();
     auto meanf = mean.contiguous(), invstdf = invstd.contiguous();
     auto gammaf = gamma.to(torch::kFloat32).contiguous();
     auto sumsf = sums.contiguous();
     DISPATCH_T(x, "bn_bwd_apply", [&] {
-        auto launch = [&](auto rl, auto dr) {
+        auto launch = [&](auto rl, auto dr, auto mk) {
             if (C % 8 == 0 && sizeof(T) == 2) {
                 long total8 = total / 8;
                 hipLaunchKernelGGL(
-                    (sodkit::bn_bwd_apply_vec_kernel<T, decltype(rl)::value, decltype(dr)::value>),
+                    (sodkit::bn_bwd_apply_vec_kernel<T, decltype(rl)::value,
+                                                     decltype(dr)::value,
+                                                     decltype(mk)::value>),
                     dim3(ew_blocks(total8)), dim3(256), 0, stream.stream(),
                     reinterpret_cast<const T *>(dy.data_ptr()),
                     reinterpret_cast<const T *>(y.data_ptr()),
                     reinterpret_cast<const T *>(x.data_ptr()),
+                    decltype(mk)::value ? mask.data_ptr<unsigned char>()
+                                        : nullptr,
                     meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
                     gammaf.data_ptr<float>(), sumsf.data_ptr<float>(), inv_n,
                     reinterpret_cast<T *>(dx.data_ptr()),
@@ -688,8 +755,13 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor dy, torch::Tensor y,
                     total, C);
         };
         using tt = std::true_type; using ff = std::false_type;
-        if (relu) { if (need_dres) launch(tt{}, tt{}); else launch(tt{}, ff{}); }
-        else      { if (need_dres) launch(ff{}, tt{}); else launch(ff{}, ff{}); }
+        if (relu && masked) {
+            if (need_dres) launch(tt{}, tt{}, tt{}); else launch(tt{}, ff{}, tt{});
+        } else if (relu) {
+            if (need_dres) launch(tt{}, tt{}, ff{}); else launch(tt{}, ff{}, ff{});
+        } else {
+            if (need_dres) launch(ff{}, tt{}, ff{}); else launch(ff{}, ff{}, ff{});
+        }
     });
     if (!need_dres) dres = torch::Tensor();
     return {dx, dres};

@@ -60,22 +60,27 @@ class _BNActFn(torch.autograd.Function):
             n_total = 0.0                                   # eval marker
             mean, invstd = ext().bn_finalize(running_mean, 0.0, momentum, eps,
                                              running_mean, running_var, False)
-        y = ext().bn_apply(x, mean, invstd, gamma, beta, relu,
-                           residual if residual is not None else torch.Tensor())
-        ctx.save_for_backward(x, y, mean, invstd, gamma)
+        # training: also emit the 1-bit ReLU-mask plane so the backward
+        # kernels read 1/16 of y's bytes for the mask instead of y itself
+        y, mask = ext().bn_apply(
+            x, mean, invstd, gamma, beta, relu,
+            residual if residual is not None else torch.Tensor(), training)
+        ctx.save_for_backward(x, y, mean, invstd, gamma,
+                              mask if mask.numel() else torch.Tensor())
         ctx.meta = (relu, residual is not None, n_total, process_group)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, mean, invstd, gamma = ctx.saved_tensors
+        x, y, mean, invstd, gamma, mask = ctx.saved_tensors
         relu, has_res, n_total, pg = ctx.meta
         dy = _cl(dy)
-        sums = ext().bn_bwd_reduce(dy, y, x, mean, invstd, relu)  # fp32 [2, C]
+        sums = ext().bn_bwd_reduce(dy, y, x, mean, invstd, relu,
+                                   mask)  # fp32 [2, C]
         if pg is not None and dist.is_initialized() and dist.get_world_size(pg) > 1:
             dist.all_reduce(sums, group=pg)
         dx, dres = ext().bn_bwd_apply(dy, y, x, mean, invstd, gamma, sums,
-                                      n_total, relu, has_res)
+                                      n_total, relu, has_res, mask)
         dgamma = sums[1].to(gamma.dtype)
         dbeta = sums[0].to(gamma.dtype)
         return (dx, dgamma, dbeta, None, None, None, None, None, None,

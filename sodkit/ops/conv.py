@@ -49,6 +49,9 @@ def _use_8p(m_tiles: int, cout: int) -> bool:
 # direct sliding-window wgrad for 3x3 s1 d1 p1 (csrc/wgrad_dir.hip): stages
 # each x row once and computes all nine taps from shifted LDS reads instead
 # of re-gathering per tap.  SODKIT_WGRAD_DIR=0 disables (A/B).
+# Measured (r2 ubench wdir column): 426 vs 166 TF at W=176, 251-253 vs
+# 160 at W=88, a wash at W=44, and a LOSS at W<=22 (k-tile padding +
+# short rows) — so the gate is W >= 64.
 _WDIR_ENV = os.environ.get("SODKIT_WGRAD_DIR", "")
 
 
@@ -58,7 +61,7 @@ def _use_wgrad_dir(w: torch.Tensor, x: torch.Tensor, stride, padding,
         return False
     return (w.shape[1] == 3 and w.shape[2] == 3 and stride == 1
             and padding == 1 and dilation == 1 and w.shape[3] % 64 == 0
-            and 8 <= x.shape[3] <= 184)
+            and 64 <= x.shape[3] <= 184)
 
 
 def _cl_ld(x: torch.Tensor):

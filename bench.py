@@ -45,6 +45,7 @@ def main() -> int:
         model=args.model, size=args.size, batch_size=args.batch,
         dtype=args.dtype, num_workers=0, output_root="/tmp/sodkit_bench",
         exp_name=f"bench_{args.model}", log_interval=10**9,
+        hip_graph=os.environ.get("SODKIT_HIPGRAPH", "0") == "1",
     )
     solver = Solver(cfg)
     solver.total_iters = 10**9  # near-constant LR over the bench window

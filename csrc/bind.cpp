@@ -63,7 +63,8 @@ std::vector<torch::Tensor> dynfilter_bwd(torch::Tensor dy, torch::Tensor x,
 
 void sgd_fused(torch::Tensor ptrs, torch::Tensor numels, torch::Tensor wds,
                torch::Tensor chunks, int64_t dtype_code, double lr, double mu,
-               double gscale, bool nesterov, torch::Tensor scale);
+               double gscale, bool nesterov, torch::Tensor scale,
+               torch::Tensor lr_t);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "sodkit CDNA4 (gfx950/MI355X) kernels";

@@ -20,7 +20,11 @@ __global__ void sgd_fused_kernel(const long *__restrict__ ptrs,
                                  const float *__restrict__ wds,
                                  const long *__restrict__ chunks,
                                  float lr, float mu, float gscale,
-                                 const float *__restrict__ scale) {
+                                 const float *__restrict__ scale,
+                                 const float *__restrict__ lrp) {
+    if (lrp != nullptr) lr = lrp[0];  // device-resident LR (hipGraph replay:
+    //                                   the host updates a tensor, not a
+    //                                   frozen kernel argument)
     if (scale != nullptr) {
         // device-side dynamic-loss-scale multiplier (fp16 path, sync-free):
         // 0.0 == overflow detected this step -> skip the whole update
@@ -54,13 +58,20 @@ __global__ void sgd_fused_kernel(const long *__restrict__ ptrs,
 
 void sgd_fused(torch::Tensor ptrs, torch::Tensor numels, torch::Tensor wds,
                torch::Tensor chunks, int64_t dtype_code, double lr, double mu,
-               double gscale, bool nesterov, torch::Tensor scale) {
+               double gscale, bool nesterov, torch::Tensor scale,
+               torch::Tensor lr_t) {
     int K = chunks.size(0);
     const float *scale_ptr = nullptr;
     if (scale.defined() && scale.numel() > 0) {
         TORCH_CHECK(scale.is_cuda() && scale.scalar_type() == torch::kFloat32,
                     "sgd scale multiplier must be a device fp32 scalar");
         scale_ptr = scale.data_ptr<float>();
+    }
+    const float *lr_ptr = nullptr;
+    if (lr_t.defined() && lr_t.numel() > 0) {
+        TORCH_CHECK(lr_t.is_cuda() && lr_t.scalar_type() == torch::kFloat32,
+                    "device lr must be a device fp32 scalar");
+        lr_ptr = lr_t.data_ptr<float>();
     }
     auto stream = at::hip::getCurrentHIPStream();
     auto launch = [&](auto t_tag, auto nv) {
@@ -69,7 +80,8 @@ void sgd_fused(torch::Tensor ptrs, torch::Tensor numels, torch::Tensor wds,
                            dim3(K), dim3(256), 0, stream.stream(),
                            ptrs.data_ptr<long>(), numels.data_ptr<long>(),
                            wds.data_ptr<float>(), chunks.data_ptr<long>(),
-                           (float)lr, (float)mu, (float)gscale, scale_ptr);
+                           (float)lr, (float)mu, (float)gscale, scale_ptr,
+                           lr_ptr);
     };
     struct F32 { using type = float; };
     struct BF { using type = __hip_bfloat16; };

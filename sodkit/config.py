@@ -50,6 +50,11 @@ class Config:
     warmup_iters: int = 0
     grad_clip: float = 0.0               # 0 = off
 
+    # hipGraph whole-step capture (single-GPU, accum_steps==1): replaces
+    # ~1900 per-step kernel dispatches with one graph launch.  LR feeds the
+    # fused SGD through a device tensor so the poly schedule survives replay.
+    hip_graph: bool = False
+
     # precision
     dtype: str = "bf16"                  # bf16 | fp16 | fp32 (compute dtype on GPU)
     loss_scale: float = 1.0              # static loss scale (fp16 only; bf16 needs none)

@@ -113,6 +113,16 @@ class Solver:
         self._accum_count = 0
         self.tb = None
 
+        # hipGraph whole-step capture (cfg.hip_graph): single-GPU,
+        # accum_steps == 1 only — the RCCL collectives stay out of capture
+        # until graph+RCCL is validated on a multi-GPU node.
+        self._graph_ok = (cfg.hip_graph and self.is_gpu
+                          and not self.distributed
+                          and max(1, cfg.accum_steps) == 1)
+        self._graph = None
+        self._graph_warm = 0
+        self._g_static = None
+
     # ------------------------------------------------------------------
     def _lr_at(self, it: int) -> float:
         cfg = self.cfg
@@ -198,6 +208,53 @@ class Solver:
             return cfg.grad_clip / (norm + 1e-6)
         return 1.0
 
+    # ------------------------------------------------------------------
+    def _step_body(self, img, mask, depth, lr) -> torch.Tensor:
+        """One full optimization step with NO host round-trips — the
+        hipGraph capture region (also runs eagerly during graph warmup).
+        ``lr`` may be a device scalar (graph replay updates it in place)."""
+        self.reducer.zero_grad()
+        logits = self.model(img, depth) if depth is not None else self.model(img)
+        loss = self.criterion(logits, mask)
+        if self._use_device_scaler and self.dynamic_scale:
+            (loss * self._scale_t).backward()
+        else:
+            loss.backward()
+        self.reducer.finalize()
+        if self._use_device_scaler:
+            mult = self._device_clip_and_scale()
+            self.optimizer.step(lr, self.reducer.grad_scale, scale_t=mult)
+        else:
+            self.optimizer.step(lr, self.reducer.grad_scale)
+        return loss
+
+    def _step_graph(self, img, mask, depth) -> torch.Tensor:
+        if self._graph_warm < 2:   # eager warmups stabilize the allocator
+            self._graph_warm += 1
+            loss = self._step_body(img, mask, depth,
+                                   self._lr_at(self.iter_count))
+            self.iter_count += 1
+            return loss.detach()
+        if self._graph is None:
+            self._g_lr = torch.zeros((), device=self.device)
+            self._g_img = img.clone()
+            self._g_mask = mask.clone()
+            self._g_depth = depth.clone() if depth is not None else None
+            torch.cuda.synchronize()
+            self._graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._graph):
+                self._g_loss = self._step_body(self._g_img, self._g_mask,
+                                               self._g_depth, self._g_lr)
+            torch.cuda.synchronize()
+        self._g_img.copy_(img, non_blocking=True)
+        self._g_mask.copy_(mask, non_blocking=True)
+        if depth is not None:
+            self._g_depth.copy_(depth, non_blocking=True)
+        self._g_lr.fill_(self._lr_at(self.iter_count))
+        self._graph.replay()
+        self.iter_count += 1
+        return self._g_loss.detach()
+
     def step(self, img: torch.Tensor, mask: torch.Tensor,
              depth: Optional[torch.Tensor] = None) -> torch.Tensor:
         """One loader batch.  With ``accum_steps == 1`` (default) this is a
@@ -206,6 +263,8 @@ class Solver:
         collectives); the N-th runs the overlapped all-reduce of the
         accumulated sums + the fused SGD with 1/(world*accum) folded in.
         Returns the (detached, device-side) unscaled micro-batch loss."""
+        if self._graph_ok:
+            return self._step_graph(img, mask, depth)
         accum = max(1, self.cfg.accum_steps)
         first = self._accum_count == 0
         final = self._accum_count == accum - 1

@@ -82,12 +82,13 @@ class SodSGD:
             self._tables.append((codes[dtype], ptr_t, numel_t, wd_t, chunk_t))
 
     @torch.no_grad()
-    def step(self, lr: float, grad_scale: float = 1.0,
+    def step(self, lr, grad_scale: float = 1.0,
              scale_t: "torch.Tensor | None" = None):
         """``scale_t`` (GPU only): a device fp32 scalar multiplied into
         grad_scale INSIDE the kernel; 0.0 skips the update entirely.  This is
         the sync-free fp16 dynamic-scaler contract — the overflow decision
-        never touches the host."""
+        never touches the host.  ``lr`` may be a device fp32 scalar tensor
+        (hipGraph capture: the host updates the tensor between replays)."""
         first = next((p for g in self.groups for p in g["params"]), None)
         if first is None:
             return
@@ -95,11 +96,16 @@ class SodSGD:
             if self._tables is None:
                 self._build_tables()
             st = scale_t if scale_t is not None else torch.Tensor()
+            if isinstance(lr, torch.Tensor):
+                lr_t, lr_f = lr, 0.0
+            else:
+                lr_t, lr_f = torch.Tensor(), float(lr)
             for dt, ptr_t, numel_t, wd_t, chunk_t in self._tables:
-                ext().sgd_fused(ptr_t, numel_t, wd_t, chunk_t, dt, float(lr),
+                ext().sgd_fused(ptr_t, numel_t, wd_t, chunk_t, dt, lr_f,
                                 self.momentum, float(grad_scale),
-                                self.nesterov, st)
+                                self.nesterov, st, lr_t)
         else:
+            lr = float(lr)
             for g in self.groups:
                 for p in g["params"]:
                     if p.grad is None:

@@ -260,9 +260,9 @@ def test_wgrad_dir_dispatch_is_used():
     from sodkit.ops import conv as conv_mod
     from sodkit import ops
     torch.manual_seed(3)
-    x = _cl(torch.randn(2, 64, 40, 40, device="cuda").to(torch.bfloat16))
+    x = _cl(torch.randn(2, 64, 64, 64, device="cuda").to(torch.bfloat16))
     w = (torch.randn(64, 3, 3, 64, device="cuda") * 0.1).to(torch.bfloat16)
-    assert conv_mod._use_wgrad_dir(w, x, 1, 1, 1)
+    assert conv_mod._use_wgrad_dir(w, x, 1, 1, 1)  # W=64 = the gate floor
     xg = x.clone().requires_grad_()
     wg = w.clone().requires_grad_()
     y = ops.conv2d(xg, wg, None, 1, 1, 1)

@@ -375,8 +375,14 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
 //   ds writes); the x tile gathers 8 consecutive output-pixels per lane
 //   (m-contiguous in LDS -> single b128 write).
 // ---------------------------------------------------------------------------
-template <typename T, int BCOUT, int BJ, int BKM>
-__global__ __launch_bounds__(256)
+// NT = threads/block.  256 => 4 waves as 2(cout-half) x 2(j-half),
+// per-wave acc 64x64.  512 => 8 waves as 4(cout-quarter) x 2(j-half) over a
+// 256x256 tile (per-wave 64x128, acc 128 VGPR): doubles the MFMA work per
+// staged LDS byte (64->131 flop/B), which is what bounds this kernel —
+// measured r2: the big 1x1/deep-3x3 wgrads sat at 111-256 TF on the
+// 128-square tile.
+template <typename T, int BCOUT, int BJ, int BKM, int NT = 256>
+__global__ __launch_bounds__(NT)
 void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                   float *__restrict__ partial,
                   int N, int H, int W, int C,     // x dims
@@ -386,7 +392,8 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                   int Kw,                          // = R*S*C (dw inner size)
                   int M_total,                     // = N*P*Q
                   int m_per_split, int tiles_c) {
-    constexpr int FC = BCOUT / 32;
+    constexpr int CW = (NT / 64) / 2;             // cout wave groups
+    constexpr int FC = BCOUT / (16 * CW);
     constexpr int FJ = BJ / 32;
 
     __shared__ __align__(16) char lds[(BCOUT + BJ) * BKM * 2];
@@ -412,12 +419,16 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     // (r,s) tap when C%8==0 -> four 16B gathers (m..m+3), transposed in
     // registers into 8 ds_write_b64 (the scattered-u16 transpose was
     // write-instruction-bound).
-    constexpr int XCH = BJ * (BKM / 4) / 8 / 256;
+    constexpr int XCH_TOT = BJ * (BKM / 4) / 8;   // total x chunks
+    constexpr int XCH = (XCH_TOT + NT - 1) / NT;  // per-thread (may idle)
     int xr[XCH], xs[XCH], xc[XCH], xj[XCH], xml[XCH];
     int xgn[XCH], xgp[XCH], xgq[XCH];   // incremental (n,p,q) of the quad base
+    bool xact[XCH];                     // chunk exists (XCH_TOT may be < NT)
 #pragma unroll
     for (int i = 0; i < XCH; ++i) {
-        int ch = t + i * 256;
+        int ch = t + i * NT;
+        xact[i] = ch < XCH_TOT;
+        if (!xact[i]) ch = 0;
         int jv8 = ch % (BJ / 8);
         xml[i] = (ch / (BJ / 8)) * 4;          // first m of the quad
         int j = tile_j + jv8 * 8;
@@ -444,7 +455,7 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
         // ---- stage dy^T tile: 4 coalesced 16B reads dy[m..m+3][cout..+7],
         //      register transpose, 8 ds_write_b64 ----
-        for (int ch = t; ch < DNCH; ch += 256) {
+        for (int ch = t; ch < DNCH; ch += NT) {
             int m_local = (ch / (BCOUT / 8)) * 4;
             int cv = (ch % (BCOUT / 8)) * 8;
             int cout = tile_c + cv;
@@ -476,6 +487,7 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
         //      register transpose, 8 ds_write_b64 ----
 #pragma unroll
         for (int i = 0; i < XCH; ++i) {
+            if (!xact[i]) continue;
             int j_local0 = (xj[i] - tile_j);
             s16x8 v[4];
 #pragma unroll
@@ -528,7 +540,7 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
             s16x8 a[FC], b[FJ];
 #pragma unroll
             for (int fc = 0; fc < FC; ++fc) {
-                int row = wv_c * (BCOUT / 2) + fc * 16 + (lane & 15);
+                int row = wv_c * (BCOUT / CW) + fc * 16 + (lane & 15);
                 a[fc] = *reinterpret_cast<const s16x8 *>(dyT_lds + swz<BKM>(row, kbyte));
             }
 #pragma unroll
@@ -549,7 +561,7 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     float *dst_base = partial + (long)z * Cout * Kw;
 #pragma unroll
     for (int fc = 0; fc < FC; ++fc) {
-        int cout = tile_c + wv_c * (BCOUT / 2) + fc * 16 + (lane >> 4) * 4;
+        int cout = tile_c + wv_c * (BCOUT / CW) + fc * 16 + (lane >> 4) * 4;
 #pragma unroll
         for (int fj = 0; fj < FJ; ++fj) {
             int j = tile_j + wv_j * (BJ / 2) + fj * 16 + (lane & 15);
@@ -742,13 +754,16 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
     int Kw = R * S * C;
     auto stream = at::hip::getCurrentHIPStream();
 
-    auto run = [&](auto t_tag, auto bc_tag) -> torch::Tensor {
+    auto run = [&](auto t_tag, auto bc_tag, auto bj_tag, auto nt_tag)
+        -> torch::Tensor {
         using T = typename decltype(t_tag)::type;
         constexpr int BC = decltype(bc_tag)::value;
-        constexpr int BJ = 128;
+        constexpr int BJ = decltype(bj_tag)::value;
+        constexpr int NT = decltype(nt_tag)::value;
         // the hot BCOUT=64 (decoder) variant runs BKM=128: half the
         // barriers per MFMA at 48KB LDS (3 blocks/CU); BCOUT=128 keeps
-        // BKM=64 (64KB would drop it to 2 blocks/CU)
+        // BKM=64 (64KB would drop it to 2 blocks/CU); the 8-wave 256-tile
+        // variants stay at BKM=64 (64KB LDS, 2 blocks/CU)
         constexpr int BKM = (BC == 64) ? 128 : 64;
         int tiles_c = (Cout + BC - 1) / BC;
         int tiles_j = (Kw + BJ - 1) / BJ;
@@ -761,8 +776,8 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
         auto partial = torch::empty({splitk, (long)Cout * Kw},
                                     x.options().dtype(torch::kFloat32));
         dim3 grid(tiles, splitk);
-        hipLaunchKernelGGL((sodkit::wgrad_kernel<T, BC, BJ, BKM>), grid, dim3(256),
-                           0, stream.stream(),
+        hipLaunchKernelGGL((sodkit::wgrad_kernel<T, BC, BJ, BKM, NT>), grid,
+                           dim3(NT), 0, stream.stream(),
                            reinterpret_cast<const T *>(x.data_ptr()),
                            reinterpret_cast<const T *>(dy.data_ptr()),
                            partial.data_ptr<float>(),
@@ -784,14 +799,26 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
     struct FP { using type = __half; };
     using c64 = std::integral_constant<int, 64>;
     using c128 = std::integral_constant<int, 128>;
+    using c256 = std::integral_constant<int, 256>;
+    using c512 = std::integral_constant<int, 512>;
     torch::Tensor dw;
+    // tile policy: intensity (MFMA flops per staged LDS byte) is what
+    // bounds this kernel, so pick the largest tile the output supports
     bool small_c = Cout <= 64;
+    bool big = Cout >= 256 && Kw >= 256;     // 256x256, 8 waves
+    bool tall = Cout >= 256 && Kw >= 128 && !big;  // 256x128, 8 waves
     AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_wgrad",
         AT_DISPATCH_CASE(at::kBFloat16, [&] {
-            dw = small_c ? run(BF{}, c64{}) : run(BF{}, c128{});
+            if (big) dw = run(BF{}, c256{}, c256{}, c512{});
+            else if (tall) dw = run(BF{}, c256{}, c128{}, c512{});
+            else if (small_c) dw = run(BF{}, c64{}, c128{}, c256{});
+            else dw = run(BF{}, c128{}, c128{}, c256{});
         })
         AT_DISPATCH_CASE(at::kHalf, [&] {
-            dw = small_c ? run(FP{}, c64{}) : run(FP{}, c128{});
+            if (big) dw = run(FP{}, c256{}, c256{}, c512{});
+            else if (tall) dw = run(FP{}, c256{}, c128{}, c512{});
+            else if (small_c) dw = run(FP{}, c64{}, c128{}, c256{});
+            else dw = run(FP{}, c128{}, c128{}, c256{});
         }));
     return dw;
 }

@@ -38,6 +38,8 @@ CONV_SHAPES = [
     (2, 32, 40, 1, 1, 1, 0, 1),        # head 1-channel (Ncout tail)
     (1, 64, 22, 9, 3, 1, 1, 1),        # DDPM kernel-gen (Ncout=9 tails)
     (2, 48, 11, 512, 3, 1, 1, 1),      # deep small-spatial large-C
+    (2, 128, 22, 512, 1, 1, 0, 1),     # wgrad 256x128 8-wave tile (tall)
+    (2, 256, 22, 1024, 1, 1, 0, 1),    # wgrad 256x256 8-wave tile (big)
 ]
 
 

@@ -811,6 +811,9 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
         // fill the chip: ~2 blocks per CU (256 CUs) regardless of tile count
         int splitk = std::max(1, std::min(512, 512 / std::max(tiles, 1)));
         splitk = (int)std::min<long>(splitk, (M + BKM - 1) / BKM);
+        if (NT == 512)   // 8-wave tiles: keep >=8 K-steps per split so the
+            //              prologue + 256KB fp32 partial store amortize
+            splitk = std::max(1, std::min<int>(splitk, (int)(M / BKM / 8)));
         long m_per_split = ((M + splitk - 1) / splitk + BKM - 1) / BKM * BKM;
         splitk = (int)((M + m_per_split - 1) / m_per_split);
         auto partial = torch::empty({splitk, (long)Cout * Kw},
@@ -842,11 +845,17 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
     using c256 = std::integral_constant<int, 256>;
     using c512 = std::integral_constant<int, 512>;
     torch::Tensor dw;
-    // tile policy: intensity (MFMA flops per staged LDS byte) is what
-    // bounds this kernel, so pick the largest tile the output supports
+    // tile policy: intensity (MFMA flops per staged LDS byte) bounds the
+    // kernel, BUT a bigger tile means fewer tiles, a larger split-K to
+    // fill the chip, and splitk x Cout x Kw x 4B of fp32 partial traffic —
+    // measured r2: at this model's small-M big-Cout shapes (M 3.8-62k)
+    // the 256-tile variants LOST 30-40% to partial-store blowup (l4_1x1e
+    // 118 vs 180 TF).  The 8-wave tiles need M large enough that split-K
+    // stays shallow; gate on M >= 200k pixels (512-res class shapes).
     bool small_c = Cout <= 64;
-    bool big = Cout >= 256 && Kw >= 256;     // 256x256, 8 waves
-    bool tall = Cout >= 256 && Kw >= 128 && !big;  // 256x128, 8 waves
+    bool mbig = M >= 200000;
+    bool big = Cout >= 256 && Kw >= 256 && mbig;     // 256x256, 8 waves
+    bool tall = Cout >= 256 && Kw >= 128 && !big && mbig;  // 256x128
     AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_wgrad",
         AT_DISPATCH_CASE(at::kBFloat16, [&] {
             if (big) dw = run(BF{}, c256{}, c256{}, c512{});

@@ -18,6 +18,13 @@ torch::Tensor conv2d_fwd_8p(torch::Tensor x, torch::Tensor w,
 torch::Tensor conv2d_dgrad_8p(torch::Tensor dy, torch::Tensor w_t, int64_t H,
                               int64_t W, int64_t stride, int64_t pad,
                               int64_t dil, int64_t lddy);
+// direct 3x3 s1 d1 p1 64-channel fwd/dgrad (csrc/conv_dir.hip)
+bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
+                       int64_t dil, int64_t Cin, int64_t Cout, int64_t W);
+torch::Tensor conv2d_fwd_dir(torch::Tensor x, torch::Tensor w,
+                             torch::Tensor bias, int64_t ld);
+torch::Tensor conv2d_dgrad_dir(torch::Tensor dy, torch::Tensor w_t,
+                               int64_t lddy);
 // direct 3x3 s1 d1 p1 wgrad (csrc/wgrad_dir.hip)
 bool wgrad_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
                         int64_t dil, int64_t C, int64_t W);
@@ -75,6 +82,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv8_eligible", &conv8_eligible);
     m.def("conv2d_fwd_8p", &conv2d_fwd_8p);
     m.def("conv2d_dgrad_8p", &conv2d_dgrad_8p);
+    m.def("conv_dir_eligible", &conv_dir_eligible);
+    m.def("conv2d_fwd_dir", &conv2d_fwd_dir);
+    m.def("conv2d_dgrad_dir", &conv2d_dgrad_dir);
     m.def("wgrad_dir_eligible", &wgrad_dir_eligible);
     m.def("conv2d_wgrad_dir", &conv2d_wgrad_dir);
     m.def("bn_stats", &bn_stats);

@@ -1,0 +1,284 @@
+// sodkit — direct (non-im2col) fwd/dgrad kernel for 3x3 stride-1
+// dilation-1 pad-1 convolutions with C == 64, Cout <= 64 on CDNA4: the
+// MINet/HDFNet decoder + AIM/SIM class, which is ~18% of the training
+// step (kernel_summary r2: the 176/88-square 64-channel convs).
+//
+// The im2col igemm kernel stages every x element ~9x (once per tap its
+// im2col row touches): B-operand traffic = M*K*2 bytes (1.1 GB per
+// 176-square conv), which L2 bandwidth bounds at ~400 TF.  Here a block
+// stages the three input rows of an output row ONCE in their natural
+// [q][c] layout and the nine taps read ROW-shifted windows — for the
+// forward direction the MFMA K dimension is j=(r,s,c), which is
+// c-contiguous in both operands, so every fragment read is a plain
+// aligned b128 (no gather, no transpose, no alignment tricks):
+//
+//   A = w[cout][(r,s,c)]      KRSC rows ARE the GEMM-A rows (LDS-resident,
+//                             staged once per block, ~72 KB)
+//   B[q][(r,s,c)] = x[p+r-1][q+s-1][c]   from 3 row images [q][c]
+//
+// dgrad is the same kernel with A = the permuted [Cin][r][s][Cout] copy
+// read at the flipped tap (8-tap) and B = dy rows (same shift form).
+//
+// Wave layout: 8 waves = 2 cout-halves x 4 q-groups; q-group frags are
+// strided (g, g+4, g+8) so ragged W wastes at most the tail fragment.
+// Row pitches padded so (pitch/4) % 64 == 4|36 (16-row b128 fragment
+// reads hit 16 distinct bank quads).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace sodkit {
+namespace cdir {
+
+typedef short s16x4_ __attribute__((ext_vector_type(4)));
+
+constexpr int XPITCH = 144;    // 64ch * 2B = 128 data + 16 pad (bank stride 36)
+constexpr int WPITCH = 1168;   // 576 j * 2B = 1152 data + 16 pad (stride 292)
+
+// MODE 0: fwd (src = x, wts = w KRSC). MODE 1: dgrad (src = dy,
+// wts = w_t [Cin][R][S][Cout], taps flipped on the A side).
+template <typename T, int MODE>
+__global__ __launch_bounds__(512, 1)
+void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
+                     const float *__restrict__ bias, T *__restrict__ out,
+                     int N, int H, int W, int ldsrc, int ldout, int Cout,
+                     int rows_per_blk, int total_rows, int nqf) {
+    // C (inner channels of src) is 64 by eligibility; K = 576.
+    extern __shared__ char lds[];
+    char *wbuf = lds;                               // [Cout<=64][WPITCH]
+    // +16 extra zero rows: the ragged last q-fragment reads rows past
+    // W+1; they must stay in-bounds and zero
+    const int xrows = W + 18;
+    char *xbuf[3];
+    xbuf[0] = lds + 64 * WPITCH;
+    xbuf[1] = xbuf[0] + xrows * XPITCH;
+    xbuf[2] = xbuf[1] + xrows * XPITCH;
+    const int lds_bytes = 64 * WPITCH + 3 * xrows * XPITCH;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wave = t >> 6;
+    const int wm = wave >> 2;          // cout half
+    const int wq = wave & 3;           // q group
+
+    const int r0 = blockIdx.x * rows_per_blk;
+    const int r1 = min(r0 + rows_per_blk, total_rows);
+
+    // ---- one-time zero fill (halo rows / pads stay zero) ----
+    for (int i = t * 8; i < lds_bytes; i += 512 * 8)
+        *reinterpret_cast<long *>(lds + i) = 0;
+    __syncthreads();
+
+    // ---- one-time weight staging: [row][1152B] -> padded pitch ----
+    {
+        int row = t >> 3, b16 = t & 7;
+        if (row < Cout) {
+            const char *g = reinterpret_cast<const char *>(wts) + (size_t)row * 1152;
+            char *d = wbuf + row * WPITCH;
+#pragma unroll
+            for (int b = 0; b < 9; ++b) {
+                int off = (b16 + b * 8) * 16;
+                *reinterpret_cast<s16x8 *>(d + off) =
+                    *reinterpret_cast<const s16x8 *>(g + off);
+            }
+        }
+    }
+
+    const int passes = (W + 63) / 64;   // x staging: 64 pixels per pass
+
+    for (int g = r0; g < r1; ++g) {
+        const int n = g / H;
+        const int p = g % H;
+
+        __syncthreads();   // previous row's MFMA reads done (also covers
+        //                    the one-time staging on the first iteration)
+
+        // ---- stage the 3 input rows, natural [q][c] layout ----
+        {
+            int c8 = t & 7, pix0 = t >> 3;   // 64 pixels x 8 c-groups
+#pragma unroll
+            for (int d = 0; d < 3; ++d) {
+                int ih = p + d - 1;
+                bool rv = ih >= 0 && ih < H;
+                for (int ps = 0; ps < passes; ++ps) {
+                    int q = pix0 + ps * 64;
+                    if (q >= W) break;
+                    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+                    if (rv)
+                        v = *reinterpret_cast<const s16x8 *>(
+                            src + ((size_t)(n * H + ih) * W + q) * ldsrc
+                                + c8 * 8);
+                    *reinterpret_cast<s16x8 *>(
+                        xbuf[d] + (q + 1) * XPITCH + c8 * 16) = v;
+                }
+            }
+        }
+        __syncthreads();
+
+        // ---- MFMA: D[cout][q] over K = 576, 18 k-tiles of 32 ----
+        f32x4 acc[2][3];
+#pragma unroll
+        for (int a = 0; a < 2; ++a)
+#pragma unroll
+            for (int b = 0; b < 3; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+        const int oct16 = ((lane >> 4) << 4);       // k-octet byte
+        const int arow = (wm * 32 + (lane & 15)) * WPITCH;
+#pragma unroll
+        for (int kt = 0; kt < 18; ++kt) {
+            const int tap = kt >> 1, chalf = kt & 1;
+            const int atap = MODE ? (8 - tap) : tap;
+            const int abyte = atap * 128 + chalf * 64 + oct16;
+            s16x8 a0 = *reinterpret_cast<const s16x8 *>(wbuf + arow + abyte);
+            s16x8 a1 = *reinterpret_cast<const s16x8 *>(wbuf + arow + 16 * WPITCH
+                                                        + abyte);
+            const int rr = tap / 3, ss = tap % 3;
+            const char *xb = xbuf[rr];
+            const int bbyte = chalf * 64 + oct16;
+#pragma unroll
+            for (int i = 0; i < 3; ++i) {
+                if (i >= nqf) break;
+                int q0 = (wq + i * 4) * 16;
+                // row q + s - 1, +1 halo base; rows beyond W+1 are zero pad
+                int row = q0 + (lane & 15) + ss;
+                s16x8 b = *reinterpret_cast<const s16x8 *>(
+                    xb + row * XPITCH + bbyte);
+                acc[0][i] = MfmaOp<T>::mma(a0, b, acc[0][i]);
+                acc[1][i] = MfmaOp<T>::mma(a1, b, acc[1][i]);
+            }
+        }
+
+        // ---- store y row (NHWC, 4 couts per lane) ----
+        const size_t obase = (size_t)(n * H + p) * W;
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+            int cout = wm * 32 + mf * 16 + ((lane >> 4) << 2);
+            if (cout >= Cout) continue;
+            float b4[4] = {0.f, 0.f, 0.f, 0.f};
+            if (bias != nullptr) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e)
+                    if (cout + e < Cout) b4[e] = bias[cout + e];
+            }
+#pragma unroll
+            for (int i = 0; i < 3; ++i) {
+                if (i >= nqf) break;
+                int q = (wq + i * 4) * 16 + (lane & 15);
+                if (q >= W) continue;
+                s16x4_ v;
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    T tv = ElemTraits<T>::from_f32(acc[mf][i][e] + b4[e]);
+                    v[e] = *reinterpret_cast<const short *>(&tv);
+                }
+                *reinterpret_cast<s16x4_ *>(out + (obase + q) * ldout + cout) = v;
+            }
+        }
+    }
+}
+
+}  // namespace cdir
+}  // namespace sodkit
+
+// ===========================================================================
+// Host glue
+// ===========================================================================
+
+bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
+                       int64_t dil, int64_t Cin, int64_t Cout, int64_t W) {
+    if (R != 3 || S != 3 || stride != 1 || pad != 1 || dil != 1) return false;
+    if (Cin != 64) return false;               // j-chunking assumes C == 64
+    if (Cout > 64 || Cout % 4 != 0) return false;
+    if (W > 184 || W < 16) return false;
+    return 64 * sodkit::cdir::WPITCH
+               + 3 * (W + 18) * sodkit::cdir::XPITCH <= 160 * 1024;
+}
+
+namespace {
+
+template <typename T, int MODE>
+void launch_conv_dir(const torch::Tensor &src, const torch::Tensor &wts,
+                     const float *bias_ptr, torch::Tensor &out,
+                     int N, int H, int W, int ldsrc, int ldout, int Cout) {
+    auto stream = at::hip::getCurrentHIPStream();
+    const int total_rows = N * H;
+    // ~2 row-sweeping blocks per CU
+    int blocks = std::min(512, total_rows);
+    int rows_per_blk = (total_rows + blocks - 1) / blocks;
+    blocks = (total_rows + rows_per_blk - 1) / rows_per_blk;
+    const int nqf_total = (W + 15) / 16;
+    const int nqf = (nqf_total + 3) / 4;     // frags per q-group (<= 3)
+    const size_t shmem = 64 * sodkit::cdir::WPITCH
+                         + 3 * (size_t)(W + 18) * sodkit::cdir::XPITCH;
+    static bool attr = false;
+    if (!attr) {
+        (void)hipFuncSetAttribute(
+            (const void *)sodkit::cdir::conv_dir_kernel<T, MODE>,
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr = true;
+    }
+    hipLaunchKernelGGL((sodkit::cdir::conv_dir_kernel<T, MODE>), dim3(blocks),
+                       dim3(512), shmem, stream.stream(),
+                       reinterpret_cast<const T *>(src.data_ptr()),
+                       reinterpret_cast<const T *>(wts.data_ptr()), bias_ptr,
+                       reinterpret_cast<T *>(out.data_ptr()),
+                       N, H, W, ldsrc, ldout, Cout, rows_per_blk, total_rows,
+                       nqf);
+}
+
+}  // namespace
+
+torch::Tensor conv2d_fwd_dir(torch::Tensor x, torch::Tensor w,
+                             torch::Tensor bias, int64_t ld) {
+    TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.dim() == 4 && w.dim() == 4);
+    int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    int Cout = w.size(0);
+    TORCH_CHECK(w.size(1) == 3 && w.size(2) == 3 && w.size(3) == C
+                && w.is_contiguous());
+    TORCH_CHECK(conv_dir_eligible(3, 3, 1, 1, 1, C, Cout, W_), "not eligible");
+    TORCH_CHECK(ld >= C, "bad channel pitch");
+    auto out = torch::empty({N, Cout, H, W_},
+                            x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    torch::Tensor bias32;
+    const float *bp = nullptr;
+    if (bias.defined() && bias.numel()) {
+        bias32 = bias.to(torch::kFloat32).contiguous();
+        bp = bias32.data_ptr<float>();
+    }
+    AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_fwd_dir",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            launch_conv_dir<__hip_bfloat16, 0>(x, w, bp, out, N, H, W_,
+                                               (int)ld, Cout, Cout);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            launch_conv_dir<__half, 0>(x, w, bp, out, N, H, W_, (int)ld,
+                                       Cout, Cout);
+        }));
+    return out;
+}
+
+torch::Tensor conv2d_dgrad_dir(torch::Tensor dy, torch::Tensor w_t,
+                               int64_t lddy) {
+    TORCH_CHECK(dy.is_cuda() && w_t.is_cuda());
+    int N = dy.size(0), Cout = dy.size(1), H = dy.size(2), W_ = dy.size(3);
+    int Cin = w_t.size(0);
+    TORCH_CHECK(w_t.size(1) == 3 && w_t.size(2) == 3 && w_t.size(3) == Cout
+                && w_t.is_contiguous());
+    // roles swap: the gemm "C" is Cout(=64), output channels are Cin
+    TORCH_CHECK(conv_dir_eligible(3, 3, 1, 1, 1, Cout, Cin, W_), "not eligible");
+    TORCH_CHECK(lddy >= Cout, "bad channel pitch");
+    auto dx = torch::empty({N, Cin, H, W_},
+                           dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+    AT_DISPATCH_SWITCH(dy.scalar_type(), "conv2d_dgrad_dir",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            launch_conv_dir<__hip_bfloat16, 1>(dy, w_t, nullptr, dx, N, H, W_,
+                                               (int)lddy, Cin, Cin);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            launch_conv_dir<__half, 1>(dy, w_t, nullptr, dx, N, H, W_,
+                                       (int)lddy, Cin, Cin);
+        }));
+    return dx;
+}

@@ -38,6 +38,7 @@ SRC = [
     "csrc/conv_igemm.hip",
     "csrc/conv_igemm8.hip",
     "csrc/wgrad_dir.hip",
+    "csrc/conv_dir.hip",
     "csrc/bnrelu.hip",
     "csrc/bilinear.hip",
     "csrc/loss.hip",

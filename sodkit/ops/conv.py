@@ -64,6 +64,21 @@ def _use_wgrad_dir(w: torch.Tensor, x: torch.Tensor, stride, padding,
             and 64 <= x.shape[3] <= 184)
 
 
+# direct fwd/dgrad for the 64-channel 3x3 decoder/AIM/SIM class
+# (csrc/conv_dir.hip): the im2col kernel re-stages every x element 9x.
+# SODKIT_CONV_DIR=0 disables (A/B).
+_CDIR_ENV = os.environ.get("SODKIT_CONV_DIR", "")
+
+
+def _use_conv_dir(cin: int, cout: int, w3: bool, x: torch.Tensor, stride,
+                  padding, dilation) -> bool:
+    if _CDIR_ENV == "0":
+        return False
+    return (w3 and stride == 1 and padding == 1 and dilation == 1
+            and cin == 64 and cout <= 64 and cout % 4 == 0
+            and 16 <= x.shape[3] <= 184)
+
+
 def _cl_ld(x: torch.Tensor):
     """Return (tensor, channel pitch).  A channel-slice of a channels_last
     tensor (torch.split / cat-backward narrow views) keeps NHWC pixel layout
@@ -99,6 +114,10 @@ class _ConvFn(torch.autograd.Function):
             use8 = False
         if use8:
             y = ext().conv2d_fwd_8p(x, w, b, stride, padding, dilation, ldx)
+        elif _use_conv_dir(w.shape[3], w.shape[0],
+                           w.shape[1] == 3 and w.shape[2] == 3,
+                           x, stride, padding, dilation):
+            y = ext().conv2d_fwd_dir(x, w, b, ldx)
         else:
             y = ext().conv2d_fwd(x, w, b, stride, padding, dilation, ldx)
         ctx.save_for_backward(x, w)
@@ -119,6 +138,11 @@ class _ConvFn(torch.autograd.Function):
                     w_t.shape[0]):
                 dx = ext().conv2d_dgrad_8p(dy, w_t, x.shape[2], x.shape[3],
                                            stride, padding, dilation, lddy)
+            elif _use_conv_dir(w_t.shape[3], w_t.shape[0],
+                               w_t.shape[1] == 3 and w_t.shape[2] == 3,
+                               x, stride, padding, dilation):
+                # dgrad roles: GEMM channels = Cout(=64), output = Cin
+                dx = ext().conv2d_dgrad_dir(dy, w_t, lddy)
             else:
                 dx = ext().conv2d_dgrad(dy, w_t, x.shape[2], x.shape[3],
                                         stride, padding, dilation, lddy)

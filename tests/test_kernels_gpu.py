@@ -40,6 +40,8 @@ CONV_SHAPES = [
     (2, 48, 11, 512, 3, 1, 1, 1),      # deep small-spatial large-C
     (2, 128, 22, 512, 1, 1, 0, 1),     # wgrad 256x128 8-wave tile (tall)
     (2, 256, 22, 1024, 1, 1, 0, 1),    # wgrad 256x256 8-wave tile (big)
+    (1, 64, 176, 64, 3, 1, 1, 1),      # conv_dir max-W decoder class
+    (2, 64, 33, 32, 3, 1, 1, 1),       # conv_dir odd W + Cout 32 (masked)
 ]
 
 

@@ -39,7 +39,10 @@ constexpr int WPITCH = 1168;   // 576 j * 2B = 1152 data + 16 pad (stride 292)
 
 // MODE 0: fwd (src = x, wts = w KRSC). MODE 1: dgrad (src = dy,
 // wts = w_t [Cin][R][S][Cout], taps flipped on the A side).
-template <typename T, int MODE>
+// ROWS = output rows per sweep: 2 for W <= 96 (H even) amortizes the
+// staging and barriers over twice the MFMA — the overlapping 4-row input
+// window serves both output rows, and the A fragments are shared.
+template <typename T, int MODE, int ROWS>
 __global__ __launch_bounds__(512, 1)
 void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
                      const float *__restrict__ bias, T *__restrict__ out,
@@ -51,11 +54,11 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
     // +16 extra zero rows: the ragged last q-fragment reads rows past
     // W+1; they must stay in-bounds and zero
     const int xrows = W + 18;
-    char *xbuf[3];
-    xbuf[0] = lds + 64 * WPITCH;
-    xbuf[1] = xbuf[0] + xrows * XPITCH;
-    xbuf[2] = xbuf[1] + xrows * XPITCH;
-    const int lds_bytes = 64 * WPITCH + 3 * xrows * XPITCH;
+    char *xbuf[ROWS + 2];
+#pragma unroll
+    for (int d = 0; d < ROWS + 2; ++d)
+        xbuf[d] = lds + 64 * WPITCH + d * xrows * XPITCH;
+    const int lds_bytes = 64 * WPITCH + (ROWS + 2) * xrows * XPITCH;
 
     const int t = threadIdx.x;
     const int lane = t & 63;
@@ -88,18 +91,18 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
 
     const int passes = (W + 63) / 64;   // x staging: 64 pixels per pass
 
-    for (int g = r0; g < r1; ++g) {
+    for (int g = r0; g < r1; g += ROWS) {
         const int n = g / H;
-        const int p = g % H;
+        const int p = g % H;   // ROWS==2: H even and g even => p, p+1 same image
 
         __syncthreads();   // previous row's MFMA reads done (also covers
         //                    the one-time staging on the first iteration)
 
-        // ---- stage the 3 input rows, natural [q][c] layout ----
+        // ---- stage the ROWS+2 input rows, natural [q][c] layout ----
         {
             int c8 = t & 7, pix0 = t >> 3;   // 64 pixels x 8 c-groups
 #pragma unroll
-            for (int d = 0; d < 3; ++d) {
+            for (int d = 0; d < ROWS + 2; ++d) {
                 int ih = p + d - 1;
                 bool rv = ih >= 0 && ih < H;
                 for (int ps = 0; ps < passes; ++ps) {
@@ -117,12 +120,14 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
         }
         __syncthreads();
 
-        // ---- MFMA: D[cout][q] over K = 576, 18 k-tiles of 32 ----
-        f32x4 acc[2][3];
+        // ---- MFMA: D[cout][q] per output row over K = 576 ----
+        f32x4 acc[ROWS][2][3];
 #pragma unroll
-        for (int a = 0; a < 2; ++a)
+        for (int rw = 0; rw < ROWS; ++rw)
 #pragma unroll
-            for (int b = 0; b < 3; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+            for (int a = 0; a < 2; ++a)
+#pragma unroll
+                for (int b = 0; b < 3; ++b) acc[rw][a][b] = {0.f, 0.f, 0.f, 0.f};
 
         const int oct16 = ((lane >> 4) << 4);       // k-octet byte
         const int arow = (wm * 32 + (lane & 15)) * WPITCH;
@@ -135,7 +140,6 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
             s16x8 a1 = *reinterpret_cast<const s16x8 *>(wbuf + arow + 16 * WPITCH
                                                         + abyte);
             const int rr = tap / 3, ss = tap % 3;
-            const char *xb = xbuf[rr];
             const int bbyte = chalf * 64 + oct16;
 #pragma unroll
             for (int i = 0; i < 3; ++i) {
@@ -143,37 +147,44 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
                 int q0 = (wq + i * 4) * 16;
                 // row q + s - 1, +1 halo base; rows beyond W+1 are zero pad
                 int row = q0 + (lane & 15) + ss;
-                s16x8 b = *reinterpret_cast<const s16x8 *>(
-                    xb + row * XPITCH + bbyte);
-                acc[0][i] = MfmaOp<T>::mma(a0, b, acc[0][i]);
-                acc[1][i] = MfmaOp<T>::mma(a1, b, acc[1][i]);
+#pragma unroll
+                for (int rw = 0; rw < ROWS; ++rw) {
+                    s16x8 b = *reinterpret_cast<const s16x8 *>(
+                        xbuf[rr + rw] + row * XPITCH + bbyte);
+                    acc[rw][0][i] = MfmaOp<T>::mma(a0, b, acc[rw][0][i]);
+                    acc[rw][1][i] = MfmaOp<T>::mma(a1, b, acc[rw][1][i]);
+                }
             }
         }
 
-        // ---- store y row (NHWC, 4 couts per lane) ----
-        const size_t obase = (size_t)(n * H + p) * W;
+        // ---- store the output rows (NHWC, 4 couts per lane) ----
 #pragma unroll
-        for (int mf = 0; mf < 2; ++mf) {
-            int cout = wm * 32 + mf * 16 + ((lane >> 4) << 2);
-            if (cout >= Cout) continue;
-            float b4[4] = {0.f, 0.f, 0.f, 0.f};
-            if (bias != nullptr) {
+        for (int rw = 0; rw < ROWS; ++rw) {
+            const size_t obase = (size_t)(n * H + p + rw) * W;
 #pragma unroll
-                for (int e = 0; e < 4; ++e)
-                    if (cout + e < Cout) b4[e] = bias[cout + e];
-            }
+            for (int mf = 0; mf < 2; ++mf) {
+                int cout = wm * 32 + mf * 16 + ((lane >> 4) << 2);
+                if (cout >= Cout) continue;
+                float b4[4] = {0.f, 0.f, 0.f, 0.f};
+                if (bias != nullptr) {
 #pragma unroll
-            for (int i = 0; i < 3; ++i) {
-                if (i >= nqf) break;
-                int q = (wq + i * 4) * 16 + (lane & 15);
-                if (q >= W) continue;
-                s16x4_ v;
-#pragma unroll
-                for (int e = 0; e < 4; ++e) {
-                    T tv = ElemTraits<T>::from_f32(acc[mf][i][e] + b4[e]);
-                    v[e] = *reinterpret_cast<const short *>(&tv);
+                    for (int e = 0; e < 4; ++e)
+                        if (cout + e < Cout) b4[e] = bias[cout + e];
                 }
-                *reinterpret_cast<s16x4_ *>(out + (obase + q) * ldout + cout) = v;
+#pragma unroll
+                for (int i = 0; i < 3; ++i) {
+                    if (i >= nqf) break;
+                    int q = (wq + i * 4) * 16 + (lane & 15);
+                    if (q >= W) continue;
+                    s16x4_ v;
+#pragma unroll
+                    for (int e = 0; e < 4; ++e) {
+                        T tv = ElemTraits<T>::from_f32(acc[rw][mf][i][e] + b4[e]);
+                        v[e] = *reinterpret_cast<const short *>(&tv);
+                    }
+                    *reinterpret_cast<s16x4_ *>(
+                        out + (obase + q) * ldout + cout) = v;
+                }
             }
         }
     }
@@ -192,40 +203,54 @@ bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
     if (Cin != 64) return false;               // j-chunking assumes C == 64
     if (Cout > 64 || Cout % 4 != 0) return false;
     if (W > 184 || W < 16) return false;
+    int bufs = (W <= 96) ? 4 : 3;   // ROWS==2 window below 96
     return 64 * sodkit::cdir::WPITCH
-               + 3 * (W + 18) * sodkit::cdir::XPITCH <= 160 * 1024;
+               + bufs * (W + 18) * sodkit::cdir::XPITCH <= 160 * 1024;
 }
 
 namespace {
 
-template <typename T, int MODE>
-void launch_conv_dir(const torch::Tensor &src, const torch::Tensor &wts,
-                     const float *bias_ptr, torch::Tensor &out,
-                     int N, int H, int W, int ldsrc, int ldout, int Cout) {
+template <typename T, int MODE, int ROWS>
+void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
+                       const float *bias_ptr, torch::Tensor &out,
+                       int N, int H, int W, int ldsrc, int ldout, int Cout) {
     auto stream = at::hip::getCurrentHIPStream();
     const int total_rows = N * H;
-    // ~2 row-sweeping blocks per CU
-    int blocks = std::min(512, total_rows);
-    int rows_per_blk = (total_rows + blocks - 1) / blocks;
+    // ~2 row-sweeping blocks per CU; ROWS==2 keeps block ranges pair-aligned
+    int blocks = std::min(512, total_rows / ROWS);
+    int rows_per_blk = (total_rows / ROWS + blocks - 1) / blocks * ROWS;
     blocks = (total_rows + rows_per_blk - 1) / rows_per_blk;
     const int nqf_total = (W + 15) / 16;
     const int nqf = (nqf_total + 3) / 4;     // frags per q-group (<= 3)
     const size_t shmem = 64 * sodkit::cdir::WPITCH
-                         + 3 * (size_t)(W + 18) * sodkit::cdir::XPITCH;
+                         + (ROWS + 2) * (size_t)(W + 18) * sodkit::cdir::XPITCH;
     static bool attr = false;
     if (!attr) {
         (void)hipFuncSetAttribute(
-            (const void *)sodkit::cdir::conv_dir_kernel<T, MODE>,
+            (const void *)sodkit::cdir::conv_dir_kernel<T, MODE, ROWS>,
             hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
         attr = true;
     }
-    hipLaunchKernelGGL((sodkit::cdir::conv_dir_kernel<T, MODE>), dim3(blocks),
-                       dim3(512), shmem, stream.stream(),
+    hipLaunchKernelGGL((sodkit::cdir::conv_dir_kernel<T, MODE, ROWS>),
+                       dim3(blocks), dim3(512), shmem, stream.stream(),
                        reinterpret_cast<const T *>(src.data_ptr()),
                        reinterpret_cast<const T *>(wts.data_ptr()), bias_ptr,
                        reinterpret_cast<T *>(out.data_ptr()),
                        N, H, W, ldsrc, ldout, Cout, rows_per_blk, total_rows,
                        nqf);
+}
+
+template <typename T, int MODE>
+void launch_conv_dir(const torch::Tensor &src, const torch::Tensor &wts,
+                     const float *bias_ptr, torch::Tensor &out,
+                     int N, int H, int W, int ldsrc, int ldout, int Cout) {
+    // pair rows when the window fits and pairs cannot straddle images
+    if (W <= 96 && H % 2 == 0)
+        launch_conv_dir_r<T, MODE, 2>(src, wts, bias_ptr, out, N, H, W,
+                                      ldsrc, ldout, Cout);
+    else
+        launch_conv_dir_r<T, MODE, 1>(src, wts, bias_ptr, out, N, H, W,
+                                      ldsrc, ldout, Cout);
 }
 
 }  // namespace

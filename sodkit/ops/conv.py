@@ -66,7 +66,10 @@ def _use_wgrad_dir(w: torch.Tensor, x: torch.Tensor, stride, padding,
 
 # direct fwd/dgrad for the 64-channel 3x3 decoder/AIM/SIM class
 # (csrc/conv_dir.hip): the im2col kernel re-stages every x element 9x.
-# SODKIT_CONV_DIR=0 disables (A/B).
+# Measured (r2 ubench fdir/ddir): wins at W=176 (fwd 478 vs 404 TF,
+# dgrad 482 vs 371) but LOSES at W=88 (293 vs 364 — the ragged q-tail
+# wastes 1/4 of the MFMA lanes and per-row overheads bite), so the gate
+# is W >= 128.  SODKIT_CONV_DIR=0 disables (A/B).
 _CDIR_ENV = os.environ.get("SODKIT_CONV_DIR", "")
 
 
@@ -76,7 +79,7 @@ def _use_conv_dir(cin: int, cout: int, w3: bool, x: torch.Tensor, stride,
         return False
     return (w3 and stride == 1 and padding == 1 and dilation == 1
             and cin == 64 and cout <= 64 and cout % 4 == 0
-            and 16 <= x.shape[3] <= 184)
+            and 128 <= x.shape[3] <= 184)
 
 
 def _cl_ld(x: torch.Tensor):

@@ -219,15 +219,31 @@ __global__ void bn_apply_vec_kernel(const T *__restrict__ x,
                                     const float *__restrict__ beta,
                                     T *__restrict__ y,
                                     unsigned char *__restrict__ mask,
-                                    long total8, int C8) {
+                                    long M, int C8) {
     // mask != nullptr (training, RELU): emit a 1-bit-per-element ReLU-mask
     // plane (one byte per 8-channel vector) so the backward kernels read
-    // 1/16 of y's bytes instead of y itself (bn_bwd_reduce was 3-tensor
-    // memory-bound at ~2.7 TB/s; this removes a third of its traffic).
-    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total8; i += stride) {
-        int c0 = (int)(i % C8) * 8;
+    // 1/16 of y's bytes instead of y itself.
+    // COLUMN thread mapping (bn_bwd_reduce's pattern): each thread owns a
+    // fixed 8-channel group and strides rows — the 4 per-channel
+    // coefficient vectors load ONCE instead of per element-group (the
+    // flat mapping ran at ~1.9 TB/s, coefficient-load bound).
+    int cc0 = blockIdx.y * 256;
+    int nch = min(C8 - cc0, 256);
+    int rows_pb = 256 / nch;
+    int lane_c = threadIdx.x % nch;
+    int row_off = threadIdx.x / nch;
+    if (row_off >= rows_pb) return;
+    int c0 = (cc0 + lane_c) * 8;
+    const long gcol = cc0 + lane_c;
+    float sc[8], sh[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        sc[e] = invstd[c0 + e] * gamma[c0 + e];
+        sh[e] = beta[c0 + e] - mean[c0 + e] * sc[e];
+    }
+    long mstep = (long)gridDim.x * rows_pb;
+    for (long m = (long)blockIdx.x * rows_pb + row_off; m < M; m += mstep) {
+        long i = m * C8 + gcol;
         s16x8 vx = *reinterpret_cast<const s16x8 *>(x + i * 8);
         s16x8 vr;
         if (RES) vr = *reinterpret_cast<const s16x8 *>(res + i * 8);
@@ -235,9 +251,7 @@ __global__ void bn_apply_vec_kernel(const T *__restrict__ x,
         unsigned int mb = 0;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-            float sc = invstd[c0 + e] * gamma[c0 + e];
-            float sh = beta[c0 + e] - mean[c0 + e] * sc;
-            float v = s16_as_f32<T>(vx[e]) * sc + sh;
+            float v = s16_as_f32<T>(vx[e]) * sc[e] + sh[e];
             if (RES) v += s16_as_f32<T>(vr[e]);
             if (RELU) v = fmaxf(v, 0.f);
             T tv = ElemTraits<T>::from_f32(v);
@@ -412,13 +426,30 @@ __global__ void bn_bwd_apply_vec_kernel(const T *__restrict__ dy,
                                         const float *__restrict__ gamma,
                                         const float *__restrict__ sums,
                                         float inv_n, T *__restrict__ dx,
-                                        T *__restrict__ dres, long total8,
+                                        T *__restrict__ dres, long M,
                                         int C8) {
+    // column thread mapping; see bn_apply_vec_kernel
     int C = C8 * 8;
-    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    long stride = (long)gridDim.x * blockDim.x;
-    for (; i < total8; i += stride) {
-        int c0 = (int)(i % C8) * 8;
+    int cc0 = blockIdx.y * 256;
+    int nch = min(C8 - cc0, 256);
+    int rows_pb = 256 / nch;
+    int lane_c = threadIdx.x % nch;
+    int row_off = threadIdx.x / nch;
+    if (row_off >= rows_pb) return;
+    int c0 = (cc0 + lane_c) * 8;
+    const long gcol = cc0 + lane_c;
+    float is8[8], mu8[8], gis8[8], s0[8], s1[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        is8[e] = invstd[c0 + e];
+        mu8[e] = mean[c0 + e];
+        gis8[e] = gamma[c0 + e] * is8[e];
+        s0[e] = sums[c0 + e] * inv_n;
+        s1[e] = sums[C + c0 + e] * inv_n;
+    }
+    long mstep = (long)gridDim.x * rows_pb;
+    for (long m = (long)blockIdx.x * rows_pb + row_off; m < M; m += mstep) {
+        long i = m * C8 + gcol;
         s16x8 vdy = *reinterpret_cast<const s16x8 *>(dy + i * 8);
         s16x8 vx = *reinterpret_cast<const s16x8 *>(x + i * 8);
         s16x8 vy;
@@ -430,7 +461,6 @@ __global__ void bn_bwd_apply_vec_kernel(const T *__restrict__ dy,
         s16x8 vo, vdr;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-            int c = c0 + e;
             float g = s16_as_f32<T>(vdy[e]);
             if (RELU) {
                 if (MASKED) { if (!((mb >> e) & 1)) g = 0.f; }
@@ -440,10 +470,9 @@ __global__ void bn_bwd_apply_vec_kernel(const T *__restrict__ dy,
                 T tg = ElemTraits<T>::from_f32(g);
                 vdr[e] = *reinterpret_cast<const short *>(&tg);
             }
-            float is = invstd[c];
-            float xh = (s16_as_f32<T>(vx[e]) - mean[c]) * is;
-            float v = g - (sums[c] + xh * sums[C + c]) * inv_n;
-            T tv = ElemTraits<T>::from_f32(gamma[c] * is * v);
+            float xh = (s16_as_f32<T>(vx[e]) - mu8[e]) * is8[e];
+            float v = g - (s0[e] + xh * s1[e]);
+            T tv = ElemTraits<T>::from_f32(gis8[e] * v);
             vo[e] = *reinterpret_cast<const short *>(&tv);
         }
         *reinterpret_cast<s16x8 *>(dx + i * 8) = vo;
@@ -502,6 +531,17 @@ inline int ew_blocks(long total) {
         AT_DISPATCH_CASE(at::kFloat,    [&] { using T = float; __VA_ARGS__(); }))
 
 }  // namespace
+
+// column-mapped elementwise grid: ~4 m-iterations per thread (hoisted
+// coefficients amortize; loads pipeline across the loop)
+static inline dim3 vec_ew_grid(long M, int C) {
+    int C8 = C / 8;
+    int nch = std::min(C8, 256);
+    int rows_pb = 256 / nch;
+    long xblocks = std::max<long>(1,
+        std::min<long>((M + rows_pb * 4 - 1) / (rows_pb * 4), 2048));
+    return dim3((unsigned)xblocks, (C8 + 255) / 256);
+}
 
 static inline dim3 vec_stats_grid(long M, int C) {
     int C8 = C / 8;
@@ -626,16 +666,16 @@ std::vector<torch::Tensor> bn_apply(torch::Tensor x, torch::Tensor mean,
                                 : nullptr;
         auto launch = [&](auto rl, auto rs) {
             if (C % 8 == 0 && sizeof(T) == 2) {
-                long total8 = total / 8;
+                long M = total / C;
                 hipLaunchKernelGGL(
                     (sodkit::bn_apply_vec_kernel<T, decltype(rl)::value, decltype(rs)::value>),
-                    dim3(ew_blocks(total8)), dim3(256), 0, stream.stream(),
+                    vec_ew_grid(M, C), dim3(256), 0, stream.stream(),
                     reinterpret_cast<const T *>(x.data_ptr()), resp,
                     meanf.data_ptr<float>(), invstdf.data_ptr<float>(),
                     gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
                     reinterpret_cast<T *>(y.data_ptr()),
                     emit_mask ? mask.data_ptr<unsigned char>() : nullptr,
-                    total8, C / 8);
+                    M, C / 8);
             } else
                 hipLaunchKernelGGL(
                     (sodkit::bn_apply_kernel<T, decltype(rl)::value, decltype(rs)::value>),
@@ -725,12 +765,12 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor dy, torch::Tensor y,
     DISPATCH_T(x, "bn_bwd_apply", [&] {
         auto launch = [&](auto rl, auto dr, auto mk) {
             if (C % 8 == 0 && sizeof(T) == 2) {
-                long total8 = total / 8;
+                long M = total / C;
                 hipLaunchKernelGGL(
                     (sodkit::bn_bwd_apply_vec_kernel<T, decltype(rl)::value,
                                                      decltype(dr)::value,
                                                      decltype(mk)::value>),
-                    dim3(ew_blocks(total8)), dim3(256), 0, stream.stream(),
+                    vec_ew_grid(M, C), dim3(256), 0, stream.stream(),
                     reinterpret_cast<const T *>(dy.data_ptr()),
                     reinterpret_cast<const T *>(y.data_ptr()),
                     reinterpret_cast<const T *>(x.data_ptr()),
@@ -740,7 +780,7 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor dy, torch::Tensor y,
                     gammaf.data_ptr<float>(), sumsf.data_ptr<float>(), inv_n,
                     reinterpret_cast<T *>(dx.data_ptr()),
                     need_dres ? reinterpret_cast<T *>(dres.data_ptr()) : nullptr,
-                    total8, C / 8);
+                    M, C / 8);
             } else
                 hipLaunchKernelGGL(
                     (sodkit::bn_bwd_apply_kernel<T, decltype(rl)::value, decltype(dr)::value>),

@@ -18,6 +18,9 @@ torch::Tensor conv2d_fwd_8p(torch::Tensor x, torch::Tensor w,
 torch::Tensor conv2d_dgrad_8p(torch::Tensor dy, torch::Tensor w_t, int64_t H,
                               int64_t W, int64_t stride, int64_t pad,
                               int64_t dil, int64_t lddy);
+torch::Tensor im2col_nhwc(torch::Tensor x, int64_t R, int64_t S,
+                          int64_t stride, int64_t pad, int64_t dil,
+                          int64_t ldx);
 // direct 3x3 s1 d1 p1 64-channel fwd/dgrad (csrc/conv_dir.hip)
 bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
                        int64_t dil, int64_t Cin, int64_t Cout, int64_t W);
@@ -82,6 +85,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv8_eligible", &conv8_eligible);
     m.def("conv2d_fwd_8p", &conv2d_fwd_8p);
     m.def("conv2d_dgrad_8p", &conv2d_dgrad_8p);
+    m.def("im2col_nhwc", &im2col_nhwc);
     m.def("conv_dir_eligible", &conv_dir_eligible);
     m.def("conv2d_fwd_dir", &conv2d_fwd_dir);
     m.def("conv2d_dgrad_dir", &conv2d_dgrad_dir);

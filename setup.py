@@ -39,6 +39,7 @@ SRC = [
     "csrc/conv_igemm8.hip",
     "csrc/wgrad_dir.hip",
     "csrc/conv_dir.hip",
+    "csrc/im2col.hip",
     "csrc/bnrelu.hip",
     "csrc/bilinear.hip",
     "csrc/loss.hip",

@@ -167,13 +167,23 @@ def conv2d(x: torch.Tensor, w_krsc: torch.Tensor,
     if x.is_cuda:
         cin = x.shape[1]
         if cin < 8 and cin % 8 != 0 and not x.requires_grad:
-            # stem-style inputs (RGB=3 / depth=1): pad channels to 8 so the
-            # kernel takes the vectorized/async-DMA path instead of the
-            # scalar gather (zero channels contribute nothing; w padding is
-            # differentiable so dw slices back automatically)
-            import torch.nn.functional as F
-            pad_c = 8 - cin
-            x = F.pad(x, (0, 0, 0, 0, 0, pad_c))
-            w_krsc = F.pad(w_krsc, (0, pad_c))
+            # stem-style inputs (RGB=3 / depth=1; no input grad): the old
+            # pad-to-8-channels route wasted 5/8 of the gather and MFMA
+            # (stem measured 41 TF fwd / 39 TF wgrad).  Materialize the
+            # im2col matrix once instead — the stem becomes a 1x1 conv over
+            # K8 channels and fwd/wgrad/bias ride the plain GEMM paths
+            # (dgrad is never needed here).  dw flows back through the
+            # differentiable reshape/pad of w.
+            R, S = w_krsc.shape[1], w_krsc.shape[2]
+            K = R * S * cin
+            K8 = (K + 7) // 8 * 8
+            xc, ldx = _cl_ld(x)
+            xim = ext().im2col_nhwc(xc, R, S, stride, padding, dilation, ldx)
+            w1 = w_krsc.reshape(w_krsc.shape[0], K)
+            if K8 != K:
+                import torch.nn.functional as F
+                w1 = F.pad(w1, (0, K8 - K))
+            w1 = w1.reshape(w_krsc.shape[0], 1, 1, K8)
+            return _ConvFn.apply(xim, w1, bias, 1, 0, 1)
         return _ConvFn.apply(x, w_krsc, bias, stride, padding, dilation)
     return reference.conv2d(x, w_krsc, bias, stride, padding, dilation)

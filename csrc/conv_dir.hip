@@ -34,21 +34,33 @@ namespace cdir {
 
 typedef short s16x4_ __attribute__((ext_vector_type(4)));
 
-constexpr int XPITCH = 144;    // 64ch * 2B = 128 data + 16 pad (bank stride 36)
-constexpr int WPITCH = 1168;   // 576 j * 2B = 1152 data + 16 pad (stride 292)
+// bank-spread pitches: (pitch/4) mod 64 has gcd 4 with 64, so 16
+// consecutive rows land on 16 distinct bank quads
+template <int CC> struct Pitch;
+template <> struct Pitch<64> {
+    static constexpr int X = 144;   // 128 data + 16 (stride 36)
+    static constexpr int W = 1168;  // 1152 data + 16 (stride 292)
+};
+template <> struct Pitch<32> {
+    static constexpr int X = 80;    // 64 data + 16 (stride 20)
+    static constexpr int W = 592;   // 576 data + 16 (stride 148)
+};
 
 // MODE 0: fwd (src = x, wts = w KRSC). MODE 1: dgrad (src = dy,
 // wts = w_t [Cin][R][S][Cout], taps flipped on the A side).
 // ROWS = output rows per sweep: 2 for W <= 96 (H even) amortizes the
 // staging and barriers over twice the MFMA — the overlapping 4-row input
 // window serves both output rows, and the A fragments are shared.
-template <typename T, int MODE, int ROWS>
+template <typename T, int MODE, int ROWS, int CC>
 __global__ __launch_bounds__(512, 1)
 void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
                      const float *__restrict__ bias, T *__restrict__ out,
                      int N, int H, int W, int ldsrc, int ldout, int Cout,
                      int rows_per_blk, int total_rows, int nqf) {
-    // C (inner channels of src) is 64 by eligibility; K = 576.
+    constexpr int XPITCH = Pitch<CC>::X;
+    constexpr int WPITCH = Pitch<CC>::W;
+    constexpr int WROW = 9 * CC * 2;     // weight row bytes
+    constexpr int NKT = 9 * CC / 32;     // k-tiles of 32
     extern __shared__ char lds[];
     char *wbuf = lds;                               // [Cout<=64][WPITCH]
     // +16 extra zero rows: the ragged last q-fragment reads rows past
@@ -74,14 +86,14 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
         *reinterpret_cast<long *>(lds + i) = 0;
     __syncthreads();
 
-    // ---- one-time weight staging: [row][1152B] -> padded pitch ----
+    // ---- one-time weight staging: [row][WROW bytes] -> padded pitch ----
     {
         int row = t >> 3, b16 = t & 7;
         if (row < Cout) {
-            const char *g = reinterpret_cast<const char *>(wts) + (size_t)row * 1152;
+            const char *g = reinterpret_cast<const char *>(wts) + (size_t)row * WROW;
             char *d = wbuf + row * WPITCH;
 #pragma unroll
-            for (int b = 0; b < 9; ++b) {
+            for (int b = 0; b < WROW / 128; ++b) {
                 int off = (b16 + b * 8) * 16;
                 *reinterpret_cast<s16x8 *>(d + off) =
                     *reinterpret_cast<const s16x8 *>(g + off);
@@ -89,7 +101,8 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
         }
     }
 
-    const int passes = (W + 63) / 64;   // x staging: 64 pixels per pass
+    constexpr int PIX_PER_PASS = 512 / (CC / 8);
+    const int passes = (W + PIX_PER_PASS - 1) / PIX_PER_PASS;
 
     for (int g = r0; g < r1; g += ROWS) {
         const int n = g / H;
@@ -100,13 +113,15 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
 
         // ---- stage the ROWS+2 input rows, natural [q][c] layout ----
         {
-            int c8 = t & 7, pix0 = t >> 3;   // 64 pixels x 8 c-groups
+            constexpr int NC8 = CC / 8;
+            constexpr int PIX = PIX_PER_PASS;
+            int c8 = t % NC8, pix0 = t / NC8;
 #pragma unroll
             for (int d = 0; d < ROWS + 2; ++d) {
                 int ih = p + d - 1;
                 bool rv = ih >= 0 && ih < H;
                 for (int ps = 0; ps < passes; ++ps) {
-                    int q = pix0 + ps * 64;
+                    int q = pix0 + ps * PIX;
                     if (q >= W) break;
                     s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
                     if (rv)
@@ -132,10 +147,11 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
         const int oct16 = ((lane >> 4) << 4);       // k-octet byte
         const int arow = (wm * 32 + (lane & 15)) * WPITCH;
 #pragma unroll
-        for (int kt = 0; kt < 18; ++kt) {
-            const int tap = kt >> 1, chalf = kt & 1;
+        for (int kt = 0; kt < NKT; ++kt) {
+            const int tap = (CC == 64) ? (kt >> 1) : kt;
+            const int chalf = (CC == 64) ? (kt & 1) : 0;
             const int atap = MODE ? (8 - tap) : tap;
-            const int abyte = atap * 128 + chalf * 64 + oct16;
+            const int abyte = atap * (CC * 2) + chalf * 64 + oct16;
             s16x8 a0 = *reinterpret_cast<const s16x8 *>(wbuf + arow + abyte);
             s16x8 a1 = *reinterpret_cast<const s16x8 *>(wbuf + arow + 16 * WPITCH
                                                         + abyte);
@@ -200,17 +216,18 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
 bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
                        int64_t dil, int64_t Cin, int64_t Cout, int64_t W) {
     if (R != 3 || S != 3 || stride != 1 || pad != 1 || dil != 1) return false;
-    if (Cin != 64) return false;               // j-chunking assumes C == 64
+    if (Cin != 64 && Cin != 32) return false;  // j-chunking needs C in {32,64}
     if (Cout > 64 || Cout % 4 != 0) return false;
     if (W > 184 || W < 16) return false;
     int bufs = (W <= 96) ? 4 : 3;   // ROWS==2 window below 96
-    return 64 * sodkit::cdir::WPITCH
-               + bufs * (W + 18) * sodkit::cdir::XPITCH <= 160 * 1024;
+    int xp = Cin == 64 ? sodkit::cdir::Pitch<64>::X : sodkit::cdir::Pitch<32>::X;
+    int wp = Cin == 64 ? sodkit::cdir::Pitch<64>::W : sodkit::cdir::Pitch<32>::W;
+    return 64 * wp + bufs * (W + 18) * xp <= 160 * 1024;
 }
 
 namespace {
 
-template <typename T, int MODE, int ROWS>
+template <typename T, int MODE, int ROWS, int CC>
 void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
                        const float *bias_ptr, torch::Tensor &out,
                        int N, int H, int W, int ldsrc, int ldout, int Cout) {
@@ -222,16 +239,16 @@ void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
     blocks = (total_rows + rows_per_blk - 1) / rows_per_blk;
     const int nqf_total = (W + 15) / 16;
     const int nqf = (nqf_total + 3) / 4;     // frags per q-group (<= 3)
-    const size_t shmem = 64 * sodkit::cdir::WPITCH
-                         + (ROWS + 2) * (size_t)(W + 18) * sodkit::cdir::XPITCH;
+    const size_t shmem = 64 * (size_t)sodkit::cdir::Pitch<CC>::W
+                         + (ROWS + 2) * (size_t)(W + 18) * sodkit::cdir::Pitch<CC>::X;
     static bool attr = false;
     if (!attr) {
         (void)hipFuncSetAttribute(
-            (const void *)sodkit::cdir::conv_dir_kernel<T, MODE, ROWS>,
+            (const void *)sodkit::cdir::conv_dir_kernel<T, MODE, ROWS, CC>,
             hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
         attr = true;
     }
-    hipLaunchKernelGGL((sodkit::cdir::conv_dir_kernel<T, MODE, ROWS>),
+    hipLaunchKernelGGL((sodkit::cdir::conv_dir_kernel<T, MODE, ROWS, CC>),
                        dim3(blocks), dim3(512), shmem, stream.stream(),
                        reinterpret_cast<const T *>(src.data_ptr()),
                        reinterpret_cast<const T *>(wts.data_ptr()), bias_ptr,
@@ -243,14 +260,21 @@ void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
 template <typename T, int MODE>
 void launch_conv_dir(const torch::Tensor &src, const torch::Tensor &wts,
                      const float *bias_ptr, torch::Tensor &out,
-                     int N, int H, int W, int ldsrc, int ldout, int Cout) {
+                     int N, int H, int W, int ldsrc, int ldout, int Cout,
+                     int CC) {
     // pair rows when the window fits and pairs cannot straddle images
-    if (W <= 96 && H % 2 == 0)
-        launch_conv_dir_r<T, MODE, 2>(src, wts, bias_ptr, out, N, H, W,
-                                      ldsrc, ldout, Cout);
-    else
-        launch_conv_dir_r<T, MODE, 1>(src, wts, bias_ptr, out, N, H, W,
-                                      ldsrc, ldout, Cout);
+    bool pair = W <= 96 && H % 2 == 0;
+    if (CC == 64) {
+        if (pair) launch_conv_dir_r<T, MODE, 2, 64>(src, wts, bias_ptr, out,
+                                                    N, H, W, ldsrc, ldout, Cout);
+        else launch_conv_dir_r<T, MODE, 1, 64>(src, wts, bias_ptr, out,
+                                               N, H, W, ldsrc, ldout, Cout);
+    } else {
+        if (pair) launch_conv_dir_r<T, MODE, 2, 32>(src, wts, bias_ptr, out,
+                                                    N, H, W, ldsrc, ldout, Cout);
+        else launch_conv_dir_r<T, MODE, 1, 32>(src, wts, bias_ptr, out,
+                                               N, H, W, ldsrc, ldout, Cout);
+    }
 }
 
 }  // namespace
@@ -275,11 +299,11 @@ torch::Tensor conv2d_fwd_dir(torch::Tensor x, torch::Tensor w,
     AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_fwd_dir",
         AT_DISPATCH_CASE(at::kBFloat16, [&] {
             launch_conv_dir<__hip_bfloat16, 0>(x, w, bp, out, N, H, W_,
-                                               (int)ld, Cout, Cout);
+                                               (int)ld, Cout, Cout, C);
         })
         AT_DISPATCH_CASE(at::kHalf, [&] {
             launch_conv_dir<__half, 0>(x, w, bp, out, N, H, W_, (int)ld,
-                                       Cout, Cout);
+                                       Cout, Cout, C);
         }));
     return out;
 }
@@ -299,11 +323,11 @@ torch::Tensor conv2d_dgrad_dir(torch::Tensor dy, torch::Tensor w_t,
     AT_DISPATCH_SWITCH(dy.scalar_type(), "conv2d_dgrad_dir",
         AT_DISPATCH_CASE(at::kBFloat16, [&] {
             launch_conv_dir<__hip_bfloat16, 1>(dy, w_t, nullptr, dx, N, H, W_,
-                                               (int)lddy, Cin, Cin);
+                                               (int)lddy, Cin, Cin, Cout);
         })
         AT_DISPATCH_CASE(at::kHalf, [&] {
             launch_conv_dir<__half, 1>(dy, w_t, nullptr, dx, N, H, W_,
-                                       (int)lddy, Cin, Cin);
+                                       (int)lddy, Cin, Cin, Cout);
         }));
     return dx;
 }

@@ -78,7 +78,7 @@ def _use_conv_dir(cin: int, cout: int, w3: bool, x: torch.Tensor, stride,
     if _CDIR_ENV == "0":
         return False
     return (w3 and stride == 1 and padding == 1 and dilation == 1
-            and cin == 64 and cout <= 64 and cout % 4 == 0
+            and cin in (32, 64) and cout <= 64 and cout % 4 == 0
             and 128 <= x.shape[3] <= 184)
 
 

@@ -42,6 +42,8 @@ CONV_SHAPES = [
     (2, 256, 22, 1024, 1, 1, 0, 1),    # wgrad 256x256 8-wave tile (big)
     (1, 64, 176, 64, 3, 1, 1, 1),      # conv_dir max-W decoder class
     (2, 64, 33, 32, 3, 1, 1, 1),       # conv_dir odd W + Cout 32 (masked)
+    (1, 32, 176, 32, 3, 1, 1, 1),      # conv_dir CC=32 (SIM inner convs)
+    (2, 32, 64, 64, 3, 1, 1, 1),       # conv_dir CC=32 paired rows
 ]
 
 

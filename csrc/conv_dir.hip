@@ -92,9 +92,10 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
         if (row < Cout) {
             const char *g = reinterpret_cast<const char *>(wts) + (size_t)row * WROW;
             char *d = wbuf + row * WPITCH;
-#pragma unroll
-            for (int b = 0; b < WROW / 128; ++b) {
-                int off = (b16 + b * 8) * 16;
+            // WROW/16 chunks per row (NOT necessarily a multiple of 8:
+            // CC=32 has 36 — the multiple-of-128 loop dropped the tail)
+            for (int c16 = b16; c16 < WROW / 16; c16 += 8) {
+                int off = c16 * 16;
                 *reinterpret_cast<s16x8 *>(d + off) =
                     *reinterpret_cast<const s16x8 *>(g + off);
             }

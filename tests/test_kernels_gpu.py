@@ -230,6 +230,46 @@ def test_sgd_fused_matches_reference():
                   what=f"sgd step {step}")
 
 
+CDIR_SHAPES = [
+    # (N, C, H, Cout): conv_dir's whole domain, direct kernel calls so
+    # every (CC, ROWS) variant is covered regardless of dispatch gates
+    (1, 64, 176, 64),    # CC=64 ROWS=1
+    (2, 64, 88, 64),     # CC=64 ROWS=2
+    (2, 64, 33, 32),     # CC=64 odd W, masked Cout
+    (1, 32, 176, 32),    # CC=32 ROWS=1
+    (2, 32, 64, 64),     # CC=32 ROWS=2
+    (2, 32, 91, 48),     # CC=32 odd W (ROWS=1: H odd), Cout 48
+]
+
+
+@pytest.mark.parametrize("shape", CDIR_SHAPES)
+@pytest.mark.parametrize("mode", ["fwd", "dgrad"])
+def test_conv_dir_matches_reference(shape, mode):
+    from sodkit import _C
+    N, C, H, Cout, = shape
+    torch.manual_seed(9)
+    if mode == "fwd":
+        assert _C.conv_dir_eligible(3, 3, 1, 1, 1, C, Cout, H)
+        x = _cl(torch.randn(N, C, H, H, device="cuda").to(torch.bfloat16))
+        w = (torch.randn(Cout, 3, 3, C, device="cuda") * 0.1).to(torch.bfloat16)
+        b = torch.randn(Cout, device="cuda").to(torch.bfloat16)
+        y = _C.conv2d_fwd_dir(x, w, b, C)
+        ref = F.conv2d(x.float(), w.float().permute(0, 3, 1, 2), b.float(),
+                       padding=1)
+        close(y, ref, what=f"conv_dir fwd {shape}")
+    else:
+        # roles swap: dy has C channels (= the GEMM inner dim), dx gets Cout
+        assert _C.conv_dir_eligible(3, 3, 1, 1, 1, C, Cout, H)
+        dy = _cl(torch.randn(N, C, H, H, device="cuda").to(torch.bfloat16))
+        w = (torch.randn(C, 3, 3, Cout, device="cuda") * 0.1).to(torch.bfloat16)
+        w_t = w.permute(3, 1, 2, 0).contiguous()   # [Cout][3][3][C]
+        dx = _C.conv2d_dgrad_dir(dy, w_t, C)
+        xr = torch.zeros(N, Cout, H, H, device="cuda", requires_grad=True)
+        yr = F.conv2d(xr, w.float().permute(0, 3, 1, 2), None, padding=1)
+        yr.backward(dy.float())
+        close(dx, xr.grad, what=f"conv_dir dgrad {shape}")
+
+
 WDIR_SHAPES = [
     # (N, C, H, Cout): 3x3 s1 d1 p1 only — the kernel's whole domain
     (2, 64, 33, 64),     # decoder/AIM class, odd W (k-tail + stage tail)

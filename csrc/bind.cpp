@@ -28,6 +28,11 @@ torch::Tensor conv2d_fwd_dir(torch::Tensor x, torch::Tensor w,
                              torch::Tensor bias, int64_t ld);
 torch::Tensor conv2d_dgrad_dir(torch::Tensor dy, torch::Tensor w_t,
                                int64_t lddy);
+// VALU 1x1 wgrad (csrc/wgrad_k1.hip)
+bool wgrad_k1_eligible(int64_t R, int64_t S, int64_t dil, int64_t pad,
+                       int64_t C, int64_t Cout);
+torch::Tensor conv2d_wgrad_k1(torch::Tensor x, torch::Tensor dy,
+                              int64_t stride, int64_t ldx, int64_t lddy);
 // direct 3x3 s1 d1 p1 wgrad (csrc/wgrad_dir.hip)
 bool wgrad_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
                         int64_t dil, int64_t C, int64_t W);
@@ -89,6 +94,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_dir_eligible", &conv_dir_eligible);
     m.def("conv2d_fwd_dir", &conv2d_fwd_dir);
     m.def("conv2d_dgrad_dir", &conv2d_dgrad_dir);
+    m.def("wgrad_k1_eligible", &wgrad_k1_eligible);
+    m.def("conv2d_wgrad_k1", &conv2d_wgrad_k1);
     m.def("wgrad_dir_eligible", &wgrad_dir_eligible);
     m.def("conv2d_wgrad_dir", &conv2d_wgrad_dir);
     m.def("bn_stats", &bn_stats);

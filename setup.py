@@ -40,6 +40,7 @@ SRC = [
     "csrc/wgrad_dir.hip",
     "csrc/conv_dir.hip",
     "csrc/im2col.hip",
+    "csrc/wgrad_k1.hip",
     "csrc/bnrelu.hip",
     "csrc/bilinear.hip",
     "csrc/loss.hip",

@@ -64,6 +64,21 @@ def _use_wgrad_dir(w: torch.Tensor, x: torch.Tensor, stride, padding,
             and 64 <= x.shape[3] <= 184)
 
 
+# VALU 1x1 wgrad (csrc/wgrad_k1.hip): the tall-skinny 1x1 weight-grad
+# reductions sat at 108-187 TF on the MFMA transpose-staging path; the
+# register-accumulating VALU kernel has no transpose.  SODKIT_WGRAD_K1=0
+# disables (A/B).
+_WK1_ENV = os.environ.get("SODKIT_WGRAD_K1", "")
+
+
+def _use_wgrad_k1(w: torch.Tensor, padding, dilation) -> bool:
+    if _WK1_ENV == "0":
+        return False
+    return (w.shape[1] == 1 and w.shape[2] == 1 and padding == 0
+            and dilation == 1 and w.shape[3] % 16 == 0
+            and w.shape[0] % 4 == 0 and w.shape[0] >= 4)
+
+
 # direct fwd/dgrad for the 64-channel 3x3 decoder/AIM/SIM class
 # (csrc/conv_dir.hip): the im2col kernel re-stages every x element 9x.
 # Measured (r2 ubench fdir/ddir): wins at W=176 (fwd 478 vs 404 TF,
@@ -152,6 +167,9 @@ class _ConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             if _use_wgrad_dir(w, x, stride, padding, dilation):
                 dw = ext().conv2d_wgrad_dir(x, dy, ldx, lddy).to(w.dtype)
+            elif _use_wgrad_k1(w, padding, dilation):
+                dw = ext().conv2d_wgrad_k1(x, dy, stride, ldx,
+                                           lddy).to(w.dtype)
             else:
                 dw = ext().conv2d_wgrad(x, dy, w.shape[1], w.shape[2],
                                         stride, padding, dilation, ldx,

@@ -230,6 +230,33 @@ def test_sgd_fused_matches_reference():
                   what=f"sgd step {step}")
 
 
+WK1_SHAPES = [
+    # (N, C, H, Cout, stride): 1x1 wgrad VALU kernel domain
+    (2, 256, 28, 64, 1),     # bottleneck reduce
+    (2, 64, 28, 256, 1),     # expand (multi cout chunk)
+    (2, 256, 14, 512, 2),    # downsample stride 2
+    (1, 2048, 11, 64, 1),    # trans conv (8 c chunks)
+    (2, 48, 20, 36, 1),      # C and Cout tails (masked lanes)
+]
+
+
+@pytest.mark.parametrize("shape", WK1_SHAPES)
+def test_wgrad_k1_matches_reference(shape):
+    from sodkit import _C
+    N, C, H, Cout, st = shape
+    torch.manual_seed(11)
+    assert _C.wgrad_k1_eligible(1, 1, 1, 0, C, Cout)
+    x = _cl(torch.randn(N, C, H, H, device="cuda").to(torch.bfloat16))
+    P = (H - 1) // st + 1
+    dy = _cl(torch.randn(N, Cout, P, P, device="cuda").to(torch.bfloat16))
+    dw = _C.conv2d_wgrad_k1(x, dy, st, C, Cout)
+    wr = torch.zeros(Cout, C, 1, 1, device="cuda", requires_grad=True)
+    yr = F.conv2d(x.float(), wr, None, stride=st)
+    yr.backward(dy.float())
+    close(dw, wr.grad.permute(0, 2, 3, 1), rtol=3e-2, atol=3e-2,
+          what=f"wgrad_k1 {shape}")
+
+
 CDIR_SHAPES = [
     # (N, C, H, Cout): conv_dir's whole domain, direct kernel calls so
     # every (CC, ROWS) variant is covered regardless of dispatch gates

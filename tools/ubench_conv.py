@@ -43,7 +43,7 @@ def bench(fn, iters=10):
 
 def main():
     dev = "cuda"
-    print(f"{'shape':>12} {'M':>8} {'K':>6} {'N':>6} | {'fwd_tf':>7} {'dgrad_tf':>8} {'wgrad_tf':>8} | {'fwd8p':>7} {'dgrad8p':>7} {'wdir':>7} {'fdir':>7} {'ddir':>7} wg8")
+    print(f"{'shape':>12} {'M':>8} {'K':>6} {'N':>6} | {'fwd_tf':>7} {'dgrad_tf':>8} {'wgrad_tf':>8} | {'fwd8p':>7} {'dgrad8p':>7} {'wdir':>7} {'wk1':>7} {'fdir':>7} {'ddir':>7} wg8")
     for name, N, Cin, H, Cout, k, st, pad, dil in SHAPES:
         x = torch.randn(N, Cin, H, H, device=dev).to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)
@@ -74,6 +74,11 @@ def main():
         if _C.wgrad_dir_eligible(k, k, st, pad, dil, Cin, H):
             t = bench(lambda: _C.conv2d_wgrad_dir(x, y, Cin, Cout))
             wd = f"{flop / t / 1e12:7.0f}"
+        wk1 = "      -"
+        if k == 1 and pad == 0 and dil == 1 and _C.wgrad_k1_eligible(
+                1, 1, 1, 0, Cin, Cout):
+            t = bench(lambda: _C.conv2d_wgrad_k1(x, y, st, Cin, Cout))
+            wk1 = f"{flop / t / 1e12:7.0f}"
         fdir = ddir = "      -"
         if _C.conv_dir_eligible(k, k, st, pad, dil, Cin, Cout, H):
             t = bench(lambda: _C.conv2d_fwd_dir(x, w, nob, Cin))
@@ -83,7 +88,7 @@ def main():
             ddir = f"{flop / t / 1e12:7.0f}"
         print(f"{name:>12} {M:>8} {K:>6} {Cout:>6} | "
               f"{flop / t_f / 1e12:7.0f} {flop / t_d / 1e12:8.0f} "
-              f"{flop / t_w / 1e12:8.0f} | {f8} {d8} {wd} {fdir} {ddir} wg8={wg8}")
+              f"{flop / t_w / 1e12:8.0f} | {f8} {d8} {wd} {wk1} {fdir} {ddir} wg8={wg8}")
 
 
 if __name__ == "__main__":

@@ -64,15 +64,18 @@ def _use_wgrad_dir(w: torch.Tensor, x: torch.Tensor, stride, padding,
             and 64 <= x.shape[3] <= 184)
 
 
-# VALU 1x1 wgrad (csrc/wgrad_k1.hip): the tall-skinny 1x1 weight-grad
-# reductions sat at 108-187 TF on the MFMA transpose-staging path; the
-# register-accumulating VALU kernel has no transpose.  SODKIT_WGRAD_K1=0
-# disables (A/B).
+# VALU 1x1 wgrad (csrc/wgrad_k1.hip) — NEGATIVE RESULT, opt-in only
+# (SODKIT_WGRAD_K1=1).  Hypothesis was that the MFMA path's transpose
+# staging (108-187 TF) left room for a register-accumulating VALU
+# reduction; measured r2: 10-38 TF — the per-pixel convert+FMA issue
+# stream is ~5x slower than the staged MFMA math even though it avoids
+# the transpose.  Kept (numerics-tested) as documentation; do not
+# re-enable without restructuring around packed dot2 and wider m-tiles.
 _WK1_ENV = os.environ.get("SODKIT_WGRAD_K1", "")
 
 
 def _use_wgrad_k1(w: torch.Tensor, padding, dilation) -> bool:
-    if _WK1_ENV == "0":
+    if _WK1_ENV != "1":
         return False
     return (w.shape[1] == 1 and w.shape[2] == 1 and padding == 0
             and dilation == 1 and w.shape[3] % 16 == 0

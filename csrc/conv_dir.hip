@@ -33,17 +33,31 @@ namespace sodkit {
 namespace cdir {
 
 typedef short s16x4_ __attribute__((ext_vector_type(4)));
+typedef __attribute__((address_space(3))) void lds_void_t;
+typedef __attribute__((address_space(1))) const void gbl_void_t;
 
-// bank-spread pitches: (pitch/4) mod 64 has gcd 4 with 64, so 16
-// consecutive rows land on 16 distinct bank quads
+SOD_DEVINL void glds16(const void *gsrc, char *lds_dst) {
+    __builtin_amdgcn_global_load_lds((const gbl_void_t *)gsrc,
+                                     (lds_void_t *)lds_dst, 16, 0, 0);
+}
+
+// x images keep their natural row size (CC*2 bytes) so `global_load_lds`
+// can stage them; bank conflicts are avoided by a 16B-block XOR swizzle
+// applied to the glds SOURCE address and inverted on the read side:
+//   block' = block ^ s(row),  s(row) = (row & (B-1)) ^ ((row >> log2B) & (B-1))
+// which makes the 16 consecutive rows of a b128 fragment read hit 16
+// distinct bank quads.  The weight image keeps a padded pitch (register-
+// staged once; rows are 1152/576 B, not glds-friendly anyway).
 template <int CC> struct Pitch;
 template <> struct Pitch<64> {
-    static constexpr int X = 144;   // 128 data + 16 (stride 36)
+    static constexpr int X = 128;
     static constexpr int W = 1168;  // 1152 data + 16 (stride 292)
+    SOD_DEVINL static int swz(int row) { return (row & 7) ^ ((row >> 3) & 7); }
 };
 template <> struct Pitch<32> {
-    static constexpr int X = 80;    // 64 data + 16 (stride 20)
+    static constexpr int X = 64;
     static constexpr int W = 592;   // 576 data + 16 (stride 148)
+    SOD_DEVINL static int swz(int row) { return (row & 3) ^ ((row >> 2) & 3); }
 };
 
 // MODE 0: fwd (src = x, wts = w KRSC). MODE 1: dgrad (src = dy,
@@ -55,6 +69,7 @@ template <typename T, int MODE, int ROWS, int CC>
 __global__ __launch_bounds__(512, 1)
 void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
                      const float *__restrict__ bias, T *__restrict__ out,
+                     const T *__restrict__ zero_page,
                      int N, int H, int W, int ldsrc, int ldout, int Cout,
                      int rows_per_blk, int total_rows, int nqf) {
     constexpr int XPITCH = Pitch<CC>::X;
@@ -112,28 +127,33 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
         __syncthreads();   // previous row's MFMA reads done (also covers
         //                    the one-time staging on the first iteration)
 
-        // ---- stage the ROWS+2 input rows, natural [q][c] layout ----
+        // ---- stage the ROWS+2 input rows via glds (async DMA): one
+        //      instruction fills 1024 B = 1024/XPITCH LDS rows; the
+        //      per-lane SOURCE is pre-swizzled so reads are conflict-free
         {
-            constexpr int NC8 = CC / 8;
-            constexpr int PIX = PIX_PER_PASS;
-            int c8 = t % NC8, pix0 = t / NC8;
+            constexpr int BLKS = CC / 8;            // 16B blocks per row
+            constexpr int RPI = 64 / BLKS;          // LDS rows per glds
+            const int my_row = lane / BLKS;         // row within the chunk
+            const int my_blk = lane % BLKS;
+            const int nchunks = (W + RPI - 1) / RPI;
 #pragma unroll
             for (int d = 0; d < ROWS + 2; ++d) {
                 int ih = p + d - 1;
                 bool rv = ih >= 0 && ih < H;
-                for (int ps = 0; ps < passes; ++ps) {
-                    int q = pix0 + ps * PIX;
-                    if (q >= W) break;
-                    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-                    if (rv)
-                        v = *reinterpret_cast<const s16x8 *>(
-                            src + ((size_t)(n * H + ih) * W + q) * ldsrc
-                                + c8 * 8);
-                    *reinterpret_cast<s16x8 *>(
-                        xbuf[d] + (q + 1) * XPITCH + c8 * 16) = v;
+                for (int ch = wave; ch < nchunks; ch += 8) {
+                    int q = ch * RPI + my_row;      // pixel this lane feeds
+                    int ldsrow = q + 1;             // +halo
+                    int sblk = my_blk ^ Pitch<CC>::swz(ldsrow);
+                    const void *g = (rv && q < W)
+                        ? (const void *)(src + ((size_t)(n * H + ih) * W + q)
+                                             * ldsrc + sblk * 8)
+                        : (const void *)zero_page;
+                    glds16(g, xbuf[d] + (size_t)(ch * RPI + 1) * XPITCH
+                                  + lane * 16);
                 }
             }
         }
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __syncthreads();
 
         // ---- MFMA: D[cout][q] per output row over K = 576 ----
@@ -157,17 +177,18 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
             s16x8 a1 = *reinterpret_cast<const s16x8 *>(wbuf + arow + 16 * WPITCH
                                                         + abyte);
             const int rr = tap / 3, ss = tap % 3;
-            const int bbyte = chalf * 64 + oct16;
+            const int bblk = (chalf * 64 + oct16) >> 4;  // 16B block index
 #pragma unroll
             for (int i = 0; i < 3; ++i) {
                 if (i >= nqf) break;
                 int q0 = (wq + i * 4) * 16;
                 // row q + s - 1, +1 halo base; rows beyond W+1 are zero pad
                 int row = q0 + (lane & 15) + ss;
+                const int bswz = (bblk ^ Pitch<CC>::swz(row)) << 4;
 #pragma unroll
                 for (int rw = 0; rw < ROWS; ++rw) {
                     s16x8 b = *reinterpret_cast<const s16x8 *>(
-                        xbuf[rr + rw] + row * XPITCH + bbyte);
+                        xbuf[rr + rw] + row * XPITCH + bswz);
                     acc[rw][0][i] = MfmaOp<T>::mma(a0, b, acc[rw][0][i]);
                     acc[rw][1][i] = MfmaOp<T>::mma(a1, b, acc[rw][1][i]);
                 }
@@ -214,6 +235,15 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
 // Host glue
 // ===========================================================================
 
+namespace {
+void *cdir_zero_page(const torch::Tensor &like) {
+    static torch::Tensor z;
+    if (!z.defined() || z.device() != like.device())
+        z = torch::zeros({32}, like.options().dtype(torch::kFloat32));
+    return z.data_ptr();
+}
+}  // namespace
+
 bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
                        int64_t dil, int64_t Cin, int64_t Cout, int64_t W) {
     if (R != 3 || S != 3 || stride != 1 || pad != 1 || dil != 1) return false;
@@ -254,6 +284,7 @@ void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
                        reinterpret_cast<const T *>(src.data_ptr()),
                        reinterpret_cast<const T *>(wts.data_ptr()), bias_ptr,
                        reinterpret_cast<T *>(out.data_ptr()),
+                       reinterpret_cast<const T *>(cdir_zero_page(src)),
                        N, H, W, ldsrc, ldout, Cout, rows_per_blk, total_rows,
                        nqf);
 }

@@ -607,16 +607,18 @@ __global__ void channel_sum_kernel(const T *__restrict__ x, float *__restrict__ 
     atomicAdd(&out[c], s);
 }
 
-// small-C variant (C <= 8): the channel-per-thread kernel leaves 255/256
-// threads idle at C=1 (the 1-channel head's bias grad measured 220 us);
-// here every thread strides rows, accumulates all C channels in registers,
-// and the block reduces through LDS once.
+// small-C variant (C <= 16): the channel-per-thread kernel leaves most
+// threads idle at tiny C (the 1-channel head's bias grad measured 220 us;
+// HDFNet's 9-channel DDPM generators 43 us); here every thread strides
+// rows, accumulates all C channels in registers, and the block reduces
+// through LDS once.
 template <typename T, bool FLAT>
 __global__ void channel_sum_small_kernel(const T *__restrict__ x,
                                          float *__restrict__ out,
                                          long M, int C, int ld) {
-    __shared__ float red[256 * 8];
-    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    __shared__ float red[256 * 16];
+    float acc[16] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f,
+                     0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
     if (FLAT) {   // C==1, ld==1: contiguous — vector reads
         const long tot8 = M >> 3;
         for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < tot8;
@@ -635,13 +637,13 @@ __global__ void channel_sum_small_kernel(const T *__restrict__ x,
                 acc[c] += ElemTraits<T>::to_f32(x[m * ld + c]);
     }
 #pragma unroll
-    for (int c = 0; c < 8; ++c) red[threadIdx.x * 8 + c] = acc[c];
+    for (int c = 0; c < 16; ++c) red[threadIdx.x * 16 + c] = acc[c];
     __syncthreads();
     for (int w = 128; w > 0; w >>= 1) {
         if (threadIdx.x < w)
 #pragma unroll
-            for (int c = 0; c < 8; ++c)
-                red[threadIdx.x * 8 + c] += red[(threadIdx.x + w) * 8 + c];
+            for (int c = 0; c < 16; ++c)
+                red[threadIdx.x * 16 + c] += red[(threadIdx.x + w) * 16 + c];
         __syncthreads();
     }
     if (threadIdx.x < C) atomicAdd(&out[threadIdx.x], red[threadIdx.x]);
@@ -878,7 +880,7 @@ torch::Tensor channel_sum(torch::Tensor x, int64_t ld) {
     long M = (long)N * H * W_;
     auto out = torch::zeros({C}, x.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
-    bool small = C <= 8;
+    bool small = C <= 16;
     bool flat = (C == 1 && ld == 1);
     dim3 grid_small(std::min<long>((M + 2047) / 2048, 512));
     dim3 grid(std::min<long>((M + 255) / 256, 1024), (C + 255) / 256);

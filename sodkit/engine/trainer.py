@@ -236,16 +236,26 @@ class Solver:
             self.iter_count += 1
             return loss.detach()
         if self._graph is None:
-            self._g_lr = torch.zeros((), device=self.device)
-            self._g_img = img.clone()
-            self._g_mask = mask.clone()
-            self._g_depth = depth.clone() if depth is not None else None
-            torch.cuda.synchronize()
-            self._graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(self._graph):
-                self._g_loss = self._step_body(self._g_img, self._g_mask,
-                                               self._g_depth, self._g_lr)
-            torch.cuda.synchronize()
+            try:
+                self._g_lr = torch.zeros((), device=self.device)
+                self._g_img = img.clone()
+                self._g_mask = mask.clone()
+                self._g_depth = depth.clone() if depth is not None else None
+                torch.cuda.synchronize()
+                self._graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(self._graph):
+                    self._g_loss = self._step_body(self._g_img, self._g_mask,
+                                                   self._g_depth, self._g_lr)
+                torch.cuda.synchronize()
+            except Exception as e:  # capture failure must not kill training
+                self.log.info(f"hipGraph capture failed ({e!r}); "
+                              f"falling back to eager steps")
+                self._graph = None
+                self._graph_ok = False
+                loss = self._step_body(img, mask, depth,
+                                       self._lr_at(self.iter_count))
+                self.iter_count += 1
+                return loss.detach()
         self._g_img.copy_(img, non_blocking=True)
         self._g_mask.copy_(mask, non_blocking=True)
         if depth is not None:

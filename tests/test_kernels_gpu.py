@@ -271,30 +271,31 @@ CDIR_SHAPES = [
 
 @pytest.mark.parametrize("shape", CDIR_SHAPES)
 @pytest.mark.parametrize("mode", ["fwd", "dgrad"])
-def test_conv_dir_matches_reference(shape, mode):
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_conv_dir_matches_reference(shape, mode, dtype):
     from sodkit import _C
     N, C, H, Cout, = shape
     torch.manual_seed(9)
     if mode == "fwd":
         assert _C.conv_dir_eligible(3, 3, 1, 1, 1, C, Cout, H)
-        x = _cl(torch.randn(N, C, H, H, device="cuda").to(torch.bfloat16))
-        w = (torch.randn(Cout, 3, 3, C, device="cuda") * 0.1).to(torch.bfloat16)
-        b = torch.randn(Cout, device="cuda").to(torch.bfloat16)
+        x = _cl(torch.randn(N, C, H, H, device="cuda").to(dtype))
+        w = (torch.randn(Cout, 3, 3, C, device="cuda") * 0.1).to(dtype)
+        b = torch.randn(Cout, device="cuda").to(dtype)
         y = _C.conv2d_fwd_dir(x, w, b, C)
         ref = F.conv2d(x.float(), w.float().permute(0, 3, 1, 2), b.float(),
                        padding=1)
-        close(y, ref, what=f"conv_dir fwd {shape}")
+        close(y, ref, what=f"conv_dir fwd {shape} {dtype}")
     else:
         # roles swap: dy has C channels (= the GEMM inner dim), dx gets Cout
         assert _C.conv_dir_eligible(3, 3, 1, 1, 1, C, Cout, H)
-        dy = _cl(torch.randn(N, C, H, H, device="cuda").to(torch.bfloat16))
-        w = (torch.randn(C, 3, 3, Cout, device="cuda") * 0.1).to(torch.bfloat16)
+        dy = _cl(torch.randn(N, C, H, H, device="cuda").to(dtype))
+        w = (torch.randn(C, 3, 3, Cout, device="cuda") * 0.1).to(dtype)
         w_t = w.permute(3, 1, 2, 0).contiguous()   # [Cout][3][3][C]
         dx = _C.conv2d_dgrad_dir(dy, w_t, C)
         xr = torch.zeros(N, Cout, H, H, device="cuda", requires_grad=True)
         yr = F.conv2d(xr, w.float().permute(0, 3, 1, 2), None, padding=1)
         yr.backward(dy.float())
-        close(dx, xr.grad, what=f"conv_dir dgrad {shape}")
+        close(dx, xr.grad, what=f"conv_dir dgrad {shape} {dtype}")
 
 
 WDIR_SHAPES = [

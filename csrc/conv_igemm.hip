@@ -452,6 +452,125 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
         for (int b = 0; b < FJ; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     constexpr int DNCH = BCOUT * (BKM / 4) / 8;   // dy chunks (4 m x 8 cout)
+    // ---- cross-K-step software pipeline (PIPE variants): the global
+    // loads for step t+1 are ISSUED before step t's MFMA, so their ~600ns
+    // latency overlaps the math.  PMC evidence (r2): the 128x128 variant
+    // ran at 8% MfmaUtil with only 6% LDS-conflict cycles — latency-bound.
+    // The staged registers cost ~50 VGPRs across the MFMA region, which
+    // only the 128x128 variant affords without spills/occupancy loss; the
+    // other variants keep the original interleaved two-barrier body.
+    constexpr bool PIPE = (BCOUT == 128 && BJ == 128 && NT == 256);
+
+    if constexpr (PIPE) {
+        static_assert(BCOUT * (BKM / 4) / 8 == NT || !PIPE, "one dy chunk/thread");
+        const int dy_ml = (t / (BCOUT / 8)) * 4;
+        const int dy_cv = (t % (BCOUT / 8)) * 8;
+        s16x8 vdy[4];
+        s16x8 vx[XCH][4];
+
+        auto load_dy = [&](long m0) {
+            int cout = tile_c + dy_cv;
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+                vdy[u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                long m = m0 + dy_ml + u;
+                if (m < m_end && cout < Cout) {
+                    const T *src = dy + m * lddy + cout;
+                    if (dyvec_ok && cout + 7 < Cout) {
+                        vdy[u] = *reinterpret_cast<const s16x8 *>(src);
+                    } else {
+#pragma unroll
+                        for (int e = 0; e < 8; ++e)
+                            if (cout + e < Cout)
+                                vdy[u][e] = reinterpret_cast<const short *>(src)[e];
+                    }
+                }
+            }
+        };
+        auto load_x = [&](long m0) {
+#pragma unroll
+            for (int i = 0; i < XCH; ++i) {
+                if (!xact[i]) continue;
+#pragma unroll
+                for (int u = 0; u < 4; ++u) {
+                    vx[i][u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                    long m = m0 + xml[i] + u;
+                    if (xj[i] < Kw && m < m_end) {
+                        int n = xgn[i], p = xgp[i], q = xgq[i] + u;
+                        if (q >= Q) { q -= Q; if (++p >= P) { p = 0; ++n; } }
+                        if (xvec_ok) {
+                            long off = gather_off<0>(n, p, q, xr[i], xs[i], xc[i],
+                                                     H, W, C, ldx, stride, pad, dil);
+                            if (off >= 0)
+                                vx[i][u] = *reinterpret_cast<const s16x8 *>(x + off);
+                        } else {
+#pragma unroll
+                            for (int e = 0; e < 8; ++e) {
+                                int j = xj[i] + e;
+                                if (j < Kw) {
+                                    int rs = j / C, c = j % C;
+                                    long off = gather_off<0>(n, p, q, rs / S,
+                                                             rs % S, c,
+                                                             H, W, C, ldx, stride, pad, dil);
+                                    if (off >= 0)
+                                        vx[i][u][e] = reinterpret_cast<const short *>(x)[off];
+                                }
+                            }
+                        }
+                    }
+                }
+                int q = xgq[i] + BKM, p = xgp[i], n = xgn[i];
+                while (q >= Q) { q -= Q; ++p; }
+                while (p >= P) { p -= P; ++n; }
+                xgq[i] = q; xgp[i] = p; xgn[i] = n;
+            }
+        };
+
+        load_dy(m_begin); load_x(m_begin);
+        for (long m0 = m_begin; m0 < m_end; m0 += BKM) {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                s16x4 w4 = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
+                *reinterpret_cast<s16x4 *>(
+                    dyT_lds + swz<BKM>(dy_cv + e, dy_ml * 2)) = w4;
+            }
+#pragma unroll
+            for (int i = 0; i < XCH; ++i) {
+                if (!xact[i]) continue;
+                int j_local0 = (xj[i] - tile_j);
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    s16x4 w4 = {vx[i][0][e], vx[i][1][e], vx[i][2][e], vx[i][3][e]};
+                    *reinterpret_cast<s16x4 *>(
+                        xT_lds + swz<BKM>(j_local0 + e, xml[i] * 2)) = w4;
+                }
+            }
+            __syncthreads();
+            if (m0 + BKM < m_end) { load_dy(m0 + BKM); load_x(m0 + BKM); }
+
+#pragma unroll
+            for (int kk = 0; kk < BKM / 32; ++kk) {
+                int kbyte = kk * 64 + (lane >> 4) * 16;
+                s16x8 a[FC], b[FJ];
+#pragma unroll
+                for (int fc = 0; fc < FC; ++fc) {
+                    int row = wv_c * (BCOUT / CW) + fc * 16 + (lane & 15);
+                    a[fc] = *reinterpret_cast<const s16x8 *>(dyT_lds + swz<BKM>(row, kbyte));
+                }
+#pragma unroll
+                for (int fj = 0; fj < FJ; ++fj) {
+                    int row = wv_j * (BJ / 2) + fj * 16 + (lane & 15);
+                    b[fj] = *reinterpret_cast<const s16x8 *>(xT_lds + swz<BKM>(row, kbyte));
+                }
+#pragma unroll
+                for (int fc = 0; fc < FC; ++fc)
+#pragma unroll
+                    for (int fj = 0; fj < FJ; ++fj)
+                        acc[fc][fj] = MfmaOp<T>::mma(a[fc], b[fj], acc[fc][fj]);
+            }
+            __syncthreads();
+        }
+    } else {
     for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
         // ---- stage dy^T tile: 4 coalesced 16B reads dy[m..m+3][cout..+7],
         //      register transpose, 8 ds_write_b64 ----
@@ -555,6 +674,8 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                     acc[fc][fj] = MfmaOp<T>::mma(a[fc], b[fj], acc[fc][fj]);
         }
         __syncthreads();
+    }
+
     }
 
     // ---- store fp32 partial [z][Cout][Kw] ----

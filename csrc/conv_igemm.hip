@@ -460,6 +460,8 @@ void wgrad_kernel(const T *__restrict__ x, const T *__restrict__ dy,
     // only the 128x128 variant affords without spills/occupancy loss; the
     // other variants keep the original interleaved two-barrier body.
     constexpr bool PIPE = (BCOUT == 128 && BJ == 128 && NT == 256);
+    // (BCOUT==64 was A/B'd: wash — its 167->206 VGPR cost one wave of
+    //  occupancy, exactly cancelling the latency hiding)
 
     if constexpr (PIPE) {
         static_assert(BCOUT * (BKM / 4) / 8 == NT || !PIPE, "one dy chunk/thread");

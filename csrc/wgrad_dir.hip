@@ -120,61 +120,64 @@ void wgrad_dir_kernel(const T *__restrict__ x, const T *__restrict__ dy,
 
     __syncthreads();
 
-    for (int g = r0; g < r1; ++g) {
+    // thread -> (c8 group, pixel quad); this split keeps the b64
+    // transpose-write bank conflicts <= 2-way (header)
+    const int s_c8 = (t >> 5) & 7;
+    const int s_pq = (t & 31) + ((t >> 8) << 5);
+    const bool s_on = s_pq < n_pq;
+
+    // staged registers for the cross-row software pipeline: row g+1's
+    // global loads are issued before row g's MFMA so their latency
+    // overlaps the math (same lever as the im2col wgrad PIPE variant;
+    // +64 VGPRs stays within the 2-waves/SIMD budget)
+    s16x8 xv[3][4], dyv[4];
+
+    auto load_rows = [&](int g) {
         const int n = g / H;
         const int p = g % H;
-
-        // ---- issue global loads for the 4 rows (x p-1,p,p+1 and dy p) ----
-        s16x8 xv[3][4], dyv[4];
-        int my_pq = -1, my_c8 = -1;
-        {
-            // thread -> (c8 group, pixel quad); this split keeps the b64
-            // transpose-write bank conflicts <= 2-way (header)
-            int c8 = (t >> 5) & 7;
-            int pq = (t & 31) + ((t >> 8) << 5);
-            bool on = pq < n_pq;
-            if (on) { my_pq = pq; my_c8 = c8; }
 #pragma unroll
-            for (int d = 0; d < 3; ++d) {
-                int ih = p + d - 1;
-                bool rv = on && ih >= 0 && ih < H;
-#pragma unroll
-                for (int u = 0; u < 4; ++u) {
-                    int q = pq * 4 + u;
-                    xv[d][u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
-                    if (rv && q < W)
-                        xv[d][u] = *reinterpret_cast<const s16x8 *>(
-                            x + ((size_t)(n * H + ih) * W + q) * ldx +
-                            cc * 64 + c8 * 8);
-                }
-            }
+        for (int d = 0; d < 3; ++d) {
+            int ih = p + d - 1;
+            bool rv = s_on && ih >= 0 && ih < H;
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
-                int q = my_pq * 4 + u;
-                dyv[u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
-                if (on && q < W && cq * 64 + c8 * 8 < Cout)
-                    dyv[u] = *reinterpret_cast<const s16x8 *>(
-                        dy + ((size_t)(n * H + p) * W + q) * lddy +
-                        cq * 64 + c8 * 8);
-            }
-            // cout tail (Cout % 8 != 0 within the last vector): zero lanes
-            if (on && cq * 64 + c8 * 8 < Cout && cq * 64 + c8 * 8 + 8 > Cout) {
-#pragma unroll
-                for (int u = 0; u < 4; ++u)
-#pragma unroll
-                    for (int e = 0; e < 8; ++e)
-                        if (cq * 64 + c8 * 8 + e >= Cout) dyv[u][e] = 0;
+                int q = s_pq * 4 + u;
+                xv[d][u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                if (rv && q < W)
+                    xv[d][u] = *reinterpret_cast<const s16x8 *>(
+                        x + ((size_t)(n * H + ih) * W + q) * ldx +
+                        cc * 64 + s_c8 * 8);
             }
         }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+            int q = s_pq * 4 + u;
+            dyv[u] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+            if (s_on && q < W && cq * 64 + s_c8 * 8 < Cout)
+                dyv[u] = *reinterpret_cast<const s16x8 *>(
+                    dy + ((size_t)(n * H + p) * W + q) * lddy +
+                    cq * 64 + s_c8 * 8);
+        }
+        // cout tail (Cout % 8 != 0 within the last vector): zero lanes
+        if (s_on && cq * 64 + s_c8 * 8 < Cout && cq * 64 + s_c8 * 8 + 8 > Cout) {
+#pragma unroll
+            for (int u = 0; u < 4; ++u)
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    if (cq * 64 + s_c8 * 8 + e >= Cout) dyv[u][e] = 0;
+        }
+    };
 
+    if (r0 < r1) load_rows(r0);
+    for (int g = r0; g < r1; ++g) {
         __syncthreads();   // previous row's MFMA reads are done
 
         // ---- register transpose -> b64 LDS writes ([row][q] images) ----
-        if (my_pq >= 0) {
-            const int qb = (8 + my_pq * 4) * 2;   // q=0 lives at element 8
+        if (s_on) {
+            const int qb = (8 + s_pq * 4) * 2;   // q=0 lives at element 8
 #pragma unroll
             for (int d = 0; d < 3; ++d) {
-                char *base = xbuf[d] + (my_c8 * 8) * xpitch + qb;
+                char *base = xbuf[d] + (s_c8 * 8) * xpitch + qb;
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
                     s16x4 w4 = {xv[d][0][e], xv[d][1][e], xv[d][2][e],
@@ -182,7 +185,7 @@ void wgrad_dir_kernel(const T *__restrict__ x, const T *__restrict__ dy,
                     *reinterpret_cast<s16x4 *>(base + e * xpitch) = w4;
                 }
             }
-            char *base = dybuf + (my_c8 * 8) * dypitch + my_pq * 4 * 2;
+            char *base = dybuf + (s_c8 * 8) * dypitch + s_pq * 4 * 2;
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
                 s16x4 w4 = {dyv[0][e], dyv[1][e], dyv[2][e], dyv[3][e]};
@@ -191,6 +194,7 @@ void wgrad_dir_kernel(const T *__restrict__ x, const T *__restrict__ dy,
         }
 
         __syncthreads();
+        if (g + 1 < r1) load_rows(g + 1);
 
         // ---- MFMA sweep over the q k-tiles ----
         for (int kt = 0; kt < ktiles; ++kt) {

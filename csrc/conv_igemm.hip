@@ -939,10 +939,10 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
         if (NT == 512)   // 8-wave tiles: keep >=8 K-steps per split so the
             //              prologue + 256KB fp32 partial store amortize
             splitk = std::max(1, std::min<int>(splitk, (int)(M / BKM / 8)));
-        else             // 4-wave tiles: >=4 K-steps — at tiny-M shapes
-            //              (trans-conv wgrads, M ~4k) the fill-the-chip
-            //              split left ONE K-step per block, all prologue
-            splitk = std::max(1, std::min<int>(splitk, (int)(M / BKM / 4)));
+        // (a >=4-K-steps cap on the 4-wave tiles was TRIED and REVERTED:
+        //  at tiny-M shapes the lost block parallelism outweighed the
+        //  prologue amortization — trans-conv wgrad 43 -> 33 TF, bench
+        //  1217 -> 1200.  Do not repeat.)
         long m_per_split = ((M + splitk - 1) / splitk + BKM - 1) / BKM * BKM;
         splitk = (int)((M + m_per_split - 1) / m_per_split);
         auto partial = torch::empty({splitk, (long)Cout * Kw},

@@ -166,6 +166,19 @@ def test_hipgraph_step_matches_eager(tmp_path):
         assert torch.allclose(a.float(), b.float(), atol=1e-2, rtol=1e-2)
 
 
+def test_hipgraph_full_train_epoch(tmp_path):
+    """Graphs must compose with the real epoch loop: dataloader batches,
+    the eager TB-panel forward between replays, checkpointing and eval."""
+    s = _mk_solver("minet_r50_352", batch_size=2, size=96, val_samples=4,
+                   train_samples=8, epoch_num=2, log_interval=2,
+                   hip_graph=True, output_root=str(tmp_path),
+                   exp_name="graph_train")
+    s.train()
+    assert s._graph is not None, "graph never captured during train()"
+    res = s.test(save_png=False)
+    assert 0.0 <= res["mae"] <= 1.0
+
+
 def test_minet_step_with_8p_backend(tmp_path):
     """Full MINet-R50 step with the opt-in 8-phase conv backend enabled:
     eligible layers (Cout%256 fwd / Cin%256 dgrad) route to

@@ -144,6 +144,10 @@ def create_loader(cfg, split: str, distributed: bool, rank: int = 0,
         num_workers=cfg.num_workers,
         pin_memory=torch.cuda.is_available(),
         drop_last=(split == "train"),
-        persistent_workers=cfg.num_workers > 0,
+        # NOT persistent: live workers hold a pickled dataset copy, so the
+        # per-epoch set_epoch() (fresh augmentation space) would never
+        # reach them and every epoch would re-draw epoch-0 samples.
+        # Respawning workers each epoch pickles the post-set_epoch state.
+        persistent_workers=False,
     )
     return loader, sampler

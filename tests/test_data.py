@@ -41,6 +41,19 @@ def test_loader_distributed_sharding():
     assert set(i0).isdisjoint(i1)
 
 
+def test_epoch_reseed_reaches_loader_workers():
+    """set_epoch must change the SAMPLES the loader yields (persistent
+    workers would pin every epoch to the epoch-0 pickled dataset state)."""
+    cfg = make_config("cpu_plumbing", train_samples=2, batch_size=2,
+                      num_workers=2)
+    loader, _ = create_loader(cfg, "train", False)
+    loader.dataset.set_epoch(0)
+    a = next(iter(loader))[0].clone()
+    loader.dataset.set_epoch(1)
+    b = next(iter(loader))[0].clone()
+    assert not torch.equal(a, b), "epoch reseed never reached the workers"
+
+
 def test_val_loader_uneven_shards_cover_exactly_once():
     # 5 val images over 2 ranks: shards of 3 + 2, no padding/duplication
     cfg = make_config("cpu_plumbing", val_samples=5, batch_size=2, num_workers=0)

@@ -230,6 +230,30 @@ def test_sgd_fused_matches_reference():
                   what=f"sgd step {step}")
 
 
+def test_direct_kernels_on_channel_slices():
+    """Res2Net feeds channel-SLICE views (NHWC pixel pitch ld > C) into the
+    convs; the direct kernels must honor the pitch like the igemm family."""
+    from sodkit import ops
+    torch.manual_seed(13)
+    parent = _cl(torch.randn(2, 128, 176, 176, device="cuda").to(torch.bfloat16))
+    xs = parent[:, 32:96]           # 64-channel slice, ld = 128
+    w = (torch.randn(64, 3, 3, 64, device="cuda") * 0.1).to(torch.bfloat16)
+    xg = xs.detach().requires_grad_()
+    wg = w.clone().requires_grad_()
+    y = ops.conv2d(xg, wg, None, 1, 1, 1)   # dispatches conv_dir (W=176)
+    gy = torch.randn_like(y)
+    y.backward(gy)                          # dgrad_dir + wgrad_dir
+
+    xr = xs.float().detach().requires_grad_()
+    wr = w.float().permute(0, 3, 1, 2).detach().requires_grad_()
+    yr = F.conv2d(xr, wr, None, padding=1)
+    yr.backward(gy.float())
+    close(y, yr, what="slice fwd")
+    close(xg.grad, xr.grad, what="slice dgrad")
+    close(wg.grad, wr.grad.permute(0, 2, 3, 1), rtol=3e-2, atol=3e-2,
+          what="slice wgrad")
+
+
 WK1_SHAPES = [
     # (N, C, H, Cout, stride): 1x1 wgrad VALU kernel domain
     (2, 256, 28, 64, 1),     # bottleneck reduce

@@ -40,6 +40,14 @@ torch::Tensor conv2d_wgrad_dir(torch::Tensor x, torch::Tensor dy,
                                int64_t ldx, int64_t lddy);
 
 torch::Tensor bn_stats(torch::Tensor x);
+bool bn_fwd_fused_eligible(int64_t C, int64_t elem_size);
+std::vector<torch::Tensor> bn_fwd_fused(torch::Tensor x, torch::Tensor gamma,
+                                        torch::Tensor beta, double n,
+                                        double momentum, double eps,
+                                        torch::Tensor running_mean,
+                                        torch::Tensor running_var, bool relu,
+                                        torch::Tensor residual,
+                                        bool want_mask);
 std::vector<torch::Tensor> bn_finalize(torch::Tensor stats, double n,
                                        double momentum, double eps,
                                        torch::Tensor running_mean,
@@ -99,6 +107,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("wgrad_dir_eligible", &wgrad_dir_eligible);
     m.def("conv2d_wgrad_dir", &conv2d_wgrad_dir);
     m.def("bn_stats", &bn_stats);
+    m.def("bn_fwd_fused_eligible", &bn_fwd_fused_eligible);
+    m.def("bn_fwd_fused", &bn_fwd_fused);
     m.def("bn_finalize", &bn_finalize);
     m.def("bn_stats_finalize", &bn_stats_finalize);
     m.def("bn_apply", &bn_apply);

@@ -42,6 +42,7 @@ SRC = [
     "csrc/im2col.hip",
     "csrc/wgrad_k1.hip",
     "csrc/bnrelu.hip",
+    "csrc/bn_fused.hip",
     "csrc/bilinear.hip",
     "csrc/loss.hip",
     "csrc/sgd.hip",

@@ -25,6 +25,14 @@ from ._backend import ext
 
 _CL = torch.channels_last
 
+# single-launch cooperative BN forward (csrc/bn_fused.hip): stats +
+# finalize + apply in one kernel with grid.sync between phases — kills
+# two ~5us latency-floor launches per BN layer.  SODKIT_BN_FUSED=0
+# disables (A/B); SyncBN keeps the split path (the RCCL all-reduce sits
+# between stats and apply).
+import os
+_BNF_ENV = os.environ.get("SODKIT_BN_FUSED", "")
+
 
 def _cl(x: torch.Tensor) -> torch.Tensor:
     return x.contiguous(memory_format=_CL)
@@ -39,6 +47,7 @@ class _BNActFn(torch.autograd.Function):
             residual = _cl(residual)
         C = x.shape[1]
         n_local = x.numel() // C
+        fused = False
         if training:
             sync = (process_group is not None and dist.is_initialized()
                     and dist.get_world_size(process_group) > 1)
@@ -51,6 +60,15 @@ class _BNActFn(torch.autograd.Function):
                 mean, invstd = ext().bn_finalize(stats, n_total, momentum,
                                                  eps, running_mean,
                                                  running_var, True)
+            elif _BNF_ENV != "0" and ext().bn_fwd_fused_eligible(
+                    C, x.element_size()):
+                n_total = float(n_local)
+                y, mask, mean, invstd = ext().bn_fwd_fused(
+                    x, gamma, beta, n_total, momentum, eps, running_mean,
+                    running_var, relu,
+                    residual if residual is not None else torch.Tensor(),
+                    True)
+                fused = True
             else:
                 n_total = float(n_local)
                 mean, invstd = ext().bn_stats_finalize(x, n_total, momentum,
@@ -60,11 +78,12 @@ class _BNActFn(torch.autograd.Function):
             n_total = 0.0                                   # eval marker
             mean, invstd = ext().bn_finalize(running_mean, 0.0, momentum, eps,
                                              running_mean, running_var, False)
-        # training: also emit the 1-bit ReLU-mask plane so the backward
-        # kernels read 1/16 of y's bytes for the mask instead of y itself
-        y, mask = ext().bn_apply(
-            x, mean, invstd, gamma, beta, relu,
-            residual if residual is not None else torch.Tensor(), training)
+        if not fused:
+            # training: also emit the 1-bit ReLU-mask plane so the backward
+            # kernels read 1/16 of y's bytes for the mask instead of y itself
+            y, mask = ext().bn_apply(
+                x, mean, invstd, gamma, beta, relu,
+                residual if residual is not None else torch.Tensor(), training)
         ctx.save_for_backward(x, y, mean, invstd, gamma,
                               mask if mask.numel() else torch.Tensor())
         ctx.meta = (relu, residual is not None, n_total, process_group)

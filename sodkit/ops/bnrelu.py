@@ -25,11 +25,14 @@ from ._backend import ext
 
 _CL = torch.channels_last
 
-# single-launch cooperative BN forward (csrc/bn_fused.hip): stats +
-# finalize + apply in one kernel with grid.sync between phases — kills
-# two ~5us latency-floor launches per BN layer.  SODKIT_BN_FUSED=0
-# disables (A/B); SyncBN keeps the split path (the RCCL all-reduce sits
-# between stats and apply).
+# single-launch cooperative BN forward (csrc/bn_fused.hip) — NEGATIVE
+# RESULT, opt-in only (SODKIT_BN_FUSED=1).  The hypothesis was that
+# collapsing stats+finalize+apply into one grid.sync'd kernel kills two
+# ~5us launch floors per layer; measured: 810 vs 1213 imgs/s — the
+# cooperative launch + two grid-wide syncs per layer cost far more than
+# the floors they save (the co-resident grid is also smaller than the
+# tuned per-kernel grids).  Numerics-identical and fully tested; kept
+# as documentation.
 import os
 _BNF_ENV = os.environ.get("SODKIT_BN_FUSED", "")
 
@@ -60,7 +63,7 @@ class _BNActFn(torch.autograd.Function):
                 mean, invstd = ext().bn_finalize(stats, n_total, momentum,
                                                  eps, running_mean,
                                                  running_var, True)
-            elif _BNF_ENV != "0" and ext().bn_fwd_fused_eligible(
+            elif _BNF_ENV == "1" and ext().bn_fwd_fused_eligible(
                     C, x.element_size()):
                 n_total = float(n_local)
                 y, mask, mean, invstd = ext().bn_fwd_fused(

@@ -230,6 +230,39 @@ def test_sgd_fused_matches_reference():
                   what=f"sgd step {step}")
 
 
+@pytest.mark.parametrize("relu,res", [(True, True), (True, False),
+                                      (False, False)])
+def test_bn_fwd_fused_matches_split_path(relu, res):
+    """The opt-in cooperative fused BN forward (a measured perf NEGATIVE,
+    kept as documentation) must stay numerics-identical to the split
+    stats+finalize+apply path it mirrors."""
+    from sodkit import _C
+    torch.manual_seed(17)
+    N, C, H = 4, 64, 24
+    if not _C.bn_fwd_fused_eligible(C, 2):
+        pytest.skip("cooperative launch unsupported")
+    x = _cl(torch.randn(N, C, H, H, device="cuda").to(torch.bfloat16))
+    r = _cl(torch.randn(N, C, H, H, device="cuda").to(torch.bfloat16)) \
+        if res else torch.Tensor()
+    gamma = torch.rand(C, device="cuda") + 0.5
+    beta = torch.randn(C, device="cuda")
+    n = float(x.numel() // C)
+
+    rm1, rv1 = torch.zeros(C, device="cuda"), torch.ones(C, device="cuda")
+    yf, mf, meanf, invf = _C.bn_fwd_fused(x, gamma, beta, n, 0.1, 1e-5,
+                                          rm1, rv1, relu, r, True)
+
+    rm2, rv2 = torch.zeros(C, device="cuda"), torch.ones(C, device="cuda")
+    mean2, inv2 = _C.bn_stats_finalize(x, n, 0.1, 1e-5, rm2, rv2)
+    ys, ms = _C.bn_apply(x, mean2, inv2, gamma, beta, relu, r, True)
+
+    assert torch.equal(yf, ys), "fused y differs from split path"
+    assert torch.equal(meanf, mean2) and torch.equal(invf, inv2)
+    assert torch.equal(rm1, rm2) and torch.equal(rv1, rv2)
+    if relu:
+        assert torch.equal(mf, ms), "fused ReLU mask differs"
+
+
 def test_direct_kernels_on_channel_slices():
     """Res2Net feeds channel-SLICE views (NHWC pixel pitch ld > C) into the
     convs; the direct kernels must honor the pitch like the igemm family."""

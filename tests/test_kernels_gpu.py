@@ -256,11 +256,22 @@ def test_bn_fwd_fused_matches_split_path(relu, res):
     mean2, inv2 = _C.bn_stats_finalize(x, n, 0.1, 1e-5, rm2, rv2)
     ys, ms = _C.bn_apply(x, mean2, inv2, gamma, beta, relu, r, True)
 
-    assert torch.equal(yf, ys), "fused y differs from split path"
-    assert torch.equal(meanf, mean2) and torch.equal(invf, inv2)
-    assert torch.equal(rm1, rm2) and torch.equal(rv1, rv2)
+    # NOT bitwise: the fused grid size comes from the occupancy query, so
+    # the slab count (and thus the fp reduction grouping) differs from the
+    # split path's grid formula — equal math, different summation order
+    close(yf, ys, rtol=1e-3, atol=1e-3, what="fused vs split y")
+    close(meanf, mean2, rtol=1e-5, atol=1e-6, what="mean")
+    close(invf, inv2, rtol=1e-4, atol=1e-5, what="invstd")
+    close(rm1, rm2, rtol=1e-5, atol=1e-6, what="running mean")
+    close(rv1, rv2, rtol=1e-4, atol=1e-5, what="running var")
     if relu:
-        assert torch.equal(mf, ms), "fused ReLU mask differs"
+        # the mask must be self-consistent with the FUSED output's sign
+        bits = torch.zeros(yf.numel(), dtype=torch.bool, device="cuda")
+        mb = mf.to(torch.int32)
+        for e in range(8):
+            bits[e::8] = ((mb >> e) & 1).bool()
+        want = (yf.permute(0, 2, 3, 1).reshape(-1).float() > 0)
+        assert torch.equal(bits, want), "mask inconsistent with fused y"
 
 
 def test_direct_kernels_on_channel_slices():

@@ -25,13 +25,13 @@ _CL = torch.channels_last
 
 # 8-phase 256-square conv backend (csrc/conv_igemm8.hip, docs/
 # IGEMM8_PLAN.md).  Measured A/B across the model shape table
-# (gpurun_out/ubench_conv r2): the 256x256 tile only beats the 128-square
+# (profiles/r2/ubench_conv_final.log): the 256x256 tile only beats the 128-square
 # kernel when the launch fills the chip — its win shapes have
 # (ceil(M/256) * Cout/256) >= ~256 workgroups (e.g. 1x1 expand 64->256
 # @88x88: 143 vs 114 TF); below that it underfills 256 CUs and LOSES
 # (3x3 512ch @11x11: 110 vs 323 TF).  Dispatch is therefore gated on the
 # workgroup count; SODKIT_IGEMM_8P=1/0 forces it on/off for A/B runs.
-# Threshold from the full r2 table (gpurun_out/ubench_conv.log): every
+# Threshold from the full r2 table (profiles/r2/ubench_conv_final.log): every
 # measured shape at >= 244 wgs wins (l3_1x1e fwd +30%, l3_1x1r dgrad
 # +23%, l2_1x1e, l1_1x1e), every shape at <= 128 loses; 232 splits them.
 _8P_ENV = os.environ.get("SODKIT_IGEMM_8P", "")

@@ -71,7 +71,14 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
                      const float *__restrict__ bias, T *__restrict__ out,
                      const T *__restrict__ zero_page,
                      int N, int H, int W, int ldsrc, int ldout, int Cout,
-                     int rows_per_blk, int total_rows, int nqf) {
+                     int rows_per_blk, int total_rows, int nq_row) {
+    // q-fragment slots: 4 q-wave-groups x 3 slots = 12.  ROWS==1 assigns
+    // ceil(nq_row/4) slots per group (round-robin).  ROWS==2 POOLS the
+    // pair's 2*nq_row fragments across all 12 slots (row = fr / nq_row):
+    // at W=88 the per-row split left 6 frags on 8 slots (75% of the MFMA
+    // lanes doing work); the pool uses 12/12.
+    const int nqf = (ROWS == 2) ? 3 : (nq_row + 3) / 4;
+    const int pool = ROWS * nq_row;
     constexpr int XPITCH = Pitch<CC>::X;
     constexpr int WPITCH = Pitch<CC>::W;
     constexpr int WROW = 9 * CC * 2;     // weight row bytes
@@ -156,14 +163,12 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __syncthreads();
 
-        // ---- MFMA: D[cout][q] per output row over K = 576 ----
-        f32x4 acc[ROWS][2][3];
+        // ---- MFMA: D[cout][q] per output row over the K = 9*CC dim ----
+        f32x4 acc[2][3];
 #pragma unroll
-        for (int rw = 0; rw < ROWS; ++rw)
+        for (int a = 0; a < 2; ++a)
 #pragma unroll
-            for (int a = 0; a < 2; ++a)
-#pragma unroll
-                for (int b = 0; b < 3; ++b) acc[rw][a][b] = {0.f, 0.f, 0.f, 0.f};
+            for (int b = 0; b < 3; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
 
         const int oct16 = ((lane >> 4) << 4);       // k-octet byte
         const int arow = (wm * 32 + (lane & 15)) * WPITCH;
@@ -181,48 +186,48 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
 #pragma unroll
             for (int i = 0; i < 3; ++i) {
                 if (i >= nqf) break;
-                int q0 = (wq + i * 4) * 16;
+                const int fr = wq + i * 4;
+                if (ROWS == 2 && fr >= pool) continue;
+                const int rw = (ROWS == 2 && fr >= nq_row) ? 1 : 0;
+                int q0 = (fr - rw * nq_row) * 16;
                 // row q + s - 1, +1 halo base; rows beyond W+1 are zero pad
                 int row = q0 + (lane & 15) + ss;
                 const int bswz = (bblk ^ Pitch<CC>::swz(row)) << 4;
-#pragma unroll
-                for (int rw = 0; rw < ROWS; ++rw) {
-                    s16x8 b = *reinterpret_cast<const s16x8 *>(
-                        xbuf[rr + rw] + row * XPITCH + bswz);
-                    acc[rw][0][i] = MfmaOp<T>::mma(a0, b, acc[rw][0][i]);
-                    acc[rw][1][i] = MfmaOp<T>::mma(a1, b, acc[rw][1][i]);
-                }
+                s16x8 b = *reinterpret_cast<const s16x8 *>(
+                    xbuf[rr + rw] + row * XPITCH + bswz);
+                acc[0][i] = MfmaOp<T>::mma(a0, b, acc[0][i]);
+                acc[1][i] = MfmaOp<T>::mma(a1, b, acc[1][i]);
             }
         }
 
         // ---- store the output rows (NHWC, 4 couts per lane) ----
 #pragma unroll
-        for (int rw = 0; rw < ROWS; ++rw) {
-            const size_t obase = (size_t)(n * H + p + rw) * W;
+        for (int mf = 0; mf < 2; ++mf) {
+            int cout = wm * 32 + mf * 16 + ((lane >> 4) << 2);
+            if (cout >= Cout) continue;
+            float b4[4] = {0.f, 0.f, 0.f, 0.f};
+            if (bias != nullptr) {
 #pragma unroll
-            for (int mf = 0; mf < 2; ++mf) {
-                int cout = wm * 32 + mf * 16 + ((lane >> 4) << 2);
-                if (cout >= Cout) continue;
-                float b4[4] = {0.f, 0.f, 0.f, 0.f};
-                if (bias != nullptr) {
+                for (int e = 0; e < 4; ++e)
+                    if (cout + e < Cout) b4[e] = bias[cout + e];
+            }
 #pragma unroll
-                    for (int e = 0; e < 4; ++e)
-                        if (cout + e < Cout) b4[e] = bias[cout + e];
+            for (int i = 0; i < 3; ++i) {
+                if (i >= nqf) break;
+                const int fr = wq + i * 4;
+                if (ROWS == 2 && fr >= pool) continue;
+                const int rw = (ROWS == 2 && fr >= nq_row) ? 1 : 0;
+                int q = (fr - rw * nq_row) * 16 + (lane & 15);
+                if (q >= W) continue;
+                const size_t obase = (size_t)(n * H + p + rw) * W;
+                s16x4_ v;
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    T tv = ElemTraits<T>::from_f32(acc[mf][i][e] + b4[e]);
+                    v[e] = *reinterpret_cast<const short *>(&tv);
                 }
-#pragma unroll
-                for (int i = 0; i < 3; ++i) {
-                    if (i >= nqf) break;
-                    int q = (wq + i * 4) * 16 + (lane & 15);
-                    if (q >= W) continue;
-                    s16x4_ v;
-#pragma unroll
-                    for (int e = 0; e < 4; ++e) {
-                        T tv = ElemTraits<T>::from_f32(acc[rw][mf][i][e] + b4[e]);
-                        v[e] = *reinterpret_cast<const short *>(&tv);
-                    }
-                    *reinterpret_cast<s16x4_ *>(
-                        out + (obase + q) * ldout + cout) = v;
-                }
+                *reinterpret_cast<s16x4_ *>(
+                    out + (obase + q) * ldout + cout) = v;
             }
         }
     }
@@ -268,8 +273,7 @@ void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
     int blocks = std::min(512, total_rows / ROWS);
     int rows_per_blk = (total_rows / ROWS + blocks - 1) / blocks * ROWS;
     blocks = (total_rows + rows_per_blk - 1) / rows_per_blk;
-    const int nqf_total = (W + 15) / 16;
-    const int nqf = (nqf_total + 3) / 4;     // frags per q-group (<= 3)
+    const int nq_row = (W + 15) / 16;        // q fragments per output row
     const size_t shmem = 64 * (size_t)sodkit::cdir::Pitch<CC>::W
                          + (ROWS + 2) * (size_t)(W + 18) * sodkit::cdir::Pitch<CC>::X;
     static bool attr = false;
@@ -286,7 +290,7 @@ void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
                        reinterpret_cast<T *>(out.data_ptr()),
                        reinterpret_cast<const T *>(cdir_zero_page(src)),
                        N, H, W, ldsrc, ldout, Cout, rows_per_blk, total_rows,
-                       nqf);
+                       nq_row);
 }
 
 template <typename T, int MODE>

@@ -84,10 +84,10 @@ def _use_wgrad_k1(w: torch.Tensor, padding, dilation) -> bool:
 
 # direct fwd/dgrad for the 64-channel 3x3 decoder/AIM/SIM class
 # (csrc/conv_dir.hip): the im2col kernel re-stages every x element 9x.
-# Measured (r2 ubench fdir/ddir): wins at W=176 (fwd 478 vs 404 TF,
-# dgrad 482 vs 371) but LOSES at W=88 (293 vs 364 — the ragged q-tail
-# wastes 1/4 of the MFMA lanes and per-row overheads bite), so the gate
-# is W >= 128.  SODKIT_CONV_DIR=0 disables (A/B).
+# Measured (r2): W=176 fwd/dgrad 517/522 vs igemm 436/400; W=88 went
+# from a loss (293 vs 364) to a WIN (377/378 vs 366/356) once the
+# ROWS=2 path pooled the row-pair's q-fragments across all wave slots,
+# so the gate is W >= 64.  SODKIT_CONV_DIR=0 disables (A/B).
 _CDIR_ENV = os.environ.get("SODKIT_CONV_DIR", "")
 
 
@@ -97,7 +97,7 @@ def _use_conv_dir(cin: int, cout: int, w3: bool, x: torch.Tensor, stride,
         return False
     return (w3 and stride == 1 and padding == 1 and dilation == 1
             and cin in (32, 64) and cout <= 64 and cout % 4 == 0
-            and 128 <= x.shape[3] <= 184)
+            and 64 <= x.shape[3] <= 184)
 
 
 def _cl_ld(x: torch.Tensor):

@@ -54,8 +54,9 @@ def test_conv_dir_gate():
     assert c._use_conv_dir(64, 64, True, _x(176), 1, 1, 1)
     assert c._use_conv_dir(32, 64, True, _x(176), 1, 1, 1)
     assert c._use_conv_dir(64, 32, True, _x(128), 1, 1, 1)
-    # measured loss at W=88 (354 vs igemm 365) and ineligible forms
-    assert not c._use_conv_dir(64, 64, True, _x(88), 1, 1, 1)
+    assert c._use_conv_dir(64, 64, True, _x(88), 1, 1, 1)  # pooled ROWS=2 win
+    # ineligible forms / measured losses
+    assert not c._use_conv_dir(64, 64, True, _x(44), 1, 1, 1)
     assert not c._use_conv_dir(128, 64, True, _x(176), 1, 1, 1)
     assert not c._use_conv_dir(64, 128, True, _x(176), 1, 1, 1)
     assert not c._use_conv_dir(64, 64, False, _x(176), 1, 0, 1)   # not 3x3

@@ -38,6 +38,10 @@ bool wgrad_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
                         int64_t dil, int64_t C, int64_t W);
 torch::Tensor conv2d_wgrad_dir(torch::Tensor x, torch::Tensor dy,
                                int64_t ldx, int64_t lddy);
+// batched grad->bucket gather for deferred-grad mode (csrc/multicopy.hip)
+int64_t multicopy_chunk_bytes();
+void multicopy(torch::Tensor src, torch::Tensor dst, torch::Tensor nbytes,
+               torch::Tensor segmap, torch::Tensor chunkmap);
 
 torch::Tensor bn_stats(torch::Tensor x);
 bool bn_fwd_fused_eligible(int64_t C, int64_t elem_size);
@@ -121,6 +125,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dynfilter_fwd", &dynfilter_fwd);
     m.def("dynfilter_bwd", &dynfilter_bwd);
     m.def("sgd_fused", &sgd_fused);
+    m.def("multicopy_chunk_bytes", &multicopy_chunk_bytes);
+    m.def("multicopy", &multicopy);
     // build metadata (diagnostics: confirms the in-tree gfx950 build is
     // the one loaded, e.g. in the driver's native-code check)
     m.attr("BUILD_ARCH") = "gfx950";

@@ -46,6 +46,7 @@ SRC = [
     "csrc/bilinear.hip",
     "csrc/loss.hip",
     "csrc/sgd.hip",
+    "csrc/multicopy.hip",
     "csrc/dynfilter.hip",
 ]
 

@@ -84,6 +84,9 @@ class Solver:
         self.optimizer = ops.SodSGD(groups, momentum=cfg.momentum,
                                     weight_decay=cfg.weight_decay,
                                     nesterov=cfg.nesterov)
+        # stable grad source for the fused-SGD pointer tables (deferred-grad
+        # mode re-lands autograd's per-step grad tensors in these views)
+        self.optimizer.grad_src = self.reducer.grad_views
 
         self.criterion = build_loss(cfg)
         self.loss_scale = cfg.loss_scale if cfg.dtype == "fp16" else 1.0

@@ -54,6 +54,12 @@ class SodSGD:
                     "momentum_buffer": torch.zeros_like(p, dtype=torch.float32),
                 }
         self._tables = None  # lazily built device pointer tables (GPU)
+        # deferred-grad mode (reducer.defer): .grad is a fresh autograd
+        # tensor each step, so the trainer points grad_src at the reducer's
+        # STABLE bucket views — the gather in finalize() lands grads there
+        # and the baked pointer tables stay valid.  In view mode the views
+        # ARE .grad, so wiring it unconditionally is a no-op.
+        self.grad_src: "dict[torch.nn.Parameter, torch.Tensor] | None" = None
 
     # -- fused GPU tables (one table + one launch per param dtype: conv
     #    weights are bf16/fp16 while BN affine params stay fp32) -----------
@@ -62,10 +68,14 @@ class SodSGD:
         for g in self.groups:
             wd = g["weight_decay"]
             for p in g["params"]:
-                assert p.grad is not None, "grads must be pre-assigned (reducer bucket views)"
+                g = (self.grad_src.get(p) if self.grad_src is not None
+                     else None)
+                if g is None:
+                    g = p.grad
+                assert g is not None, "grads must be pre-assigned (reducer bucket views)"
                 st = self.state[p]
                 groups_by_dtype.setdefault(p.dtype, []).append(
-                    ((p.data_ptr(), p.grad.data_ptr(), st["master"].data_ptr(),
+                    ((p.data_ptr(), g.data_ptr(), st["master"].data_ptr(),
                       st["momentum_buffer"].data_ptr()), p.numel(), wd, p.device))
         self._tables = []
         codes = {torch.float32: 0, torch.bfloat16: 1, torch.float16: 2}

@@ -91,6 +91,7 @@ class GradReducer:
 
         # --- flat storage + grad views ---
         self._p2bucket: dict[torch.nn.Parameter, _Bucket] = {}
+        self.grad_views: dict[torch.nn.Parameter, torch.Tensor] = {}
         for b in self.buckets:
             total = sum(p.numel() for p in b.params)
             dev = b.params[0].device
@@ -99,8 +100,28 @@ class GradReducer:
             off = 0
             for p in b.params:
                 p.grad = b.flat[off:off + p.numel()].view_as(p)
+                self.grad_views[p] = p.grad
                 off += p.numel()
                 self._p2bucket[p] = b
+
+        # --- deferred-grad mode (single-GPU): AccumulateGrad's per-param
+        # add_ into the bucket views is ~280 launches of <=512 threads per
+        # step at the ~4.5us launch floor (profiles/r2/
+        # kernel_summary_close.txt, ~3.5% of the step).  At world==1 there
+        # is no overlap to preserve, so instead: zero_grad() sets .grad=None
+        # (autograd then ASSIGNS fresh grad tensors — zero kernels) and
+        # finalize() lands them in the flats with ONE multicopy kernel.
+        # Bit-identical to the view path (assignment == add-into-zero; the
+        # gather is an exact copy).  Accumulation-safe: micro-steps 2..k add
+        # into the live .grad and the gather copies the accumulated value.
+        # The fused SGD keeps reading the stable bucket views (grad_views).
+        # SODKIT_DEFER_GRADS=0 disables (A/B).
+        self.defer = (_world(self.pg) == 1 and len(params) > 0
+                      and params[0].is_cuda
+                      and os.environ.get("SODKIT_DEFER_GRADS", "1") == "1")
+        self._mc = None          # lazily built static copy tables
+        self._ptr_cache = None   # last step's grad pointers (skip H2D)
+        self._cache_captured = False
 
         self.comm_stream = (torch.cuda.Stream()
                             if self.overlap and _world(self.pg) > 1 else None)
@@ -177,11 +198,86 @@ class GradReducer:
                     f"not plausible for a {w}-way sum")
 
     # ------------------------------------------------------------------
+    def _defer_gather(self) -> None:
+        """One-kernel gather of every autograd-assigned .grad into its
+        bucket slot (deferred-grad mode).  Segment sizes and destinations
+        are static (bucket layout); only the source pointers change —
+        and with the caching allocator they are step-stable, so the tiny
+        H2D pointer refresh is skipped in steady state.  hipGraph-safe:
+        under capture the refresh goes through a pinned staging buffer
+        (async H2D only), and replays re-run the baked copy+kernel against
+        the capture pool's stable addresses."""
+        from ..ops._backend import ext
+
+        if self._mc is None:
+            chunk = int(ext().multicopy_chunk_bytes())
+            plist: list[torch.nn.Parameter] = []
+            dst_ptrs: list[int] = []
+            sizes: list[int] = []
+            for b in self.buckets:
+                esz = b.params[0].element_size()
+                base = b.flat.data_ptr()
+                off = 0
+                for p in b.params:
+                    plist.append(p)
+                    dst_ptrs.append(base + off * esz)
+                    sizes.append(p.numel() * esz)
+                    off += p.numel()
+            dev = self.buckets[0].flat.device
+            segmap, chunkmap = [], []
+            for i, nb in enumerate(sizes):
+                for c in range((nb + chunk - 1) // chunk):
+                    segmap.append(i)
+                    chunkmap.append(c)
+            self._mc = (
+                plist, dst_ptrs,
+                torch.tensor(dst_ptrs, dtype=torch.int64, device=dev),
+                torch.tensor(sizes, dtype=torch.int64, device=dev),
+                torch.tensor(segmap, dtype=torch.int32, device=dev),
+                torch.tensor(chunkmap, dtype=torch.int32, device=dev),
+                torch.empty(len(plist), dtype=torch.int64, device=dev),
+                torch.empty(len(plist), dtype=torch.int64, pin_memory=True),
+            )
+        (plist, dst_ptrs, dst_dev, nb_dev, segmap, chunkmap, src_dev,
+         src_pin) = self._mc
+        ptrs = []
+        for i, p in enumerate(plist):
+            g = p.grad
+            if g is not None and g.is_contiguous() and g.dtype == p.dtype:
+                ptrs.append(g.data_ptr())
+            else:
+                v = self.grad_views[p]
+                if g is None:
+                    v.detach().zero_()     # param got no grad this step
+                else:                      # odd layout: land it directly
+                    v.detach().copy_(g.reshape(v.shape).to(p.dtype))
+                ptrs.append(dst_ptrs[i])   # self-copy no-op in the kernel
+        capturing = torch.cuda.is_current_stream_capturing()
+        if ptrs != self._ptr_cache or (self._cache_captured and not capturing):
+            # second clause: a FAILED capture may have cached pointers whose
+            # device-side refresh never executed — eager must re-copy
+            self._ptr_cache = ptrs
+            self._cache_captured = capturing
+            t = torch.tensor(ptrs, dtype=torch.int64)
+            if capturing:
+                src_pin.copy_(t)
+                src_dev.copy_(src_pin, non_blocking=True)
+            else:
+                src_dev.copy_(t)  # blocking: stream-ordered after last gather
+        ext().multicopy(src_dev, dst_dev, nb_dev, segmap, chunkmap)
+
+    # ------------------------------------------------------------------
     def finalize(self) -> None:
         """Call after the last (sync) backward of the step: wait for every
         in-flight collective (and launch any straggler buckets — a bucket
         whose params got no grad this backward has arrived != 0 ...== 0 via
         the cycling counter only when it completed)."""
+        if self.defer:
+            self._defer_gather()
+            for b in self.buckets:
+                b.arrived = 0
+                b.launched = False
+            return
         for b in self.buckets:
             if not b.launched and _world(self.pg) > 1 and self._sync:
                 # straggler: either some param got no grad this step
@@ -204,6 +300,15 @@ class GradReducer:
             b.launched = False
 
     def zero_grad(self) -> None:
+        if self.defer:
+            # no memsets: .grad=None makes autograd ASSIGN the next grads
+            # (zero kernels) and the finalize gather overwrites every slot
+            for p in self._params:
+                p.grad = None
+            for b in self.buckets:
+                b.arrived = 0
+                b.launched = False
+            return
         for b in self.buckets:
             b.flat.zero_()
             b.arrived = 0

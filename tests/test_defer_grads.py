@@ -1,0 +1,105 @@
+"""Deferred-grad mode (sodkit/parallel/reducer.py): at world==1 on GPU,
+zero_grad() sets .grad=None (autograd assigns fresh grads — no per-param
+add_ launches) and finalize() lands them in the bucket flats with one
+multicopy kernel.  Must be BIT-identical to the grad-as-view path:
+assignment == add-into-zero, and the gather is an exact byte copy."""
+
+import copy
+import os
+
+import pytest
+import torch
+
+from sodkit.parallel.reducer import GradReducer
+
+
+def _model(device="cpu"):
+    torch.manual_seed(7)
+    m = torch.nn.Sequential(
+        torch.nn.Conv2d(3, 8, 3, padding=1),
+        torch.nn.BatchNorm2d(8),
+        torch.nn.ReLU(),
+        torch.nn.Conv2d(8, 4, 1),
+    )
+    return m.to(device)
+
+
+def test_defer_off_on_cpu():
+    r = GradReducer(_model())
+    assert not r.defer  # CPU: gather kernel is GPU-only; views stay
+
+
+def test_defer_env_kill_switch(monkeypatch):
+    monkeypatch.setenv("SODKIT_DEFER_GRADS", "0")
+    r = GradReducer(_model())
+    assert not r.defer
+
+
+def _flats_after(model, x, defer: bool, steps: int = 2, accum: int = 1):
+    prev = os.environ.get("SODKIT_DEFER_GRADS")
+    os.environ["SODKIT_DEFER_GRADS"] = "1" if defer else "0"
+    try:
+        r = GradReducer(model)
+    finally:
+        if prev is None:
+            os.environ.pop("SODKIT_DEFER_GRADS", None)
+        else:
+            os.environ["SODKIT_DEFER_GRADS"] = prev
+    assert r.defer == (defer and x.is_cuda)
+    for _ in range(steps):
+        r.zero_grad()
+        for _ in range(accum):   # micro-batches accumulate into .grad
+            model(x).square().mean().backward()
+        r.finalize()
+    return [b.flat.clone() for b in r.buckets]
+
+
+@pytest.mark.gpu
+def test_defer_bitwise_vs_views():
+    m1 = _model("cuda")
+    m2 = copy.deepcopy(m1)
+    x = torch.randn(2, 3, 16, 16, device="cuda")
+    ref = _flats_after(m1, x, defer=False)
+    got = _flats_after(m2, x, defer=True)
+    assert len(ref) == len(got)
+    for a, b in zip(ref, got):
+        assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+def test_defer_bitwise_with_accumulation():
+    m1 = _model("cuda")
+    m2 = copy.deepcopy(m1)
+    x = torch.randn(2, 3, 16, 16, device="cuda")
+    ref = _flats_after(m1, x, defer=False, steps=1, accum=3)
+    got = _flats_after(m2, x, defer=True, steps=1, accum=3)
+    for a, b in zip(ref, got):
+        assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+def test_defer_missing_grad_zeroes_slot():
+    m = _model("cuda")
+    os.environ["SODKIT_DEFER_GRADS"] = "1"
+    try:
+        r = GradReducer(m)
+    finally:
+        os.environ.pop("SODKIT_DEFER_GRADS", None)
+    assert r.defer
+    x = torch.randn(2, 3, 16, 16, device="cuda")
+    r.zero_grad()
+    m(x).square().mean().backward()
+    r.finalize()
+    # second step: only the tail conv gets grads -> every other slot must
+    # read ZERO (stale grads applied by SGD would corrupt training)
+    r.zero_grad()
+    feat = m[0](x).detach().requires_grad_(True)
+    m[3](m[2](m[1](feat))).square().mean().backward()
+    r.finalize()
+    zeroed = total = 0
+    for p in m[0].parameters():
+        total += 1
+        if torch.equal(r.grad_views[p].detach(),
+                       torch.zeros_like(r.grad_views[p])):
+            zeroed += 1
+    assert zeroed == total

@@ -10,6 +10,16 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t R,
                            int64_t S, int64_t stride, int64_t pad, int64_t dil,
                            int64_t ldx, int64_t lddy);
 torch::Tensor channel_sum(torch::Tensor x, int64_t ld);
+// fwd + fused BN stat partials (csrc/conv_igemm.hip ES epilogue)
+std::vector<torch::Tensor> conv2d_fwd_stats(torch::Tensor x, torch::Tensor w,
+                                            torch::Tensor bias,
+                                            int64_t stride, int64_t pad,
+                                            int64_t dil, int64_t ld);
+std::vector<torch::Tensor> bn_finalize_slabs(torch::Tensor partial, double n,
+                                             double momentum, double eps,
+                                             torch::Tensor running_mean,
+                                             torch::Tensor running_var);
+torch::Tensor bn_slab_sum(torch::Tensor partial);
 // 8-phase opt-in conv backend (SODKIT_IGEMM_8P=1; docs/IGEMM8_PLAN.md)
 bool conv8_eligible(int64_t cout, int64_t cin);
 torch::Tensor conv2d_fwd_8p(torch::Tensor x, torch::Tensor w,
@@ -111,6 +121,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("wgrad_dir_eligible", &wgrad_dir_eligible);
     m.def("conv2d_wgrad_dir", &conv2d_wgrad_dir);
     m.def("bn_stats", &bn_stats);
+    m.def("conv2d_fwd_stats", &conv2d_fwd_stats);
+    m.def("bn_finalize_slabs", &bn_finalize_slabs);
+    m.def("bn_slab_sum", &bn_slab_sum);
     m.def("bn_fwd_fused_eligible", &bn_fwd_fused_eligible);
     m.def("bn_fwd_fused", &bn_fwd_fused);
     m.def("bn_finalize", &bn_finalize);

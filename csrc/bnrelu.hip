@@ -644,6 +644,48 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x, double n,
     return {mean, invstd};
 }
 
+// finalize conv-emitted stat slabs (conv2d_fwd_stats): partial[nb][2C]
+// has the same layout as bn_stats_vec's output, so this is exactly the
+// second half of bn_stats_finalize — the stats pass never touches y.
+std::vector<torch::Tensor> bn_finalize_slabs(torch::Tensor partial, double n,
+                                             double momentum, double eps,
+                                             torch::Tensor running_mean,
+                                             torch::Tensor running_var) {
+    TORCH_CHECK(partial.is_cuda() && partial.dim() == 2
+                && partial.scalar_type() == at::kFloat);
+    int nb = partial.size(0);
+    int C = (int)(partial.size(1) / 2);
+    TORCH_CHECK(C == running_mean.size(0), "slab/param channel mismatch");
+    auto mean = torch::empty({C}, running_mean.options());
+    auto invstd = torch::empty({C}, running_mean.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL((sodkit::bn_reduce_finalize_kernel<16, 16>),
+                       dim3((C + 15) / 16), dim3(16, 16), 0, stream.stream(),
+                       partial.data_ptr<float>(), nb, C, (float)n,
+                       (float)momentum, (float)eps,
+                       running_mean.data_ptr<float>(),
+                       running_var.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>());
+    return {mean, invstd};
+}
+
+// reduce conv-emitted stat slabs to the coalesced [2, C] stats tensor
+// (SyncBN: this is what gets all-reduced before bn_finalize)
+torch::Tensor bn_slab_sum(torch::Tensor partial) {
+    TORCH_CHECK(partial.is_cuda() && partial.dim() == 2
+                && partial.scalar_type() == at::kFloat);
+    int nb = partial.size(0);
+    long R = partial.size(1);
+    auto out = torch::empty({2, R / 2},
+                            partial.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL((sodkit::slab_reduce_kernel<16, 16>),
+                       dim3((R + 15) / 16), dim3(16, 16), 0, stream.stream(),
+                       partial.data_ptr<float>(), out.data_ptr<float>(), nb,
+                       R);
+    return out;
+}
+
 std::vector<torch::Tensor> bn_apply(torch::Tensor x, torch::Tensor mean,
                                     torch::Tensor invstd, torch::Tensor gamma,
                                     torch::Tensor beta, bool relu,

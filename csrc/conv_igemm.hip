@@ -98,11 +98,18 @@ SOD_DEVINL void glds16(const void *gsrc, char *lds_dst) {
 
 SOD_DEVINL int swz_nib(int row) { return ((row & 7) ^ ((row >> 3) & 7)) << 4; }
 
-template <typename T, int MODE, int BCOUT, int BM>
+// ES=1 (fwd only): the epilogue also emits per-channel (sum, sumsq) BN
+// partials over the ROUNDED output values into stats_out[tiles_m][2*Ncout]
+// — the exact input format of bn_reduce_finalize_kernel — so a following
+// BatchNorm never re-reads y for its stats pass (bn_stats_vec was ~1.2
+// ms/step of pure y re-read traffic).  Cost in the conv: ~8 live VGPRs,
+// a 16-lane shuffle tree per fragment and a 2 KB static-LDS combine.
+template <typename T, int MODE, int BCOUT, int BM, int ES = 0>
 __global__ __launch_bounds__(256, 3)   // force >= 3 waves/SIMD (171-reg
                                        // natural allocation lands at 2)
 void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                   const float *__restrict__ bias, T *__restrict__ out,
+                  float *__restrict__ stats_out,
                   const T *__restrict__ zero_page,
                   int Ns, int Hs, int Ws, int Cs,   // gather-source dims
                   int ld_src,                       // pixel pitch of B_src
@@ -336,6 +343,12 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     }
 
     // ---- epilogue: D[cout][m] -> NHWC out[m][cout], 8B packed stores ----
+    // ES: static-LDS combine plane for the two wv_m wave-halves that share
+    // each channel quad (discarded entirely for ES=0 instantiations)
+    struct ESRed { float v[2][2][BCOUT]; };  // [wv_m][sum|ssq][c_local]
+    __shared__ char es_red_raw[ES ? sizeof(ESRed) : 1];
+    ESRed *es_red = reinterpret_cast<ESRed *>(es_red_raw);
+    if constexpr (ES) __syncthreads();  // LDS tiles done before reuse
 #pragma unroll
     for (int fc = 0; fc < FC; ++fc) {
         int cout = tile_c + wv_c * (BCOUT / 2) + fc * 16 + (lane >> 4) * 4;
@@ -345,6 +358,8 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
             for (int e = 0; e < 4; ++e)
                 if (cout + e < Ncout) b4[e] = bias[cout + e];
         }
+        float ssum[4] = {0.f, 0.f, 0.f, 0.f};
+        float sssq[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int fm = 0; fm < FM; ++fm) {
             int m = tile_m + wv_m * (BM / 2) + fm * 16 + (lane & 15);
@@ -353,6 +368,18 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
             if (bias != nullptr) {
 #pragma unroll
                 for (int e = 0; e < 4; ++e) v[e] += b4[e];
+            }
+            if constexpr (ES) {
+                // stats over the ROUNDED stored values: identical numbers
+                // to what bn_stats_vec would re-read from y
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    T te = ElemTraits<T>::from_f32(v[e]);
+                    float f = s16_as_f32<T>(
+                        *reinterpret_cast<const short *>(&te));
+                    ssum[e] += f;
+                    sssq[e] += f * f;
+                }
             }
             T *dst = out + (long)m * Ncout + cout;
             // vector store needs 8B alignment: holds iff Ncout % 4 == 0
@@ -364,6 +391,36 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
                     if (cout + e < Ncout)
                         dst[e] = ElemTraits<T>::from_f32(v[e]);
             }
+        }
+        if constexpr (ES) {
+            // fold the 16 m-lanes (lane bits 0..3; channel bits untouched)
+#pragma unroll
+            for (int d = 1; d < 16; d <<= 1)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    ssum[e] += __shfl_xor(ssum[e], d, 64);
+                    sssq[e] += __shfl_xor(sssq[e], d, 64);
+                }
+            if ((lane & 15) == 0) {
+                int cl = wv_c * (BCOUT / 2) + fc * 16 + (lane >> 4) * 4;
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    es_red->v[wv_m][0][cl + e] = ssum[e];
+                    es_red->v[wv_m][1][cl + e] = sssq[e];
+                }
+            }
+        }
+    }
+    if constexpr (ES) {
+        __syncthreads();
+        // one slab row per block: stats_out[m_tile][{sum,ssq} x Ncout]
+        int mtile = tile_m / BM;
+        long base = (long)mtile * 2 * Ncout;
+        for (int i = t; i < 2 * BCOUT; i += 256) {
+            int row = i / BCOUT, c = i % BCOUT;
+            if (tile_c + c < Ncout)
+                stats_out[base + (long)row * Ncout + tile_c + c] =
+                    es_red->v[0][row][c] + es_red->v[1][row][c];
         }
     }
 }
@@ -800,7 +857,8 @@ void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
                   int Ns, int Hs, int Ws, int Cs, int ld_src,
                   int P, int Q, int Ncout,
                   int R, int S, int stride, int pad, int dil,
-                  long M_total, int K_total) {
+                  long M_total, int K_total,
+                  torch::Tensor *stats = nullptr) {
     using T = scalar_t;
     auto stream = at::hip::getCurrentHIPStream();
     const float *bias_ptr = bias.defined() && bias.numel()
@@ -824,14 +882,32 @@ void launch_igemm(const torch::Tensor &src, const torch::Tensor &w,
         }();
         if (force_sb) dbuf = false;
         size_t shmem = (size_t)(dbuf ? 2 : 1) * (BC + BMv) * 64 * 2;
-        hipLaunchKernelGGL((sodkit::igemm_kernel<T, MODE, BC, BMv>), grid,
-                           dim3(256), shmem, stream.stream(),
-                           reinterpret_cast<const T *>(src.data_ptr()),
-                           reinterpret_cast<const T *>(w.data_ptr()), bias_ptr,
-                           reinterpret_cast<T *>(out.data_ptr()), zpage,
-                           Ns, Hs, Ws, Cs, ld_src, P, Q, Ncout, R, S,
-                           stride, pad, dil,
-                           (int)M_total, K_total, tiles_m, (int)dbuf);
+        float *stats_ptr = nullptr;
+        if (stats != nullptr) {
+            *stats = torch::empty({tiles_m, 2L * Ncout},
+                                  src.options().dtype(torch::kFloat32));
+            stats_ptr = stats->data_ptr<float>();
+        }
+        auto kern = [&](auto es_tag) {
+            constexpr int ESv = decltype(es_tag)::value;
+            hipLaunchKernelGGL((sodkit::igemm_kernel<T, MODE, BC, BMv, ESv>),
+                               grid, dim3(256), shmem, stream.stream(),
+                               reinterpret_cast<const T *>(src.data_ptr()),
+                               reinterpret_cast<const T *>(w.data_ptr()),
+                               bias_ptr,
+                               reinterpret_cast<T *>(out.data_ptr()),
+                               stats_ptr, zpage,
+                               Ns, Hs, Ws, Cs, ld_src, P, Q, Ncout, R, S,
+                               stride, pad, dil,
+                               (int)M_total, K_total, tiles_m, (int)dbuf);
+        };
+        if constexpr (MODE == 0) {
+            if (stats_ptr != nullptr) {
+                kern(std::integral_constant<int, 1>{});
+                return;
+            }
+        }
+        kern(std::integral_constant<int, 0>{});
     };
     using c64 = std::integral_constant<int, 64>;
     using c128 = std::integral_constant<int, 128>;
@@ -878,6 +954,43 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                                     P, Q, Cout, R, S, stride, pad, dil, M, K);
         }));
     return out;
+}
+
+// fwd + fused BN stats partials: returns {y, slabs[tiles_m, 2*Cout]} —
+// slabs feed bn_finalize_slabs / bn_slab_sum (csrc/bnrelu.hip) so the BN
+// stats pass never re-reads y.
+std::vector<torch::Tensor> conv2d_fwd_stats(torch::Tensor x, torch::Tensor w,
+                                            torch::Tensor bias,
+                                            int64_t stride, int64_t pad,
+                                            int64_t dil, int64_t ld) {
+    CHECK_GPU(x); CHECK_GPU(w);
+    TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
+    int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    TORCH_CHECK(ld >= C, "bad channel pitch");
+    int Cout = w.size(0), R = w.size(1), S = w.size(2);
+    TORCH_CHECK(w.size(3) == C, "weight Cin mismatch");
+    TORCH_CHECK(w.is_contiguous());
+    int P = out_dim(H, R, stride, pad, dil);
+    int Q = out_dim(W_, S, stride, pad, dil);
+    long M = (long)N * P * Q;
+    int K = R * S * C;
+    auto out = torch::empty({N, Cout, P, Q},
+                            x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    torch::Tensor bias32;
+    if (bias.defined() && bias.numel()) bias32 = bias.to(torch::kFloat32);
+    torch::Tensor slabs;
+    AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_fwd_stats",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            launch_igemm<__hip_bfloat16, 0>(x, w, bias32, out, N, H, W_, C,
+                                            (int)ld, P, Q, Cout, R, S,
+                                            stride, pad, dil, M, K, &slabs);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            launch_igemm<__half, 0>(x, w, bias32, out, N, H, W_, C, (int)ld,
+                                    P, Q, Cout, R, S, stride, pad, dil, M, K,
+                                    &slabs);
+        }));
+    return {out, slabs};
 }
 
 torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w_t, int64_t H,

@@ -44,7 +44,7 @@ def _cl(x: torch.Tensor) -> torch.Tensor:
 class _BNActFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
-                momentum, eps, relu, residual, process_group):
+                momentum, eps, relu, residual, process_group, slabs):
         x = _cl(x)
         if residual is not None:
             residual = _cl(residual)
@@ -55,7 +55,9 @@ class _BNActFn(torch.autograd.Function):
             sync = (process_group is not None and dist.is_initialized()
                     and dist.get_world_size(process_group) > 1)
             if sync:
-                stats = ext().bn_stats(x)                  # fp32 [2, C]
+                # conv-emitted slabs skip the stats re-read of x here too
+                stats = (ext().bn_slab_sum(slabs) if slabs is not None
+                         else ext().bn_stats(x))           # fp32 [2, C]
                 dist.all_reduce(stats, group=process_group)
                 # equal per-rank shapes by construction (DistributedSampler
                 # with drop_last) -> no count exchange, no host sync
@@ -63,6 +65,14 @@ class _BNActFn(torch.autograd.Function):
                 mean, invstd = ext().bn_finalize(stats, n_total, momentum,
                                                  eps, running_mean,
                                                  running_var, True)
+            elif slabs is not None:
+                # conv->BN stats fusion (conv2d_fwd_stats): finalize the
+                # conv's per-tile (sum, sumsq) partials — no y re-read
+                n_total = float(n_local)
+                mean, invstd = ext().bn_finalize_slabs(slabs, n_total,
+                                                       momentum, eps,
+                                                       running_mean,
+                                                       running_var)
             elif _BNF_ENV == "1" and ext().bn_fwd_fused_eligible(
                     C, x.element_size()):
                 n_total = float(n_local)
@@ -106,7 +116,7 @@ class _BNActFn(torch.autograd.Function):
         dgamma = sums[1].to(gamma.dtype)
         dbeta = sums[0].to(gamma.dtype)
         return (dx, dgamma, dbeta, None, None, None, None, None, None,
-                dres if has_res else None, None)
+                dres if has_res else None, None, None)
 
 
 def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
@@ -115,9 +125,17 @@ def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
            relu: bool = True, residual: Optional[torch.Tensor] = None,
            process_group=None) -> torch.Tensor:
     if x.is_cuda:
+        # conv-emitted stat slabs (sodkit.ops.conv._apply_conv) ride the
+        # producing tensor; consume-and-remove so they can never go stale
+        slabs = getattr(x, "_sodkit_stats", None)
+        if slabs is not None:
+            del x._sodkit_stats
+            if not (training and slabs.dim() == 2
+                    and slabs.shape[1] == 2 * x.shape[1]):
+                slabs = None
         return _BNActFn.apply(x, gamma, beta, running_mean, running_var,
                               training, momentum, eps, relu, residual,
-                              process_group)
+                              process_group, slabs)
     return reference.bn_act(x, gamma, beta, running_mean, running_var,
                             training, momentum, eps, relu, residual,
                             process_group)

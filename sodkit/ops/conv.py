@@ -121,7 +121,7 @@ def _cl(x: torch.Tensor) -> torch.Tensor:
 
 class _ConvFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding, dilation):
+    def forward(ctx, x, w, bias, stride, padding, dilation, emit_stats):
         x, ldx = _cl_ld(x)
         b = bias if bias is not None else torch.Tensor()
         if w.shape[0] % 256 == 0 and w.shape[3] % 8 == 0:
@@ -133,20 +133,30 @@ class _ConvFn(torch.autograd.Function):
             use8 = _use_8p(m_tiles, w.shape[0])
         else:
             use8 = False
+        slabs = None
         if use8:
             y = ext().conv2d_fwd_8p(x, w, b, stride, padding, dilation, ldx)
         elif _use_conv_dir(w.shape[3], w.shape[0],
                            w.shape[1] == 3 and w.shape[2] == 3,
                            x, stride, padding, dilation):
             y = ext().conv2d_fwd_dir(x, w, b, ldx)
+        elif emit_stats:
+            # ES epilogue: the kernel also writes per-channel BN stat
+            # partials [tiles_m, 2*Cout] so a following BatchNorm skips its
+            # stats re-read of y entirely (docs/KERNELS.md)
+            y, slabs = ext().conv2d_fwd_stats(x, w, b, stride, padding,
+                                              dilation, ldx)
         else:
             y = ext().conv2d_fwd(x, w, b, stride, padding, dilation, ldx)
+        if slabs is None:
+            slabs = torch.empty(0, device=x.device, dtype=torch.float32)
         ctx.save_for_backward(x, w)
         ctx.meta = (stride, padding, dilation, bias is not None, ldx)
-        return y
+        ctx.mark_non_differentiable(slabs)
+        return y, slabs
 
     @staticmethod
-    def backward(ctx, dy):
+    def backward(ctx, dy, _dslabs):
         x, w = ctx.saved_tensors
         stride, padding, dilation, has_bias, ldx = ctx.meta
         dy, lddy = _cl_ld(dy)
@@ -179,7 +189,22 @@ class _ConvFn(torch.autograd.Function):
                                         lddy).to(w.dtype)
         if has_bias and ctx.needs_input_grad[2]:
             db = ext().channel_sum(dy, lddy).to(w.dtype)
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None
+
+
+# conv->BN stats fusion (SODKIT_CONV_STATS=0 disables): the slabs ride the
+# output tensor as `._sodkit_stats`; sodkit.ops.bn_act consumes (and
+# removes) them when the very next op is a BatchNorm over that tensor.
+_CST_ENV = os.environ.get("SODKIT_CONV_STATS", "")
+
+
+def _apply_conv(x, w, bias, stride, padding, dilation):
+    emit = (_CST_ENV != "0" and torch.is_grad_enabled()
+            and (x.requires_grad or w.requires_grad))
+    y, slabs = _ConvFn.apply(x, w, bias, stride, padding, dilation, emit)
+    if slabs.numel():
+        y._sodkit_stats = slabs
+    return y
 
 
 def conv2d(x: torch.Tensor, w_krsc: torch.Tensor,
@@ -205,6 +230,6 @@ def conv2d(x: torch.Tensor, w_krsc: torch.Tensor,
                 import torch.nn.functional as F
                 w1 = F.pad(w1, (0, K8 - K))
             w1 = w1.reshape(w_krsc.shape[0], 1, 1, K8)
-            return _ConvFn.apply(xim, w1, bias, 1, 0, 1)
-        return _ConvFn.apply(x, w_krsc, bias, stride, padding, dilation)
+            return _apply_conv(xim, w1, bias, 1, 0, 1)
+        return _apply_conv(x, w_krsc, bias, stride, padding, dilation)
     return reference.conv2d(x, w_krsc, bias, stride, padding, dilation)

@@ -748,3 +748,95 @@ def test_conv8p_matches_default_kernel_e2e():
         conv_mod._8P_ENV = prev
     close(y0, ref, what="default vs ref")
     close(y1, ref, what="8p vs ref")
+
+
+# ---- conv->BN stats fusion (conv2d_fwd_stats ES epilogue) ----------------
+
+STATS_SHAPES = [
+    (2, 64, 56, 64, 3, 1, 1, 1),     # 3x3, single cout tile
+    (2, 256, 28, 64, 1, 1, 0, 1),    # 1x1 reduce
+    (2, 64, 28, 256, 1, 1, 0, 1),    # 1x1 expand, multiple cout tiles
+    (2, 48, 11, 512, 3, 1, 1, 1),    # Ncout tail + small M (masked rows)
+    (1, 64, 22, 9, 3, 1, 1, 1),      # ragged Ncout=9 (scalar store path)
+    (2, 256, 14, 512, 1, 2, 0, 1),   # strided
+]
+
+
+@pytest.mark.parametrize("shape", STATS_SHAPES)
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_conv2d_fwd_stats_matches_bn_stats(shape, dtype):
+    """The ES epilogue's slabs, reduced, must equal the per-channel
+    sum/sumsq that bn_stats reads back from the stored y (identical values,
+    different fp32 summation order)."""
+    from sodkit.ops._backend import ext
+    N, Cin, H, Cout, k, st, pad, dil = shape
+    torch.manual_seed(3)
+    x = _cl(torch.randn(N, Cin, H, H, device="cuda").to(dtype))
+    w = (torch.randn(Cout, k, k, Cin, device="cuda") * 0.1).to(dtype)
+    b = torch.randn(Cout, device="cuda").to(dtype)
+    y1, slabs = ext().conv2d_fwd_stats(x, w, b, st, pad, dil, Cin)
+    y0 = ext().conv2d_fwd(x, w, b, st, pad, dil, Cin)
+    assert torch.equal(y1, y0), "ES epilogue must not change y"
+    red = slabs.sum(dim=0)
+    C = Cout
+    ref_sum = y0.float().sum(dim=(0, 2, 3))
+    ref_ssq = y0.float().square().sum(dim=(0, 2, 3))
+    close(red[:C], ref_sum, rtol=1e-3, atol=1e-2, what=f"sum {shape}")
+    close(red[C:], ref_ssq, rtol=1e-3, atol=1e-2, what=f"ssq {shape}")
+
+
+def test_bn_finalize_slabs_matches_stats_finalize():
+    from sodkit.ops._backend import ext
+    torch.manual_seed(4)
+    x = _cl(torch.randn(2, 64, 56, 56, device="cuda").to(torch.bfloat16))
+    w = (torch.randn(64, 3, 3, 64, device="cuda") * 0.1).to(torch.bfloat16)
+    y, slabs = ext().conv2d_fwd_stats(x, w, torch.Tensor(), 1, 1, 1, 64)
+    n = float(y.numel() // y.shape[1])
+    rm0 = torch.zeros(64, device="cuda"); rv0 = torch.ones(64, device="cuda")
+    rm1 = rm0.clone(); rv1 = rv0.clone()
+    m0, i0 = ext().bn_stats_finalize(y, n, 0.1, 1e-5, rm0, rv0)
+    m1, i1 = ext().bn_finalize_slabs(slabs, n, 0.1, 1e-5, rm1, rv1)
+    close(m1, m0, rtol=1e-4, atol=1e-5, what="mean")
+    close(i1, i0, rtol=1e-4, atol=1e-5, what="invstd")
+    close(rm1, rm0, rtol=1e-4, atol=1e-5, what="running_mean")
+    close(rv1, rv0, rtol=1e-4, atol=1e-5, what="running_var")
+    s = ext().bn_slab_sum(slabs)
+    ref = ext().bn_stats(y)
+    close(s, ref, rtol=1e-3, atol=1e-2, what="slab_sum")
+
+
+def test_conv_bn_stats_fusion_end_to_end():
+    """ConvBNReLU under grad: the conv attaches slabs, bn consumes them;
+    output and BN running stats match the unfused path closely."""
+    from sodkit.models.blocks import ConvBNReLU
+    from sodkit import ops
+    torch.manual_seed(5)
+    m1 = ConvBNReLU(64, 128, 3, padding=1).cuda().train()
+    import copy
+    m2 = copy.deepcopy(m1)
+    from sodkit.models.blocks import convert_precision
+    convert_precision(m1, torch.bfloat16)
+    convert_precision(m2, torch.bfloat16)
+    x = _cl(torch.randn(2, 64, 40, 40, device="cuda").to(torch.bfloat16))
+    x1 = x.clone().requires_grad_()
+    y1 = m1(x1)                          # fused (default)
+    # verify the attr actually flowed: conv output inside had slabs
+    yc = ops.conv2d(x1, m1.conv.weight, None, 1, 1, 1)
+    assert getattr(yc, "_sodkit_stats", None) is not None, \
+        "conv under grad must attach stats slabs"
+    from sodkit.ops import conv as conv_mod
+    prev = conv_mod._CST_ENV
+    conv_mod._CST_ENV = "0"
+    try:
+        x2 = x.clone().requires_grad_()
+        y2 = m2(x2)                      # unfused
+    finally:
+        conv_mod._CST_ENV = prev
+    close(y1, y2, rtol=5e-3, atol=5e-3, what="y fused vs split")
+    close(m1.bn.running_mean, m2.bn.running_mean, rtol=1e-3, atol=1e-4,
+          what="running_mean")
+    close(m1.bn.running_var, m2.bn.running_var, rtol=1e-3, atol=1e-4,
+          what="running_var")
+    y1.square().mean().backward()
+    y2.square().mean().backward()
+    close(x1.grad, x2.grad, rtol=5e-2, atol=5e-3, what="dx fused vs split")

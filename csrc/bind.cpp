@@ -25,6 +25,11 @@ bool conv8_eligible(int64_t cout, int64_t cin);
 torch::Tensor conv2d_fwd_8p(torch::Tensor x, torch::Tensor w,
                             torch::Tensor bias, int64_t stride, int64_t pad,
                             int64_t dil, int64_t ld);
+std::vector<torch::Tensor> conv2d_fwd_8p_stats(torch::Tensor x,
+                                               torch::Tensor w,
+                                               torch::Tensor bias,
+                                               int64_t stride, int64_t pad,
+                                               int64_t dil, int64_t ld);
 torch::Tensor conv2d_dgrad_8p(torch::Tensor dy, torch::Tensor w_t, int64_t H,
                               int64_t W, int64_t stride, int64_t pad,
                               int64_t dil, int64_t lddy);
@@ -36,6 +41,10 @@ bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
                        int64_t dil, int64_t Cin, int64_t Cout, int64_t W);
 torch::Tensor conv2d_fwd_dir(torch::Tensor x, torch::Tensor w,
                              torch::Tensor bias, int64_t ld);
+std::vector<torch::Tensor> conv2d_fwd_dir_stats(torch::Tensor x,
+                                                torch::Tensor w,
+                                                torch::Tensor bias,
+                                                int64_t ld);
 torch::Tensor conv2d_dgrad_dir(torch::Tensor dy, torch::Tensor w_t,
                                int64_t lddy);
 // VALU 1x1 wgrad (csrc/wgrad_k1.hip)
@@ -122,6 +131,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv2d_wgrad_dir", &conv2d_wgrad_dir);
     m.def("bn_stats", &bn_stats);
     m.def("conv2d_fwd_stats", &conv2d_fwd_stats);
+    m.def("conv2d_fwd_8p_stats", &conv2d_fwd_8p_stats);
+    m.def("conv2d_fwd_dir_stats", &conv2d_fwd_dir_stats);
     m.def("bn_finalize_slabs", &bn_finalize_slabs);
     m.def("bn_slab_sum", &bn_slab_sum);
     m.def("bn_fwd_fused_eligible", &bn_fwd_fused_eligible);

@@ -65,10 +65,14 @@ template <> struct Pitch<32> {
 // ROWS = output rows per sweep: 2 for W <= 96 (H even) amortizes the
 // staging and barriers over twice the MFMA — the overlapping 4-row input
 // window serves both output rows, and the A fragments are shared.
-template <typename T, int MODE, int ROWS, int CC>
+// ES=1 (fwd): also accumulate per-channel BN stat partials over the
+// ROUNDED outputs across the block's whole row sweep; one slab row per
+// block in stats_out[blocks][2*Cout] (bn_reduce_finalize layout).
+template <typename T, int MODE, int ROWS, int CC, int ES = 0>
 __global__ __launch_bounds__(512, 1)
 void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
                      const float *__restrict__ bias, T *__restrict__ out,
+                     float *__restrict__ stats_out,
                      const T *__restrict__ zero_page,
                      int N, int H, int W, int ldsrc, int ldout, int Cout,
                      int rows_per_blk, int total_rows, int nq_row) {
@@ -126,6 +130,16 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
 
     constexpr int PIX_PER_PASS = 512 / (CC / 8);
     const int passes = (W + PIX_PER_PASS - 1) / PIX_PER_PASS;
+
+    // ES: per-lane stat partials, live across the whole row sweep (+16
+    // VGPRs; the kernel is LDS-bound at 1 block/CU, not register-bound)
+    float dsum[2][4], dssq[2][4];
+    if constexpr (ES) {
+#pragma unroll
+        for (int a = 0; a < 2; ++a)
+#pragma unroll
+            for (int e = 0; e < 4; ++e) { dsum[a][e] = 0.f; dssq[a][e] = 0.f; }
+    }
 
     for (int g = r0; g < r1; g += ROWS) {
         const int n = g / H;
@@ -225,11 +239,50 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
                 for (int e = 0; e < 4; ++e) {
                     T tv = ElemTraits<T>::from_f32(acc[mf][i][e] + b4[e]);
                     v[e] = *reinterpret_cast<const short *>(&tv);
+                    if constexpr (ES) {
+                        float fr = s16_as_f32<T>(v[e]);
+                        dsum[mf][e] += fr;
+                        dssq[mf][e] += fr * fr;
+                    }
                 }
                 *reinterpret_cast<s16x4_ *>(
                     out + (obase + q) * ldout + cout) = v;
             }
         }
+    }
+
+    if constexpr (ES) {
+        struct ESR { float v[4][2][64]; };          // [wq][sum|ssq][c]
+        __shared__ char esr_raw[sizeof(ESR)];
+        ESR *esr = reinterpret_cast<ESR *>(esr_raw);
+        __syncthreads();
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+            float s4[4], q4[4];
+#pragma unroll
+            for (int e = 0; e < 4; ++e) { s4[e] = dsum[mf][e]; q4[e] = dssq[mf][e]; }
+#pragma unroll
+            for (int d = 1; d < 16; d <<= 1)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    s4[e] += __shfl_xor(s4[e], d, 64);
+                    q4[e] += __shfl_xor(q4[e], d, 64);
+                }
+            int cl = wm * 32 + mf * 16 + ((lane >> 4) << 2);
+            if ((lane & 15) == 0 && cl < Cout) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    esr->v[wq][0][cl + e] = s4[e];
+                    esr->v[wq][1][cl + e] = q4[e];
+                }
+            }
+        }
+        __syncthreads();
+        long base = (long)blockIdx.x * 2 * Cout;
+        for (int i = t; i < 2 * Cout; i += 512)
+            stats_out[base + i] =
+                ((esr->v[0][i / Cout][i % Cout] + esr->v[1][i / Cout][i % Cout]) +
+                 (esr->v[2][i / Cout][i % Cout] + esr->v[3][i / Cout][i % Cout]));
     }
 }
 
@@ -263,10 +316,11 @@ bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
 
 namespace {
 
-template <typename T, int MODE, int ROWS, int CC>
+template <typename T, int MODE, int ROWS, int CC, int ES = 0>
 void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
                        const float *bias_ptr, torch::Tensor &out,
-                       int N, int H, int W, int ldsrc, int ldout, int Cout) {
+                       int N, int H, int W, int ldsrc, int ldout, int Cout,
+                       torch::Tensor *stats = nullptr) {
     auto stream = at::hip::getCurrentHIPStream();
     const int total_rows = N * H;
     // ~2 row-sweeping blocks per CU; ROWS==2 keeps block ranges pair-aligned
@@ -276,40 +330,46 @@ void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
     const int nq_row = (W + 15) / 16;        // q fragments per output row
     const size_t shmem = 64 * (size_t)sodkit::cdir::Pitch<CC>::W
                          + (ROWS + 2) * (size_t)(W + 18) * sodkit::cdir::Pitch<CC>::X;
+    float *stats_ptr = nullptr;
+    if constexpr (ES) {
+        *stats = torch::empty({blocks, 2L * Cout},
+                              src.options().dtype(torch::kFloat32));
+        stats_ptr = stats->data_ptr<float>();
+    }
     static bool attr = false;
     if (!attr) {
         (void)hipFuncSetAttribute(
-            (const void *)sodkit::cdir::conv_dir_kernel<T, MODE, ROWS, CC>,
+            (const void *)sodkit::cdir::conv_dir_kernel<T, MODE, ROWS, CC, ES>,
             hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
         attr = true;
     }
-    hipLaunchKernelGGL((sodkit::cdir::conv_dir_kernel<T, MODE, ROWS, CC>),
+    hipLaunchKernelGGL((sodkit::cdir::conv_dir_kernel<T, MODE, ROWS, CC, ES>),
                        dim3(blocks), dim3(512), shmem, stream.stream(),
                        reinterpret_cast<const T *>(src.data_ptr()),
                        reinterpret_cast<const T *>(wts.data_ptr()), bias_ptr,
-                       reinterpret_cast<T *>(out.data_ptr()),
+                       reinterpret_cast<T *>(out.data_ptr()), stats_ptr,
                        reinterpret_cast<const T *>(cdir_zero_page(src)),
                        N, H, W, ldsrc, ldout, Cout, rows_per_blk, total_rows,
                        nq_row);
 }
 
-template <typename T, int MODE>
+template <typename T, int MODE, int ES = 0>
 void launch_conv_dir(const torch::Tensor &src, const torch::Tensor &wts,
                      const float *bias_ptr, torch::Tensor &out,
                      int N, int H, int W, int ldsrc, int ldout, int Cout,
-                     int CC) {
+                     int CC, torch::Tensor *stats = nullptr) {
     // pair rows when the window fits and pairs cannot straddle images
     bool pair = W <= 96 && H % 2 == 0;
     if (CC == 64) {
-        if (pair) launch_conv_dir_r<T, MODE, 2, 64>(src, wts, bias_ptr, out,
-                                                    N, H, W, ldsrc, ldout, Cout);
-        else launch_conv_dir_r<T, MODE, 1, 64>(src, wts, bias_ptr, out,
-                                               N, H, W, ldsrc, ldout, Cout);
+        if (pair) launch_conv_dir_r<T, MODE, 2, 64, ES>(src, wts, bias_ptr, out,
+                                                    N, H, W, ldsrc, ldout, Cout, stats);
+        else launch_conv_dir_r<T, MODE, 1, 64, ES>(src, wts, bias_ptr, out,
+                                               N, H, W, ldsrc, ldout, Cout, stats);
     } else {
-        if (pair) launch_conv_dir_r<T, MODE, 2, 32>(src, wts, bias_ptr, out,
-                                                    N, H, W, ldsrc, ldout, Cout);
-        else launch_conv_dir_r<T, MODE, 1, 32>(src, wts, bias_ptr, out,
-                                               N, H, W, ldsrc, ldout, Cout);
+        if (pair) launch_conv_dir_r<T, MODE, 2, 32, ES>(src, wts, bias_ptr, out,
+                                                    N, H, W, ldsrc, ldout, Cout, stats);
+        else launch_conv_dir_r<T, MODE, 1, 32, ES>(src, wts, bias_ptr, out,
+                                               N, H, W, ldsrc, ldout, Cout, stats);
     }
 }
 
@@ -342,6 +402,40 @@ torch::Tensor conv2d_fwd_dir(torch::Tensor x, torch::Tensor w,
                                        Cout, Cout, C);
         }));
     return out;
+}
+
+// fwd + fused BN stat partials (same contract as conv2d_fwd_stats)
+std::vector<torch::Tensor> conv2d_fwd_dir_stats(torch::Tensor x,
+                                                torch::Tensor w,
+                                                torch::Tensor bias,
+                                                int64_t ld) {
+    TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.dim() == 4 && w.dim() == 4);
+    int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    int Cout = w.size(0);
+    TORCH_CHECK(w.size(1) == 3 && w.size(2) == 3 && w.size(3) == C
+                && w.is_contiguous());
+    TORCH_CHECK(conv_dir_eligible(3, 3, 1, 1, 1, C, Cout, W_), "not eligible");
+    TORCH_CHECK(ld >= C, "bad channel pitch");
+    auto out = torch::empty({N, Cout, H, W_},
+                            x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    torch::Tensor bias32;
+    const float *bp = nullptr;
+    if (bias.defined() && bias.numel()) {
+        bias32 = bias.to(torch::kFloat32).contiguous();
+        bp = bias32.data_ptr<float>();
+    }
+    torch::Tensor slabs;
+    AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_fwd_dir_stats",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            launch_conv_dir<__hip_bfloat16, 0, 1>(x, w, bp, out, N, H, W_,
+                                                  (int)ld, Cout, Cout, C,
+                                                  &slabs);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            launch_conv_dir<__half, 0, 1>(x, w, bp, out, N, H, W_, (int)ld,
+                                          Cout, Cout, C, &slabs);
+        }));
+    return {out, slabs};
 }
 
 torch::Tensor conv2d_dgrad_dir(torch::Tensor dy, torch::Tensor w_t,

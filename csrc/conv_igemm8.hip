@@ -47,10 +47,15 @@ SOD_DEVINL int img_off(int buf, int op, int kh) {
 
 // MODE=0 fwd; MODE=1 dgrad (x=dy over the OH x OW grid, w permuted to
 // [Cin][R][S][Cout], gather inverts the conv geometry per tap).
-template <typename T, int MODE>
+// ES=1 (fwd): also emit per-channel BN stat partials over the ROUNDED
+// outputs into stats_out[tiles_m][2*Cout] (bn_reduce_finalize layout) —
+// same scheme as the 128-square igemm kernel; the l1/l2/l3 1x1 expands
+// this backend owns are the LARGEST BatchNorm inputs in the model.
+template <typename T, int MODE, int ES = 0>
 __global__ __launch_bounds__(512, 1)
 void conv8_kernel(const T *__restrict__ x, const T *__restrict__ w,
                   const float *__restrict__ bias, T *__restrict__ out,
+                  float *__restrict__ stats_out,
                   const T *__restrict__ zero_page,
                   int N_, int H, int W, int Cin, int ld, int Cout,
                   int R, int S, int stride, int pad, int dil,
@@ -201,9 +206,16 @@ void conv8_kernel(const T *__restrict__ x, const T *__restrict__ w,
     }
 
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    // ES: [wn][sum|ssq][256 c_local] combine plane (the 4 wn waves share
+    // each channel); static LDS on top of the 128 KB dynamic tile = 136 KB
+    struct ESR { float v[4][2][256]; };
+    __shared__ char esr_raw[ES ? sizeof(ESR) : 1];
+    ESR *esr = reinterpret_cast<ESR *>(esr_raw);
     const int c0 = tc * 256 + wm * 128;
 #pragma unroll
-    for (int mf = 0; mf < 8; ++mf)
+    for (int mf = 0; mf < 8; ++mf) {
+        float ssum[4] = {0.f, 0.f, 0.f, 0.f};
+        float sssq[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int nf = 0; nf < 4; ++nf) {
             long pix = tp * 256 + wn * 64 + nf * 16 + (lane & 15);
@@ -216,9 +228,42 @@ void conv8_kernel(const T *__restrict__ x, const T *__restrict__ w,
                 if (bias != nullptr) f += bias[cc + e];
                 T t = ElemTraits<T>::from_f32(f);
                 v[e] = *reinterpret_cast<const short *>(&t);
+                if constexpr (ES) {
+                    float fr = s16_as_f32<T>(v[e]);  // the ROUNDED value
+                    ssum[e] += fr;
+                    sssq[e] += fr * fr;
+                }
             }
             *reinterpret_cast<s16x4_ *>(out + pix * Cout + cc) = v;
         }
+        if constexpr (ES) {
+#pragma unroll
+            for (int d = 1; d < 16; d <<= 1)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    ssum[e] += __shfl_xor(ssum[e], d, 64);
+                    sssq[e] += __shfl_xor(sssq[e], d, 64);
+                }
+            if ((lane & 15) == 0) {
+                int cl = wm * 128 + mf * 16 + ((lane >> 4) << 2);
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    esr->v[wn][0][cl + e] = ssum[e];
+                    esr->v[wn][1][cl + e] = sssq[e];
+                }
+            }
+        }
+    }
+    if constexpr (ES) {
+        __syncthreads();
+        long base = tp * 2 * Cout;   // slab row = this block's m tile
+        for (int i = tid; i < 2 * 256; i += 512) {
+            int row = i >> 8, c = i & 255;
+            stats_out[base + (long)row * Cout + tc * 256 + c] =
+                ((esr->v[0][row][c] + esr->v[1][row][c]) +
+                 (esr->v[2][row][c] + esr->v[3][row][c]));
+        }
+    }
 }
 
 }  // namespace ig8
@@ -242,25 +287,41 @@ void launch_conv8(const torch::Tensor &src, const torch::Tensor &w,
                   const float *bias_ptr, torch::Tensor &out,
                   int N, int H, int W, int Cin, int ld, int Cout,
                   int R, int S, int stride, int pad, int dil,
-                  int OH, int OW, long M) {
+                  int OH, int OW, long M, torch::Tensor *stats = nullptr) {
     auto stream = at::hip::getCurrentHIPStream();
     constexpr size_t SHMEM = 128 * 1024;
-    static bool attr_set = false;   // per (T, MODE) instantiation
-    if (!attr_set) {
-        (void)hipFuncSetAttribute(
-            (const void *)sodkit::ig8::conv8_kernel<T, MODE>,
-            hipFuncAttributeMaxDynamicSharedMemorySize, SHMEM);
-        attr_set = true;
+    int tiles_m = (int)((M + 255) / 256);
+    dim3 grid((unsigned)tiles_m * (Cout / 256));
+    float *stats_ptr = nullptr;
+    if constexpr (MODE == 0) {
+        if (stats != nullptr) {
+            *stats = torch::empty({tiles_m, 2L * Cout},
+                                  src.options().dtype(torch::kFloat32));
+            stats_ptr = stats->data_ptr<float>();
+        }
     }
-    dim3 grid((unsigned)((M + 255) / 256) * (Cout / 256));
-    hipLaunchKernelGGL((sodkit::ig8::conv8_kernel<T, MODE>), grid, dim3(512),
-                       SHMEM, stream.stream(),
-                       reinterpret_cast<const T *>(src.data_ptr()),
-                       reinterpret_cast<const T *>(w.data_ptr()), bias_ptr,
-                       reinterpret_cast<T *>(out.data_ptr()),
-                       reinterpret_cast<const T *>(zero_page_ptr8(src)),
-                       N, H, W, Cin, ld, Cout, R, S, stride, pad, dil,
-                       OH, OW);
+    auto kern = [&](auto es_tag) {
+        constexpr int ESv = decltype(es_tag)::value;
+        static bool attr_set = false;   // per (T, MODE, ES) instantiation
+        if (!attr_set) {
+            (void)hipFuncSetAttribute(
+                (const void *)sodkit::ig8::conv8_kernel<T, MODE, ESv>,
+                hipFuncAttributeMaxDynamicSharedMemorySize, SHMEM);
+            attr_set = true;
+        }
+        hipLaunchKernelGGL((sodkit::ig8::conv8_kernel<T, MODE, ESv>), grid,
+                           dim3(512), SHMEM, stream.stream(),
+                           reinterpret_cast<const T *>(src.data_ptr()),
+                           reinterpret_cast<const T *>(w.data_ptr()), bias_ptr,
+                           reinterpret_cast<T *>(out.data_ptr()), stats_ptr,
+                           reinterpret_cast<const T *>(zero_page_ptr8(src)),
+                           N, H, W, Cin, ld, Cout, R, S, stride, pad, dil,
+                           OH, OW);
+    };
+    if (stats_ptr != nullptr)
+        kern(std::integral_constant<int, 1>{});
+    else
+        kern(std::integral_constant<int, 0>{});
 }
 
 inline int out_dim8(int in, int k, int stride, int pad, int dil) {
@@ -304,6 +365,44 @@ torch::Tensor conv2d_fwd_8p(torch::Tensor x, torch::Tensor w,
                                     Cout, R, S, stride, pad, dil, P, Q, M);
         }));
     return out;
+}
+
+// fwd + fused BN stat partials (same contract as conv2d_fwd_stats)
+std::vector<torch::Tensor> conv2d_fwd_8p_stats(torch::Tensor x,
+                                               torch::Tensor w,
+                                               torch::Tensor bias,
+                                               int64_t stride, int64_t pad,
+                                               int64_t dil, int64_t ld) {
+    TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.dim() == 4 && w.dim() == 4);
+    int N = x.size(0), C = x.size(1), H = x.size(2), W_ = x.size(3);
+    int Cout = w.size(0), R = w.size(1), S = w.size(2);
+    TORCH_CHECK(w.size(3) == C && w.is_contiguous());
+    TORCH_CHECK(conv8_eligible(Cout, C), "8p: needs Cout%256==0, Cin%8==0");
+    TORCH_CHECK(ld >= C, "bad channel pitch");
+    int P = out_dim8(H, R, stride, pad, dil);
+    int Q = out_dim8(W_, S, stride, pad, dil);
+    long M = (long)N * P * Q;
+    auto out = torch::empty({N, Cout, P, Q},
+                            x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    torch::Tensor bias32;
+    const float *bp = nullptr;
+    if (bias.defined() && bias.numel()) {
+        bias32 = bias.to(torch::kFloat32).contiguous();
+        bp = bias32.data_ptr<float>();
+    }
+    torch::Tensor slabs;
+    AT_DISPATCH_SWITCH(x.scalar_type(), "conv2d_fwd_8p_stats",
+        AT_DISPATCH_CASE(at::kBFloat16, [&] {
+            launch_conv8<__hip_bfloat16, 0>(x, w, bp, out, N, H, W_, C,
+                                            (int)ld, Cout, R, S, stride, pad,
+                                            dil, P, Q, M, &slabs);
+        })
+        AT_DISPATCH_CASE(at::kHalf, [&] {
+            launch_conv8<__half, 0>(x, w, bp, out, N, H, W_, C, (int)ld,
+                                    Cout, R, S, stride, pad, dil, P, Q, M,
+                                    &slabs);
+        }));
+    return {out, slabs};
 }
 
 torch::Tensor conv2d_dgrad_8p(torch::Tensor dy, torch::Tensor w_t, int64_t H,

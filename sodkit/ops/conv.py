@@ -135,11 +135,19 @@ class _ConvFn(torch.autograd.Function):
             use8 = False
         slabs = None
         if use8:
-            y = ext().conv2d_fwd_8p(x, w, b, stride, padding, dilation, ldx)
+            if emit_stats:
+                y, slabs = ext().conv2d_fwd_8p_stats(x, w, b, stride, padding,
+                                                     dilation, ldx)
+            else:
+                y = ext().conv2d_fwd_8p(x, w, b, stride, padding, dilation,
+                                        ldx)
         elif _use_conv_dir(w.shape[3], w.shape[0],
                            w.shape[1] == 3 and w.shape[2] == 3,
                            x, stride, padding, dilation):
-            y = ext().conv2d_fwd_dir(x, w, b, ldx)
+            if emit_stats:
+                y, slabs = ext().conv2d_fwd_dir_stats(x, w, b, ldx)
+            else:
+                y = ext().conv2d_fwd_dir(x, w, b, ldx)
         elif emit_stats:
             # ES epilogue: the kernel also writes per-channel BN stat
             # partials [tiles_m, 2*Cout] so a following BatchNorm skips its

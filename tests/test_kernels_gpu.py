@@ -840,3 +840,33 @@ def test_conv_bn_stats_fusion_end_to_end():
     y1.square().mean().backward()
     y2.square().mean().backward()
     close(x1.grad, x2.grad, rtol=5e-2, atol=5e-3, what="dx fused vs split")
+
+
+@pytest.mark.parametrize("shape,fn", [
+    ((2, 64, 88, 256, 1, 1, 0, 1), "conv2d_fwd_8p_stats"),   # l1 expand (ig8)
+    ((2, 128, 44, 512, 1, 1, 0, 1), "conv2d_fwd_8p_stats"),  # l2 expand
+    ((1, 64, 176, 64, 3, 1, 1, 1), "conv2d_fwd_dir_stats"),  # decoder W=176
+    ((2, 64, 88, 64, 3, 1, 1, 1), "conv2d_fwd_dir_stats"),   # ROWS=2 pair
+    ((1, 32, 96, 32, 3, 1, 1, 1), "conv2d_fwd_dir_stats"),   # CC=32
+])
+def test_backend_stats_epilogues(shape, fn):
+    """ig8 / conv_dir ES epilogues: slabs reduce to the per-channel
+    sum/sumsq of the stored y (the largest BN inputs ride these backends)."""
+    from sodkit.ops._backend import ext
+    N, Cin, H, Cout, k, st, pad, dil = shape
+    torch.manual_seed(6)
+    x = _cl(torch.randn(N, Cin, H, H, device="cuda").to(torch.bfloat16))
+    w = (torch.randn(Cout, k, k, Cin, device="cuda") * 0.1).to(torch.bfloat16)
+    b = torch.randn(Cout, device="cuda").to(torch.bfloat16)
+    if fn == "conv2d_fwd_8p_stats":
+        y1, slabs = ext().conv2d_fwd_8p_stats(x, w, b, st, pad, dil, Cin)
+        y0 = ext().conv2d_fwd_8p(x, w, b, st, pad, dil, Cin)
+    else:
+        y1, slabs = ext().conv2d_fwd_dir_stats(x, w, b, Cin)
+        y0 = ext().conv2d_fwd_dir(x, w, b, Cin)
+    assert torch.equal(y1, y0), "ES epilogue must not change y"
+    red = slabs.sum(dim=0)
+    ref_sum = y0.float().sum(dim=(0, 2, 3))
+    ref_ssq = y0.float().square().sum(dim=(0, 2, 3))
+    close(red[:Cout], ref_sum, rtol=1e-3, atol=1e-2, what=f"sum {shape}")
+    close(red[Cout:], ref_ssq, rtol=1e-3, atol=1e-2, what=f"ssq {shape}")

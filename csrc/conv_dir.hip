@@ -253,7 +253,7 @@ void conv_dir_kernel(const T *__restrict__ src, const T *__restrict__ wts,
 
     if constexpr (ES) {
         struct ESR { float v[4][2][64]; };          // [wq][sum|ssq][c]
-        __shared__ char esr_raw[sizeof(ESR)];
+        __shared__ __attribute__((aligned(16))) char esr_raw[sizeof(ESR)];
         ESR *esr = reinterpret_cast<ESR *>(esr_raw);
         __syncthreads();
 #pragma unroll

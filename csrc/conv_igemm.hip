@@ -346,7 +346,7 @@ void igemm_kernel(const T *__restrict__ B_src, const T *__restrict__ A_w,
     // ES: static-LDS combine plane for the two wv_m wave-halves that share
     // each channel quad (discarded entirely for ES=0 instantiations)
     struct ESRed { float v[2][2][BCOUT]; };  // [wv_m][sum|ssq][c_local]
-    __shared__ char es_red_raw[ES ? sizeof(ESRed) : 1];
+    __shared__ __attribute__((aligned(16))) char es_red_raw[ES ? sizeof(ESRed) : 1];
     ESRed *es_red = reinterpret_cast<ESRed *>(es_red_raw);
     if constexpr (ES) __syncthreads();  // LDS tiles done before reuse
 #pragma unroll

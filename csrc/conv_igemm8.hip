@@ -209,7 +209,7 @@ void conv8_kernel(const T *__restrict__ x, const T *__restrict__ w,
     // ES: [wn][sum|ssq][256 c_local] combine plane (the 4 wn waves share
     // each channel); static LDS on top of the 128 KB dynamic tile = 136 KB
     struct ESR { float v[4][2][256]; };
-    __shared__ char esr_raw[ES ? sizeof(ESR) : 1];
+    __shared__ __attribute__((aligned(16))) char esr_raw[ES ? sizeof(ESR) : 1];
     ESR *esr = reinterpret_cast<ESR *>(esr_raw);
     const int c0 = tc * 256 + wm * 128;
 #pragma unroll

@@ -311,7 +311,9 @@ bool conv_dir_eligible(int64_t R, int64_t S, int64_t stride, int64_t pad,
     int bufs = (W <= 96) ? 4 : 3;   // ROWS==2 window below 96
     int xp = Cin == 64 ? sodkit::cdir::Pitch<64>::X : sodkit::cdir::Pitch<32>::X;
     int wp = Cin == 64 ? sodkit::cdir::Pitch<64>::W : sodkit::cdir::Pitch<32>::W;
-    return 64 * wp + bufs * (W + 18) * xp <= 160 * 1024;
+    // margin: the ES=1 fwd variant adds a 2 KB static LDS combine plane
+    // and its dynamic opt-in is 160K-4K (see launch_conv_dir_r)
+    return 64 * wp + bufs * (W + 18) * xp <= 156 * 1024;
 }
 
 namespace {
@@ -338,9 +340,16 @@ void launch_conv_dir_r(const torch::Tensor &src, const torch::Tensor &wts,
     }
     static bool attr = false;
     if (!attr) {
+        // the ES=1 kernel carries a 2 KB static combine plane: the dynamic
+        // opt-in must leave room for it or hipFuncSetAttribute FAILS — and
+        // a swallowed failure here leaves hipErrorInvalidValue in the
+        // thread's error slot, which torch's next launch check reports as
+        // a phantom launch failure (first call of each instantiation only)
         (void)hipFuncSetAttribute(
             (const void *)sodkit::cdir::conv_dir_kernel<T, MODE, ROWS, CC, ES>,
-            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+            hipFuncAttributeMaxDynamicSharedMemorySize,
+            160 * 1024 - (ES ? 4096 : 0));
+        (void)hipGetLastError();   // never let an opt-in failure poison it
         attr = true;
     }
     hipLaunchKernelGGL((sodkit::cdir::conv_dir_kernel<T, MODE, ROWS, CC, ES>),

@@ -304,9 +304,13 @@ void launch_conv8(const torch::Tensor &src, const torch::Tensor &w,
         constexpr int ESv = decltype(es_tag)::value;
         static bool attr_set = false;   // per (T, MODE, ES) instantiation
         if (!attr_set) {
+            // 128K dynamic + 8K ES static = 136K <= 160K, so this opt-in is
+            // valid either way; clear the error slot regardless so a failure
+            // can never surface as a phantom launch error downstream
             (void)hipFuncSetAttribute(
                 (const void *)sodkit::ig8::conv8_kernel<T, MODE, ESv>,
                 hipFuncAttributeMaxDynamicSharedMemorySize, SHMEM);
+            (void)hipGetLastError();
             attr_set = true;
         }
         hipLaunchKernelGGL((sodkit::ig8::conv8_kernel<T, MODE, ESv>), grid,

@@ -161,10 +161,16 @@ class _ConvFn(torch.autograd.Function):
         ctx.save_for_backward(x, w)
         ctx.meta = (stride, padding, dilation, bias is not None, ldx)
         ctx.mark_non_differentiable(slabs)
+        # without this, autograd MATERIALIZES a zero-filled grad for the
+        # slab output on every backward (~110 FillFunctor launches/step,
+        # ~0.7 ms — seen in profiles/r2/kernel_summary_final.txt)
+        ctx.set_materialize_grads(False)
         return y, slabs
 
     @staticmethod
     def backward(ctx, dy, _dslabs):
+        if dy is None:   # materialize_grads(False): y unused in the graph
+            return None, None, None, None, None, None, None
         x, w = ctx.saved_tensors
         stride, padding, dilation, has_bias, ldx = ctx.meta
         dy, lddy = _cl_ld(dy)

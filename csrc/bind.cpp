@@ -61,6 +61,9 @@ torch::Tensor conv2d_wgrad_dir(torch::Tensor x, torch::Tensor dy,
 int64_t multicopy_chunk_bytes();
 void multicopy(torch::Tensor src, torch::Tensor dst, torch::Tensor nbytes,
                torch::Tensor segmap, torch::Tensor chunkmap);
+void multipermute(torch::Tensor src, torch::Tensor dst, torch::Tensor K,
+                  torch::Tensor C, torch::Tensor RS, torch::Tensor seg,
+                  torch::Tensor rsb, torch::Tensor ktb, torch::Tensor ctb);
 
 torch::Tensor bn_stats(torch::Tensor x);
 bool bn_fwd_fused_eligible(int64_t C, int64_t elem_size);
@@ -151,6 +154,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sgd_fused", &sgd_fused);
     m.def("multicopy_chunk_bytes", &multicopy_chunk_bytes);
     m.def("multicopy", &multicopy);
+    m.def("multipermute", &multipermute);
     // build metadata (diagnostics: confirms the in-tree gfx950 build is
     // the one loaded, e.g. in the driver's native-code check)
     m.attr("BUILD_ARCH") = "gfx950";

@@ -77,3 +77,69 @@ void multicopy(torch::Tensor src, torch::Tensor dst, torch::Tensor nbytes,
                        nbytes.data_ptr<long>(), segmap.data_ptr<int>(),
                        chunkmap.data_ptr<int>());
 }
+
+// ---------------------------------------------------------------------------
+// multipermute: batched KRSC -> CRSK weight transpose (dgrad w_t cache).
+// The dgrad A-operand is w.permute(3,1,2,0).contiguous(); materializing it
+// per conv per step was ~110 ATen copy launches (~0.64 ms).  Weights only
+// change in the fused SGD step, so the cache layer (sodkit/ops/conv.py
+// _WtCache) refreshes every cached w_t with ONE launch of this kernel
+// right after the update.  Per block: one 64x64 (k, c) tile of one (r,s)
+// slice, staged through LDS (int slots: column reads stay bank-free).
+// ---------------------------------------------------------------------------
+
+namespace sodkit {
+namespace mp {
+
+__global__ void __launch_bounds__(256)
+multipermute_kernel(const long *__restrict__ src, const long *__restrict__ dst,
+                    const int *__restrict__ Kd, const int *__restrict__ Cd,
+                    const int *__restrict__ RSd, const int *__restrict__ seg,
+                    const int *__restrict__ rsb, const int *__restrict__ ktb,
+                    const int *__restrict__ ctb) {
+    __shared__ int tile[64][65];
+    int b = blockIdx.x;
+    int s = seg[b], rs = rsb[b];
+    int k0 = ktb[b] * 64, c0 = ctb[b] * 64;
+    int K = Kd[s], C = Cd[s], RS = RSd[s];
+    const unsigned short *sp = reinterpret_cast<const unsigned short *>(src[s]);
+    unsigned short *dp = reinterpret_cast<unsigned short *>(dst[s]);
+    int tx = threadIdx.x & 63, ty = threadIdx.x >> 6;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) {
+        int k = k0 + ty + i * 4;
+        unsigned short v = 0;
+        if (k < K && c0 + tx < C)
+            v = sp[((long)k * RS + rs) * C + c0 + tx];
+        tile[ty + i * 4][tx] = v;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < 16; ++i) {
+        int c = c0 + ty + i * 4;
+        if (c < C && k0 + tx < K)
+            dp[((long)c * RS + rs) * K + k0 + tx] =
+                (unsigned short)tile[tx][ty + i * 4];
+    }
+}
+
+}  // namespace mp
+}  // namespace sodkit
+
+void multipermute(torch::Tensor src, torch::Tensor dst, torch::Tensor K,
+                  torch::Tensor C, torch::Tensor RS, torch::Tensor seg,
+                  torch::Tensor rsb, torch::Tensor ktb, torch::Tensor ctb) {
+    TORCH_CHECK(src.is_cuda() && seg.is_cuda());
+    TORCH_CHECK(src.scalar_type() == at::kLong &&
+                seg.scalar_type() == at::kInt);
+    int blocks = (int)seg.numel();
+    if (blocks == 0) return;
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(sodkit::mp::multipermute_kernel, dim3(blocks),
+                       dim3(256), 0, stream.stream(),
+                       src.data_ptr<long>(), dst.data_ptr<long>(),
+                       K.data_ptr<int>(), C.data_ptr<int>(),
+                       RS.data_ptr<int>(), seg.data_ptr<int>(),
+                       rsb.data_ptr<int>(), ktb.data_ptr<int>(),
+                       ctb.data_ptr<int>());
+}

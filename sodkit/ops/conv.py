@@ -176,8 +176,8 @@ class _ConvFn(torch.autograd.Function):
         dy, lddy = _cl_ld(dy)
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            # dgrad A-operand layout [Cin, R, S, Cout] (a small permuted copy)
-            w_t = w.permute(3, 1, 2, 0).contiguous()
+            # dgrad A-operand layout [Cin, R, S, Cout] (cached; see _WtCache)
+            w_t = _WtCache.get(w)
             if w_t.shape[0] % 256 == 0 and w_t.shape[3] % 8 == 0 and _use_8p(
                     -(-x.shape[0] * x.shape[2] * x.shape[3] // 256),
                     w_t.shape[0]):
@@ -210,6 +210,83 @@ class _ConvFn(torch.autograd.Function):
 # output tensor as `._sodkit_stats`; sodkit.ops.bn_act consumes (and
 # removes) them when the very next op is a BatchNorm over that tensor.
 _CST_ENV = os.environ.get("SODKIT_CONV_STATS", "")
+
+# dgrad w_t cache (SODKIT_WT_CACHE=0 disables): materializing
+# w.permute(3,1,2,0).contiguous() per conv per step was ~110 ATen copy
+# launches (~0.64 ms/step, profiles/r2/kernel_summary_final.txt).  Weights
+# only change inside the fused SGD, which calls refresh() — ONE
+# multipermute kernel re-transposes every cached buffer in place.  Any
+# OTHER mutation is caught: a torch-op write bumps `_version`, a `.data`
+# replacement changes `data_ptr` — both invalidate the entry and the
+# backward falls back to the eager permute (and re-registers).
+_WT_ENV = os.environ.get("SODKIT_WT_CACHE", "")
+
+
+class _WtCache:
+    token = 0
+    entries: dict = {}   # data_ptr -> entry dict
+    tables = None
+    dirty = False
+
+    @classmethod
+    def get(cls, w: torch.Tensor) -> torch.Tensor:
+        if _WT_ENV == "0" or w.element_size() != 2:
+            return w.permute(3, 1, 2, 0).contiguous()
+        key = w.data_ptr()
+        ent = cls.entries.get(key)
+        if (ent is not None and ent["version"] == w._version
+                and ent["token"] == cls.token):
+            return ent["buf"]
+        buf = w.permute(3, 1, 2, 0).contiguous()
+        cls.entries[key] = {
+            "version": w._version, "token": cls.token, "buf": buf,
+            # keep the source storage alive: if `.data` is ever replaced,
+            # a pending refresh must not read freed memory (the stale
+            # entry is then never USED — the data_ptr key has moved on)
+            "src_t": w.detach(),
+            "K": w.shape[0], "RS": w.shape[1] * w.shape[2], "C": w.shape[3],
+        }
+        cls.dirty = True
+        return buf
+
+    @classmethod
+    def refresh(cls) -> None:
+        """Called by SodSGD.step right after the weight update: bump the
+        epoch token and re-transpose every cached w_t in one launch."""
+        cls.token += 1
+        if not cls.entries:
+            return
+        if cls.dirty or cls.tables is None:
+            if torch.cuda.is_current_stream_capturing():
+                return   # stale entries -> eager fallback inside capture
+            ents = list(cls.entries.values())
+            dev = ents[0]["buf"].device
+
+            def mk(vals, dt):
+                return torch.tensor(vals, dtype=dt, device=dev)
+
+            seg, rsb, ktb, ctb = [], [], [], []
+            for i, e in enumerate(ents):
+                for rs in range(e["RS"]):
+                    for kt in range((e["K"] + 63) // 64):
+                        for ct in range((e["C"] + 63) // 64):
+                            seg.append(i)
+                            rsb.append(rs)
+                            ktb.append(kt)
+                            ctb.append(ct)
+            cls.tables = (
+                mk([e["src_t"].data_ptr() for e in ents], torch.int64),
+                mk([e["buf"].data_ptr() for e in ents], torch.int64),
+                mk([e["K"] for e in ents], torch.int32),
+                mk([e["C"] for e in ents], torch.int32),
+                mk([e["RS"] for e in ents], torch.int32),
+                mk(seg, torch.int32), mk(rsb, torch.int32),
+                mk(ktb, torch.int32), mk(ctb, torch.int32),
+            )
+            cls.dirty = False
+        ext().multipermute(*cls.tables)
+        for e in cls.entries.values():
+            e["token"] = cls.token
 
 
 def _apply_conv(x, w, bias, stride, padding, dilation):

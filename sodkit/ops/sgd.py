@@ -114,6 +114,10 @@ class SodSGD:
                 ext().sgd_fused(ptr_t, numel_t, wd_t, chunk_t, dt, lr_f,
                                 self.momentum, float(grad_scale),
                                 self.nesterov, st, lr_t)
+            # weights just changed: re-transpose every cached dgrad w_t in
+            # one launch (sodkit/ops/conv.py _WtCache)
+            from . import conv as _conv
+            _conv._WtCache.refresh()
         else:
             lr = float(lr)
             for g in self.groups:

@@ -870,3 +870,62 @@ def test_backend_stats_epilogues(shape, fn):
     ref_ssq = y0.float().square().sum(dim=(0, 2, 3))
     close(red[:Cout], ref_sum, rtol=1e-3, atol=1e-2, what=f"sum {shape}")
     close(red[Cout:], ref_ssq, rtol=1e-3, atol=1e-2, what=f"ssq {shape}")
+
+
+def test_wt_cache_refresh_and_invalidation():
+    """dgrad w_t cache: the SGD-step refresh (one multipermute launch) must
+    keep every cached transpose bit-identical to an eager permute, and any
+    torch-op weight mutation must invalidate the entry."""
+    from sodkit import ops
+    from sodkit.ops.conv import _WtCache
+    torch.manual_seed(8)
+    w = torch.nn.Parameter(
+        (torch.randn(96, 3, 3, 64, device="cuda") * 0.1).to(torch.bfloat16))
+    opt = ops.SodSGD([{"params": [w]}], momentum=0.9)
+    x = _cl(torch.randn(2, 64, 24, 24, device="cuda").to(torch.bfloat16))
+    for _ in range(3):
+        xg = x.clone().requires_grad_()
+        y = ops.conv2d(xg, w, None, 1, 1, 1)
+        y.square().mean().backward()
+        ent = _WtCache.entries.get(w.data_ptr())
+        assert ent is not None, "dgrad must register the weight"
+        assert torch.equal(ent["buf"], w.permute(3, 1, 2, 0).contiguous()), \
+            "cached w_t stale after SGD refresh"
+        opt.step(0.05, 1.0)
+    # refreshed after the last step too
+    ent = _WtCache.entries[w.data_ptr()]
+    assert torch.equal(ent["buf"], w.permute(3, 1, 2, 0).contiguous())
+    # a torch-op mutation bumps _version -> entry invalid -> fresh permute
+    with torch.no_grad():
+        w.mul_(2.0)
+    assert ent["version"] != w._version
+    got = _WtCache.get(w)
+    assert torch.equal(got, w.permute(3, 1, 2, 0).contiguous())
+
+
+def test_multipermute_ragged_shapes():
+    """direct kernel check on non-multiple-of-64 K/C (DDPM Cout=9 class)."""
+    from sodkit.ops._backend import ext
+    torch.manual_seed(9)
+    ws = [torch.randn(9, 3, 3, 64, device="cuda").to(torch.bfloat16),
+          torch.randn(130, 1, 1, 40, device="cuda").to(torch.bfloat16)]
+    bufs = [torch.empty(w.shape[3], w.shape[1], w.shape[2], w.shape[0],
+                        device="cuda", dtype=w.dtype) for w in ws]
+    seg, rsb, ktb, ctb = [], [], [], []
+    for i, w in enumerate(ws):
+        K, RS, C = w.shape[0], w.shape[1] * w.shape[2], w.shape[3]
+        for rs in range(RS):
+            for kt in range((K + 63) // 64):
+                for ct in range((C + 63) // 64):
+                    seg.append(i); rsb.append(rs); ktb.append(kt); ctb.append(ct)
+    mk = lambda v, dt: torch.tensor(v, dtype=dt, device="cuda")
+    ext().multipermute(
+        mk([w.data_ptr() for w in ws], torch.int64),
+        mk([b.data_ptr() for b in bufs], torch.int64),
+        mk([w.shape[0] for w in ws], torch.int32),
+        mk([w.shape[3] for w in ws], torch.int32),
+        mk([w.shape[1] * w.shape[2] for w in ws], torch.int32),
+        mk(seg, torch.int32), mk(rsb, torch.int32),
+        mk(ktb, torch.int32), mk(ctb, torch.int32))
+    for w, b in zip(ws, bufs):
+        assert torch.equal(b, w.permute(3, 1, 2, 0).contiguous())

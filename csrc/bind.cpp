@@ -64,6 +64,10 @@ void multicopy(torch::Tensor src, torch::Tensor dst, torch::Tensor nbytes,
 void multipermute(torch::Tensor src, torch::Tensor dst, torch::Tensor K,
                   torch::Tensor C, torch::Tensor RS, torch::Tensor seg,
                   torch::Tensor rsb, torch::Tensor ktb, torch::Tensor ctb);
+// NHWC 3x3 s2 p1 max pool with packed argmax (csrc/maxpool.hip)
+std::vector<torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x);
+torch::Tensor maxpool3x3s2_bwd(torch::Tensor dy, torch::Tensor idx,
+                               int64_t H, int64_t W);
 
 torch::Tensor bn_stats(torch::Tensor x);
 bool bn_fwd_fused_eligible(int64_t C, int64_t elem_size);
@@ -155,6 +159,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("multicopy_chunk_bytes", &multicopy_chunk_bytes);
     m.def("multicopy", &multicopy);
     m.def("multipermute", &multipermute);
+    m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);
+    m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
     // build metadata (diagnostics: confirms the in-tree gfx950 build is
     // the one loaded, e.g. in the driver's native-code check)
     m.attr("BUILD_ARCH") = "gfx950";

@@ -47,6 +47,7 @@ SRC = [
     "csrc/loss.hip",
     "csrc/sgd.hip",
     "csrc/multicopy.hip",
+    "csrc/maxpool.hip",
     "csrc/dynfilter.hip",
 ]
 

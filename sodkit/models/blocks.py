@@ -84,8 +84,9 @@ class ConvBNReLU(nn.Module):
 
 
 def maxpool2d_3x3s2(x: torch.Tensor) -> torch.Tensor:
-    """Stem maxpool (SURVEY K7 — ATen-ROCm routed; rocprof showed it cold)."""
-    return torch.nn.functional.max_pool2d(x, 3, stride=2, padding=1)
+    """Stem maxpool: custom NHWC argmax-gather pair on GPU
+    (sodkit/ops/pool.py — ATen's nhwc backward ran at ~0.3 TB/s)."""
+    return ops.maxpool2d_3x3s2(x)
 
 
 def convert_precision(model: nn.Module, dtype: torch.dtype) -> nn.Module:

@@ -929,3 +929,24 @@ def test_multipermute_ragged_shapes():
         mk(ktb, torch.int32), mk(ctb, torch.int32))
     for w, b in zip(ws, bufs):
         assert torch.equal(b, w.permute(3, 1, 2, 0).contiguous())
+
+
+@pytest.mark.parametrize("shape", [(2, 64, 176, 176), (1, 64, 88, 88),
+                                   (2, 8, 23, 31)])
+def test_maxpool3x3s2(shape):
+    """Custom stem maxpool vs torch fp32: values AND gradients (the fwd's
+    row-major strict-> argmax matches ATen's tie-break, so dx is exact)."""
+    from sodkit import ops
+    torch.manual_seed(10)
+    N, C, H, W = shape
+    x0 = torch.randn(N, C, H, W, device="cuda").to(torch.bfloat16)
+    x = _cl(x0).requires_grad_()
+    y = ops.maxpool2d_3x3s2(x)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    xr = x0.float().clone().requires_grad_()
+    yr = F.max_pool2d(xr, 3, stride=2, padding=1)
+    yr.backward(gy.float())
+    assert y.shape == yr.shape
+    assert torch.equal(y.float(), yr.detach()), "fwd values must be exact"
+    assert torch.equal(x.grad.float(), xr.grad), "dx must be exact"

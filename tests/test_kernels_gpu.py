@@ -980,7 +980,7 @@ def test_maxpool3x3s2(shape):
                                accumulate=True)
     assert torch.allclose(x.grad.float(), dx_ref, rtol=1e-2, atol=1e-2), \
         (x.grad.float() - dx_ref).abs().max().item()
-    # argmax points at a true maximum
-    xv = x0.float().permute(0, 2, 3, 1)  # N H W C
-    pick = xv.reshape(-1)[flat.reshape(-1)].view(N, P, Q, C).permute(0, 3, 1, 2)
+    # argmax points at a true maximum (flat is an NCHW linear index)
+    pick = x0.float().reshape(-1)[flat.reshape(-1)] \
+        .view(N, P, Q, C).permute(0, 3, 1, 2)
     assert torch.equal(pick, yr), "argmax must select a maximal element"

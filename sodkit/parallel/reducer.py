@@ -265,6 +265,14 @@ class GradReducer:
             else:
                 src_dev.copy_(t)  # blocking: stream-ordered after last gather
         ext().multicopy(src_dev, dst_dev, nb_dev, segmap, chunkmap)
+        # restore the public contract: after the gather, .grad IS the bucket
+        # view again (host-only pointer stores).  Any consumer that bakes
+        # .grad pointers — the fused SGD's tables, user code — sees the
+        # STABLE views, not step-transient autograd tensors.  (Dropping the
+        # last ref to an autograd grad here is safe: the caching allocator
+        # is stream-ordered, so any reuse is enqueued after the gather.)
+        for p in plist:
+            p.grad = self.grad_views[p]
 
     # ------------------------------------------------------------------
     def finalize(self) -> None:

@@ -133,3 +133,30 @@ def test_defer_missing_grad_zeroes_slot():
                        torch.zeros_like(r.grad_views[p])):
             zeroed += 1
     assert zeroed == total
+
+
+@pytest.mark.gpu
+def test_defer_with_direct_sgd_tables():
+    """GradReducer + SodSGD WITHOUT the trainer's grad_src wiring (the
+    tools/rccl_validate oracle pattern): the fused SGD bakes .grad pointer
+    tables on first step — defer mode must leave .grad pointing at the
+    STABLE bucket views after finalize, or the tables bake step-transient
+    autograd tensors and later steps read garbage (caught as NaN weights
+    in the oracle's 1-rank reference)."""
+    from sodkit.ops import SodSGD
+    torch.manual_seed(11)
+    m = _model("cuda")
+    r = GradReducer(m)
+    assert r.defer
+    opt = SodSGD([{"params": list(m.parameters()), "weight_decay": 1e-4}])
+    x = torch.randn(2, 3, 16, 16, device="cuda")
+    for _ in range(4):
+        r.zero_grad()
+        m(x).square().mean().backward()
+        r.finalize()
+        for p in m.parameters():
+            assert p.grad is not None and p.grad.data_ptr() == \
+                r.grad_views[p].data_ptr(), "finalize must restore the views"
+        opt.step(0.05, r.grad_scale)
+    ck = sum(p.detach().float().sum().item() for p in m.parameters())
+    assert ck == ck, "weights went NaN: SGD tables read stale grad memory"

@@ -125,6 +125,9 @@ class Solver:
         self._graph = None
         self._graph_warm = 0
         self._g_static = None
+        self._g_ev = None      # input-copy retirement fence (one step ahead)
+        self._g_hold = None    # keeps the in-flight batch's buffers alive
+        self._h2d_pending = []  # CPU(pinned) source tensors of enqueued H2Ds
 
     # ------------------------------------------------------------------
     def _lr_at(self, it: int) -> float:
@@ -137,6 +140,11 @@ class Solver:
         return cfg.lr
 
     def _to_device(self, t: torch.Tensor, cast: bool) -> torch.Tensor:
+        if self._graph_ok and not t.is_cuda:
+            # graph mode: the replay loop's host side runs far ahead of the
+            # device, so the loader would recycle this pinned buffer before
+            # the async H2D executes — hold it until the fence retires it
+            self._h2d_pending.append(t)
         t = t.to(self.device, non_blocking=True)
         if cast and self.is_gpu and self.dtype != torch.float32:
             t = t.to(self.dtype)
@@ -255,14 +263,30 @@ class Solver:
                               f"falling back to eager steps")
                 self._graph = None
                 self._graph_ok = False
+                self._h2d_pending = []
                 loss = self._step_body(img, mask, depth,
                                        self._lr_at(self.iter_count))
                 self.iter_count += 1
                 return loss.detach()
+        # Bound the host to ONE step ahead: without this, replay() lets the
+        # host free-run and the loader RECYCLES its pinned staging buffers
+        # before the async input copies execute — replays then train on
+        # garbage (or DMA from freed pinned memory and fault).  The event
+        # retires the PREVIOUS step's copies before this step's begin, and
+        # _g_hold keeps the previous batch's buffers alive until then.
+        if self._g_ev is None:
+            self._g_ev = torch.cuda.Event()
+        else:
+            self._g_ev.synchronize()
         self._g_img.copy_(img, non_blocking=True)
         self._g_mask.copy_(mask, non_blocking=True)
         if depth is not None:
             self._g_depth.copy_(depth, non_blocking=True)
+        self._g_ev.record()
+        # hold BOTH the device tensors and the CPU(pinned) sources of this
+        # step's H2D copies until the next fence retires them
+        self._g_hold = (self._h2d_pending, img, mask, depth)
+        self._h2d_pending = []
         self._g_lr.fill_(self._lr_at(self.iter_count))
         self._graph.replay()
         self.iter_count += 1

@@ -199,3 +199,19 @@ def test_minet_step_with_8p_backend(tmp_path):
         conv_mod._8P_ENV = prev
     assert l0 == l0 and l1 == l1 and l2 == l2, "NaN loss under 8p"
     assert l2 < l0 * 1.5, (l0, l1, l2)   # still optimizing, not diverging
+
+
+def test_hipgraph_with_loader_workers(tmp_path):
+    """Regression: under hipGraph replay the host free-runs, and the
+    loader's pinned staging buffers were recycled before the async H2D
+    copies executed — replays trained on garbage (constant loss, random
+    eval).  The one-step-ahead fence + batch holds must keep real-loader
+    graph training learning."""
+    s = _mk_solver("minet_r50_352", batch_size=4, size=96, train_samples=48,
+                   val_samples=8, epoch_num=2, output_root=str(tmp_path),
+                   hip_graph=True, num_workers=2, log_interval=4)
+    assert s._graph_ok
+    s.train()
+    assert s._graph is not None, "graph capture must have engaged"
+    res = s.test()
+    assert res["mae"] < 0.3, f"graph-mode loader training did not learn: {res}"

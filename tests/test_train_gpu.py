@@ -12,7 +12,8 @@ pytestmark = pytest.mark.gpu
 def _mk_solver(preset, **kw):
     from sodkit.config import make_config
     from sodkit.engine import Solver
-    cfg = make_config(preset, num_workers=0, **kw)
+    kw.setdefault("num_workers", 0)
+    cfg = make_config(preset, **kw)
     return Solver(cfg)
 
 

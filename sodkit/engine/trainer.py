@@ -254,7 +254,13 @@ class Solver:
                 self._g_depth = depth.clone() if depth is not None else None
                 torch.cuda.synchronize()
                 self._graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(self._graph):
+                # thread_local: the loader's pin-memory thread calls
+                # hipHostMalloc while we capture; in the default "global"
+                # mode that INVALIDATES the capture (the real-loader soak
+                # died with hipErrorStreamCaptureInvalidated).  The pin
+                # thread does no device work, so relaxing is safe.
+                with torch.cuda.graph(self._graph,
+                                      capture_error_mode="thread_local"):
                     self._g_loss = self._step_body(self._g_img, self._g_mask,
                                                    self._g_depth, self._g_lr)
                 torch.cuda.synchronize()

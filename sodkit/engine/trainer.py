@@ -394,6 +394,7 @@ class Solver:
                 sampler.set_epoch(epoch)
             loader.dataset.set_epoch(epoch)
             t_last = time.perf_counter()
+            win_imgs = 0
             for i, batch in enumerate(loader):
                 if len(batch) == 4:
                     img, depth, mask, _ = batch
@@ -405,12 +406,17 @@ class Solver:
                 mask = self._to_device(mask, cast=False)
                 loss = self.step(img, mask, depth)
                 prof_step()
-                now = time.perf_counter()
-                ips = img.shape[0] * self.world / (now - t_last)
-                t_last = now
-                ips_meter.update(ips)
+                win_imgs += img.shape[0] * self.world
                 if (i + 1) % cfg.log_interval == 0 or i + 1 == len(loader):
-                    loss_meter.update(loss.item())
+                    # .item() syncs to the last step's completion, so the
+                    # window wall-clock is honest even under hipGraph replay
+                    # (per-step timing would measure async enqueue rates)
+                    lv = loss.item()
+                    now = time.perf_counter()
+                    ips_meter.update(win_imgs / (now - t_last))
+                    t_last = now
+                    win_imgs = 0
+                    loss_meter.update(lv)
                     comm = ""
                     if self.reducer.stats_mode and self.distributed:
                         st = self.reducer.pop_stats()

@@ -53,7 +53,12 @@ class Config:
     # hipGraph whole-step capture (single-GPU, accum_steps==1): replaces
     # ~1900 per-step kernel dispatches with one graph launch.  LR feeds the
     # fused SGD through a device tensor so the poly schedule survives replay.
-    hip_graph: bool = False
+    # Default ON since late r2: real-loader training measured +14% (1148.9
+    # vs 1008.9 imgs/s, identical MAE) once capture moved to thread_local
+    # error mode and the replay loop fenced the loader's pinned buffers;
+    # multi-GPU and accum configs fall back automatically, and a failed
+    # capture falls back to eager with a log line.
+    hip_graph: bool = True
 
     # precision
     dtype: str = "bf16"                  # bf16 | fp16 | fp32 (compute dtype on GPU)

@@ -240,9 +240,15 @@ class _WtCache:
         buf = w.permute(3, 1, 2, 0).contiguous()
         cls.entries[key] = {
             "version": w._version, "token": cls.token, "buf": buf,
-            # keep the source storage alive: if `.data` is ever replaced,
-            # a pending refresh must not read freed memory (the stale
-            # entry is then never USED — the data_ptr key has moved on)
+            # keep the source storage alive — this is load-bearing TWICE:
+            # (1) a pending refresh after a `.data` swap must not read
+            # freed memory, and (2) pinning the storage means its data_ptr
+            # can never be recycled to a NEW same-shape weight, so the
+            # data_ptr KEY uniquely identifies this weight forever (a
+            # recycled pointer + fresh _version could otherwise alias a
+            # stale entry and serve another model's transposed weights).
+            # Cost: a dead model's cached weights stay resident until the
+            # process exits — noise against 288 GB of HBM.
             "src_t": w.detach(),
             "K": w.shape[0], "RS": w.shape[1] * w.shape[2], "C": w.shape[3],
         }
